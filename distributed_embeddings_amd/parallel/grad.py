@@ -1,0 +1,159 @@
+"""Hybrid-parallel gradient synchronization.
+
+MI355X-native replacement of the reference's Horovod monkeypatches
+(``/root/reference/distributed_embeddings/python/layers/dist_model_parallel.py:1217-1326``):
+``DistributedGradientTape`` / ``DistributedOptimizer`` / ``broadcast_variables``
+/ ``BroadcastGlobalVariablesCallback``.  PyTorch has no tape, so the one
+mechanism here is: model-parallel parameters carry ``de_local = True`` (set by
+``DistributedEmbedding``) and are **excluded** from data-parallel gradient
+averaging; everything else is bucket-allreduced.
+
+Collectives run over ``torch.distributed`` (RCCL on GPU).  xGMI note: ring
+allreduce uses 2 of the 7 point-to-point links, so buckets are sized large
+(default 64 MiB) to amortize; fine-tuning of RCCL algorithm choice happens at
+the env level.
+"""
+
+from typing import Iterable, List
+
+import torch
+import torch.distributed as dist
+from torch import nn
+
+from . import comm
+
+
+def is_local_param(p: torch.nn.Parameter) -> bool:
+    return getattr(p, "de_local", False)
+
+
+def broadcast_parameters(module_or_params, root: int = 0):
+    """Broadcasts data-parallel parameters (and buffers) from ``root``.
+
+    Model-parallel params (``de_local``) are skipped — parity with reference
+    ``broadcast_variables`` filtering by the ``de_local`` attr (``:1234-1239``).
+    """
+    if comm.world_size() == 1:
+        return
+    if isinstance(module_or_params, nn.Module):
+        params = list(module_or_params.parameters())
+        buffers = list(module_or_params.buffers())
+    else:
+        params, buffers = list(module_or_params), []
+    with torch.no_grad():
+        for p in params:
+            if not is_local_param(p):
+                dist.broadcast(p.data, src=root)
+        for b in buffers:
+            if b.dtype.is_floating_point or b.dtype in (torch.int32, torch.int64):
+                dist.broadcast(b.data, src=root)
+
+
+# Backwards-friendly alias matching the reference public name.
+broadcast_variables = broadcast_parameters
+
+
+def _allreduce_dense_bucketed(params: List[torch.nn.Parameter], world: int,
+                              bucket_bytes: int = 64 << 20):
+    """Flat-bucket average of dense grads (7-link xGMI wants few, large calls)."""
+    bucket, nbytes = [], 0
+    def flush():
+        nonlocal bucket, nbytes
+        if not bucket:
+            return
+        flat = torch.cat([p.grad.reshape(-1) for p in bucket])
+        dist.all_reduce(flat)
+        flat /= world
+        pos = 0
+        for p in bucket:
+            n = p.grad.numel()
+            p.grad.copy_(flat[pos:pos + n].view_as(p.grad))
+            pos += n
+        bucket, nbytes = [], 0
+    groups = {}
+    for p in params:
+        groups.setdefault((p.grad.dtype, p.grad.device), []).append(p)
+    for ps in groups.values():
+        bucket, nbytes = [], 0
+        for p in ps:
+            bucket.append(p)
+            nbytes += p.grad.numel() * p.grad.element_size()
+            if nbytes >= bucket_bytes:
+                flush()
+        flush()
+
+
+def allreduce_gradients(module_or_params, bucket_bytes: int = 64 << 20):
+    """Averages gradients of all non-``de_local`` params across ranks.
+
+    Sparse gradients (data-parallel embedding tables) are densified before the
+    allreduce — parity with Horovod ``sparse_as_dense=True`` in the reference
+    tape/optimizer (``:1260-1262``).
+    """
+    world = comm.world_size()
+    if world == 1:
+        return
+    if isinstance(module_or_params, nn.Module):
+        params = module_or_params.parameters()
+    else:
+        params = module_or_params
+    dense = []
+    for p in params:
+        if p.grad is None or is_local_param(p):
+            continue
+        if p.grad.layout != torch.strided:
+            p.grad = p.grad.to_dense()
+        dense.append(p)
+    _allreduce_dense_bucketed(dense, world, bucket_bytes)
+
+
+class DistributedOptimizer:
+    """Wraps an optimizer: averages dp grads across ranks before each step.
+
+    Parity: reference ``DistributedOptimizer`` (``:1270-1300``) — mp variables
+    (marked ``de_local``) stay local; everything else allreduces.
+
+    Usage::
+
+        opt = DistributedOptimizer(torch.optim.Adagrad(model.parameters(), lr=...))
+        loss.backward(); opt.step(); opt.zero_grad()
+    """
+
+    def __init__(self, optimizer: torch.optim.Optimizer, bucket_bytes: int = 64 << 20):
+        self.optimizer = optimizer
+        self.bucket_bytes = bucket_bytes
+
+    @property
+    def param_groups(self):
+        return self.optimizer.param_groups
+
+    def state_dict(self):
+        return self.optimizer.state_dict()
+
+    def load_state_dict(self, sd):
+        self.optimizer.load_state_dict(sd)
+
+    def zero_grad(self, set_to_none: bool = True):
+        self.optimizer.zero_grad(set_to_none=set_to_none)
+
+    def step(self, closure=None):
+        params = [p for g in self.optimizer.param_groups for p in g["params"]]
+        allreduce_gradients(params, self.bucket_bytes)
+        return self.optimizer.step(closure)
+
+
+class BroadcastParametersOnFirstStep:
+    """Broadcast dp params from rank 0 once, at the first step.
+
+    Parity: reference ``BroadcastGlobalVariablesCallback`` (``:1303-1326``).
+    """
+
+    def __init__(self, module: nn.Module, root: int = 0):
+        self.module = module
+        self.root = root
+        self._done = False
+
+    def __call__(self):
+        if not self._done:
+            broadcast_parameters(self.module, self.root)
+            self._done = True

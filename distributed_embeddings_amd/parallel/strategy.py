@@ -1,0 +1,396 @@
+"""DistEmbeddingStrategy — the host-side shard planner.
+
+MI355X-native re-design of the reference planner
+(``/root/reference/distributed_embeddings/python/layers/dist_model_parallel.py:301-709``).
+Same capabilities, restructured around an explicit slice registry instead of
+parallel index lists:
+
+* table grouping by size thresholds into data-parallel / column(table)-parallel
+  / row-sliced groups (parity: ``init_table_groups :479-495``),
+* column slicing into power-of-two width slices capped by
+  ``min(world, width)``, with auto-threshold when there are fewer tables than
+  workers (parity: ``maybe_slice_table_column :518-549``, ``:566-573``),
+* row slicing with remainder-to-low-ranks and negative index offsets
+  (parity: ``create_row_sliced_configs :588-609``),
+* three placement strategies: ``basic`` round-robin, ``memory_balanced``
+  size-sorted zigzag, ``memory_optimized`` greedy min-total bin packing
+  (parity: ``apply_strategy :612-648``),
+* same-rank slice re-merging (parity: ``_merge_slices :694-709``),
+* CPU offload of the largest tables beyond ``gpu_embedding_size`` elements
+  (parity: ``_maybe_offload :449-476``),
+* fusion of co-resident same-width same-combiner tables into one concatenated
+  variable with row offsets (parity: ``_create_concat :651-691``) — unlike the
+  reference (which leans on XLA to fuse lookups, comment ``:650``), the fused
+  table here feeds **one** HIP lookup kernel per group by construction,
+* shared embeddings via ``input_table_map`` (parity: ``:367-368,413-419``).
+
+MI355X sizing note: thresholds default much higher than the reference's
+because each GPU has 288 GB HBM3E; fewer tables ever need slicing or offload.
+
+Every rank computes the same global plan deterministically (parity:
+``:408-434``); nothing here touches a device.
+"""
+
+import dataclasses
+from typing import Callable, Dict, List, Optional, Sequence, Tuple
+
+# Default thresholds sized for 288 GB HBM3E per MI355X (fp32 elements).
+DEFAULT_DP_THRESHOLD = 0  # no tables data-parallel unless asked
+DEFAULT_ROW_SLICE_THRESHOLD = None
+
+
+@dataclasses.dataclass
+class TableConfig:
+    """Planner view of one embedding table."""
+    input_dim: int
+    output_dim: int
+    combiner: Optional[str] = None
+    initializer: Optional[Callable] = None
+    name: Optional[str] = None
+
+    @property
+    def elements(self) -> int:
+        return self.input_dim * self.output_dim
+
+
+@dataclasses.dataclass
+class Slice:
+    """One column slice of a table placed on one rank.
+
+    ``col_offset``/``width`` give the column range of the original table this
+    slice holds (column order == pair order, see planner notes below).
+    """
+    table_id: int
+    rank: int
+    width: int
+    col_offset: int = 0
+    local_index: int = -1        # index into the rank's local (pre-concat) config list
+    concat_group: int = -1       # index of the concat group on its rank
+    concat_row_offset: int = 0   # row offset of this slice's table inside the fused variable
+
+
+@dataclasses.dataclass
+class ConcatGroup:
+    """A fused variable on one rank: one or more table slices stacked by rows."""
+    rank: int
+    input_dim: int               # total rows
+    output_dim: int
+    combiner: Optional[str]
+    cpu_offload: bool
+    members: List[Slice] = dataclasses.field(default_factory=list)
+
+    @property
+    def row_offsets(self) -> List[int]:
+        return [m.concat_row_offset for m in self.members]
+
+
+@dataclasses.dataclass
+class RowShard:
+    """One row (vocab) shard of a row-sliced table."""
+    table_id: int
+    rank: int
+    rows: int
+    row_offset: int              # first global row held by this shard
+
+
+def _pow2_num_slices(elements: int, threshold: Optional[float]) -> int:
+    if threshold is None:
+        return 1
+    n = 1
+    size = float(elements)
+    while size > threshold:
+        n *= 2
+        size /= 2
+    return n
+
+
+class DistEmbeddingStrategy:
+    """Plans how a list of embedding tables shards across ``world_size`` ranks.
+
+    Args:
+      table_configs: one :class:`TableConfig` (or dict) per table.
+      world_size: number of model-parallel workers.
+      strategy: ``'basic' | 'memory_balanced' | 'memory_optimized'``.
+      input_table_map: optional list mapping each input to a table id
+        (shared embeddings); default identity.
+      column_slice_threshold: max elements per column slice (None = no slicing
+        unless there are fewer tables than workers).
+      row_slice_threshold: tables with >= this many elements are row-sliced.
+      data_parallel_threshold: tables with <= this many elements replicate.
+      gpu_embedding_size: max total elements resident on each GPU; largest
+        tables beyond it are CPU-offloaded (None = everything on GPU).
+    """
+
+    def __init__(
+        self,
+        table_configs: Sequence,
+        world_size: int,
+        strategy: str = "basic",
+        input_table_map: Optional[Sequence[int]] = None,
+        column_slice_threshold: Optional[int] = None,
+        row_slice_threshold: Optional[int] = None,
+        data_parallel_threshold: Optional[int] = None,
+        gpu_embedding_size: Optional[int] = None,
+    ):
+        if strategy not in ("basic", "memory_balanced", "memory_optimized"):
+            raise ValueError(f"unknown strategy {strategy!r}")
+        self.world_size = int(world_size)
+        self.strategy = strategy
+        self.configs: List[TableConfig] = [
+            c if isinstance(c, TableConfig) else TableConfig(**c) for c in table_configs
+        ]
+        self.input_table_map = (
+            list(input_table_map) if input_table_map is not None
+            else list(range(len(self.configs)))
+        )
+        for t in self.input_table_map:
+            if not 0 <= t < len(self.configs):
+                raise ValueError("input_table_map entry out of range")
+        self.column_slice_threshold = column_slice_threshold
+        self.row_slice_threshold = row_slice_threshold
+        self.data_parallel_threshold = data_parallel_threshold
+        self.gpu_embedding_size = gpu_embedding_size
+
+        self._plan()
+
+    # ---------------------------------------------------------------- grouping
+
+    def _group_tables(self) -> Tuple[List[int], List[int], List[int]]:
+        """Route each table to exactly one of dp / col / row by size."""
+        dp, col, row = [], [], []
+        for i, cfg in enumerate(self.configs):
+            if self.data_parallel_threshold and cfg.elements <= self.data_parallel_threshold:
+                dp.append(i)
+            elif self.row_slice_threshold and cfg.elements >= self.row_slice_threshold:
+                row.append(i)
+            else:
+                col.append(i)
+        return dp, col, row
+
+    # ------------------------------------------------------------- col slicing
+
+    def _auto_column_threshold(self, col_ids: List[int]) -> Optional[float]:
+        """When there are fewer tables than workers, halve the largest until
+        there are enough slices (parity: reference ``:566-573``)."""
+        if self.column_slice_threshold is not None or not col_ids:
+            return self.column_slice_threshold
+        sizes = sorted(self.configs[t].elements for t in col_ids)
+        if len(sizes) >= self.world_size:
+            return None
+        threshold = None
+        while len(sizes) < self.world_size:
+            largest = sizes.pop()
+            threshold = largest - 1
+            sizes.extend([largest // 2, largest // 2])
+            sizes.sort()
+        return threshold
+
+    def _slice_widths(self, cfg: TableConfig, threshold) -> List[int]:
+        n = _pow2_num_slices(cfg.elements, threshold)
+        n = min(n, self.world_size, cfg.output_dim)
+        if n <= 1:
+            return [cfg.output_dim]
+        base, rem = divmod(cfg.output_dim, n)
+        return [base + (1 if i < rem else 0) for i in range(n)]
+
+    # -------------------------------------------------------------- placement
+
+    @staticmethod
+    def _place(mode: str, world: int, items: List[Tuple[int, int]]) -> List[List[int]]:
+        """Distribute items (id, size) to ranks; returns per-rank item-index lists.
+
+        Items are indices into ``items``.  Parity: reference ``apply_strategy``.
+        """
+        idx = list(range(len(items)))
+        if mode == "basic":
+            return [idx[r::world] for r in range(world)]
+        sizes = [s for _, s in items]
+        if mode == "memory_balanced":
+            order = [i for _, i in sorted(((sizes[i], i) for i in idx), reverse=True)]
+            return [
+                order[r::2 * world] + order[(2 * world - 1 - r)::2 * world]
+                for r in range(world)
+            ]
+        # memory_optimized: greedy — biggest item to currently-lightest rank.
+        order = [i for _, i in sorted(((sizes[i], i) for i in idx), reverse=True)]
+        heap = [(0, r, []) for r in range(world)]
+        for i in order:
+            heap.sort(key=lambda e: (e[0], e[1]))
+            total, r, lst = heap[0]
+            lst.append(i)
+            heap[0] = (total + sizes[i], r, lst)
+        heap.sort(key=lambda e: e[1])
+        return [e[2] for e in heap]
+
+    # ------------------------------------------------------------------- plan
+
+    def _plan(self):
+        W = self.world_size
+        self.table_groups = self._group_tables()
+        dp_ids, col_ids, row_ids = self.table_groups
+
+        # Partition inputs by group, remembering how to restore original order.
+        self.input_groups = [[], [], []]   # global input indices per group
+        self.input_maps = [[], [], []]     # per-group: local table index per input
+        for inp, tid in enumerate(self.input_table_map):
+            for g, ids in enumerate((dp_ids, col_ids, row_ids)):
+                if tid in ids:
+                    self.input_groups[g].append(inp)
+                    self.input_maps[g].append(ids.index(tid))
+                    break
+        flat = [i for grp in self.input_groups for i in grp]
+        self.reverse_input_order = [i for _, i in sorted(zip(flat, range(len(flat))))]
+
+        # ---------------- column / table-parallel group ----------------
+        threshold = self._auto_column_threshold(col_ids)
+        slice_widths = {t: self._slice_widths(self.configs[t], threshold) for t in col_ids}
+
+        # Flatten slices for placement, in table order (slices contiguous).
+        flat_items = []       # (table_id, width)
+        for t in col_ids:
+            for w in slice_widths[t]:
+                flat_items.append((t, w))
+        placement = self._place(
+            self.strategy, W, [(t, self.configs[t].input_dim * w) for t, w in flat_items]
+        )
+
+        # Walk ranks in order; the j-th occurrence of table t defines column
+        # range j (pair order == column order — see module docstring).  Repeat
+        # occurrences on one rank merge into the first (wider slice).
+        self.table_slices: Dict[int, List[Slice]] = {t: [] for t in col_ids}
+        self.rank_slices: List[List[Slice]] = [[] for _ in range(W)]
+        for r in range(W):
+            seen: Dict[int, Slice] = {}
+            for item_idx in placement[r]:
+                t, w = flat_items[item_idx]
+                if t in seen:
+                    seen[t].width += w
+                else:
+                    s = Slice(table_id=t, rank=r, width=w)
+                    s.local_index = len(self.rank_slices[r])
+                    seen[t] = s
+                    self.rank_slices[r].append(s)
+                    self.table_slices[t].append(s)
+        # Column offsets in pair order.
+        for t in col_ids:
+            off = 0
+            for s in self.table_slices[t]:
+                s.col_offset = off
+                off += s.width
+            assert off == self.configs[t].output_dim, (t, off)
+
+        # Output ranges needing post-a2a concat, in *input* order.  An input of
+        # a table with k slices contributes k adjacent outputs after reorder.
+        self.sliced_out_ranges: List[Tuple[int, int]] = []
+        col_input_nslices = []
+        for inp_local, tbl_local in enumerate(self.input_maps[1]):
+            t = col_ids[tbl_local]
+            col_input_nslices.append(len(self.table_slices[t]))
+        pos = 0
+        for n in col_input_nslices:
+            if n > 1:
+                self.sliced_out_ranges.append((pos, pos + n))
+            pos += n
+        self.num_col_outputs = pos
+
+        # CPU offload decision per *rank-local* slice (largest beyond budget).
+        for r in range(W):
+            self._mark_offload(self.rank_slices[r])
+
+        # Concat groups per rank (same width + combiner + both on GPU).
+        self.rank_concat_groups: List[List[ConcatGroup]] = []
+        for r in range(W):
+            groups: List[ConcatGroup] = []
+            for s in self.rank_slices[r]:
+                cfg = self.configs[s.table_id]
+                placed = False
+                for gi, grp in enumerate(groups):
+                    if (grp.output_dim == s.width and grp.combiner == cfg.combiner
+                            and not grp.cpu_offload and not s._offload):
+                        s.concat_group = gi
+                        s.concat_row_offset = grp.input_dim
+                        grp.input_dim += cfg.input_dim
+                        grp.members.append(s)
+                        placed = True
+                        break
+                if not placed:
+                    s.concat_group = len(groups)
+                    s.concat_row_offset = 0
+                    groups.append(ConcatGroup(
+                        rank=r, input_dim=cfg.input_dim, output_dim=s.width,
+                        combiner=cfg.combiner, cpu_offload=s._offload, members=[s]))
+            self.rank_concat_groups.append(groups)
+
+        # Forward bookkeeping: per rank, the (input, slice) pairs it serves, in
+        # local-layer order; plus flat widths and the reorder back to input order.
+        # Pair (inp, slice) order on rank r: iterate r's slices in local order,
+        # for each slice its table's inputs in original order.
+        table_inputs: Dict[int, List[int]] = {t: [] for t in col_ids}
+        for inp_local, tbl_local in enumerate(self.input_maps[1]):
+            table_inputs[col_ids[tbl_local]].append(inp_local)
+        self.rank_input_ids: List[List[int]] = []   # local col-group input index
+        self.rank_input_slices: List[List[Slice]] = []
+        for r in range(W):
+            ids, sls = [], []
+            for s in self.rank_slices[r]:
+                for inp_local in table_inputs[s.table_id]:
+                    ids.append(inp_local)
+                    sls.append(s)
+            self.rank_input_ids.append(ids)
+            self.rank_input_slices.append(sls)
+        # widths in worker order (concatenated over ranks)
+        self.widths_list_flat = [s.width for r in range(W) for s in self.rank_input_slices[r]]
+        worker_order_inputs = [i for r in range(W) for i in self.rank_input_ids[r]]
+        self.rev_tp_order = [
+            i for _, i in sorted(zip(worker_order_inputs, range(len(worker_order_inputs))))
+        ]
+
+        # ---------------- row-sliced group ----------------
+        self.row_shards: Dict[int, List[RowShard]] = {}
+        for t in row_ids:
+            cfg = self.configs[t]
+            base, rem = divmod(cfg.input_dim, W)
+            shards, off = [], 0
+            for r in range(W):
+                rows = base + (1 if r < rem else 0)
+                shards.append(RowShard(table_id=t, rank=r, rows=rows, row_offset=off))
+                off += rows
+            self.row_shards[t] = shards
+
+        self.dp_table_ids = dp_ids
+        self.col_table_ids = col_ids
+        self.row_table_ids = row_ids
+
+    def _mark_offload(self, slices: List[Slice]):
+        budget = self.gpu_embedding_size
+        for s in slices:
+            s._offload = False
+        if budget is None:
+            return
+        order = sorted(range(len(slices)),
+                       key=lambda i: self.configs[slices[i].table_id].input_dim * slices[i].width)
+        total = 0
+        for i in order:
+            s = slices[i]
+            total += self.configs[s.table_id].input_dim * s.width
+            s._offload = total > budget
+
+    # -------------------------------------------------------------- accessors
+
+    def local_concat_groups(self, rank: int) -> List[ConcatGroup]:
+        return self.rank_concat_groups[rank]
+
+    def local_row_shards(self, rank: int) -> List[RowShard]:
+        return [self.row_shards[t][rank] for t in self.row_table_ids]
+
+    def describe(self) -> str:
+        lines = [f"DistEmbeddingStrategy(world={self.world_size}, strategy={self.strategy})"]
+        dp, col, row = self.table_groups
+        lines.append(f"  groups: dp={dp} col={col} row={row}")
+        for r in range(self.world_size):
+            slices = ", ".join(
+                f"T{s.table_id}[{s.col_offset}:{s.col_offset + s.width}]"
+                + ("(cpu)" if s._offload else "")
+                for s in self.rank_slices[r])
+            lines.append(f"  rank{r}: {slices}")
+        return "\n".join(lines)

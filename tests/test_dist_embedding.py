@@ -1,0 +1,249 @@
+"""Distributed equivalence tests (gloo, real multi-process, world=2).
+
+The core pattern mirrors the reference suite (SURVEY.md §4 / reference
+``python/layers/dist_model_parallel_test.py:244-291``): build an
+undistributed reference model and a distributed twin, inject identical
+weights through ``set_weights``-style assignment, compare forward exactly,
+then apply one SGD step on both and compare the full reassembled weights.
+"""
+
+import pytest
+import torch
+
+from conftest import run_distributed
+
+
+def _make_inputs(table_sizes, batch, hotness=1, seed=7, world=1):
+    g = torch.Generator().manual_seed(seed)
+    inputs = []
+    for size in table_sizes:
+        if hotness == 1:
+            inputs.append(torch.randint(0, size, (world * batch,), generator=g))
+        else:
+            inputs.append(torch.randint(0, size, (world * batch, hotness), generator=g))
+    return inputs
+
+
+def _ref_weights(table_sizes, width, seed=3):
+    g = torch.Generator().manual_seed(seed)
+    return [torch.randn(s, width, generator=g) for s in table_sizes]
+
+
+def _dist_forward_backward(rank, world, table_sizes, width, strategy, kwargs,
+                           hotness, combiner):
+    import distributed_embeddings_amd as de
+
+    torch.manual_seed(100 + rank)
+    tables = [de.TableConfig(s, width, combiner) for s in table_sizes]
+    model = de.DistributedEmbedding(tables, strategy=strategy, **kwargs)
+
+    weights = _ref_weights(table_sizes, width)
+    model.set_weights([w.numpy() for w in weights])
+
+    inputs = _make_inputs(table_sizes, 4, hotness, world=world)
+    local = [x[rank * 4:(rank + 1) * 4] for x in inputs]
+
+    outs = model(local)
+    loss = sum((o * o).sum() for o in outs)
+    # whole-job loss for comparison (sum over ranks)
+    loss_t = torch.tensor([float(loss)])
+    de.comm.allreduce_sum_(loss_t)
+
+    opt = de.DistributedOptimizer(torch.optim.SGD(model.parameters(), lr=0.1))
+    loss.backward()
+    opt.step()
+
+    new_weights = model.get_weights(all_ranks=True)
+    return {
+        "outs": [o.detach() for o in outs],
+        "loss": float(loss_t),
+        "weights": [torch.as_tensor(w) for w in new_weights],
+    }
+
+
+def _single_forward_backward(table_sizes, width, hotness, combiner, world):
+    import distributed_embeddings_amd as de
+    tables = [de.Embedding(s, width, combiner) for s in table_sizes]
+    weights = _ref_weights(table_sizes, width)
+    with torch.no_grad():
+        for t, w in zip(tables, weights):
+            t.weight.copy_(w)
+    inputs = _make_inputs(table_sizes, 4, hotness, world=world)
+    outs = [t(x) for t, x in zip(tables, inputs)]
+    # data-parallel grads average over ranks, and each rank's loss covers its
+    # local batch: total grad = sum over full batch. With lr scaled the same,
+    # run SGD on summed loss.
+    loss = sum((o * o).sum() for o in outs)
+    params = [t.weight for t in tables]
+    opt = torch.optim.SGD(params, lr=0.1)
+    loss.backward()
+    # dp grads in the distributed run are averaged over ranks; mp grads are
+    # exact sums. The reference sidesteps this by comparing under optimizers
+    # with allreduce-average + per-rank loss; here the undistributed twin
+    # applies grad/1 for mp tables and grad/world for dp tables.
+    opt.step()
+    return outs, [p.detach() for p in params]
+
+
+CASES = [
+    ("basic", {}, 1, None),
+    ("memory_balanced", {}, 1, None),
+    ("memory_optimized", {}, 1, None),
+    ("basic", {"column_slice_threshold": 64 * 100}, 1, None),
+    ("basic", {}, 4, "sum"),
+    ("basic", {}, 4, "mean"),
+    ("basic", {"row_slice_threshold": 1}, 1, None),
+    ("basic", {"row_slice_threshold": 1}, 4, "sum"),
+    ("basic", {"data_parallel_threshold": 70 * 16}, 1, None),
+]
+
+
+@pytest.mark.parametrize("strategy,kwargs,hotness,combiner", CASES)
+def test_forward_equivalence_world2(strategy, kwargs, hotness, combiner):
+    table_sizes = [67, 130, 259, 40]
+    width = 16
+    world = 2
+    results = run_distributed(
+        _dist_forward_backward, world=world,
+        args=(table_sizes, width, strategy, kwargs, hotness, combiner))
+
+    ref_outs, _ = _single_forward_backward(table_sizes, width, hotness, combiner, world)
+
+    for rank in range(world):
+        outs = results[rank]["outs"]
+        assert len(outs) == len(table_sizes)
+        for t, o in enumerate(outs):
+            ref = ref_outs[t][rank * 4:(rank + 1) * 4]
+            assert torch.allclose(o, ref, atol=1e-5), \
+                f"rank {rank} table {t}: max err {(o - ref).abs().max()}"
+
+
+@pytest.mark.parametrize("strategy,kwargs,hotness,combiner", [
+    ("basic", {}, 1, None),
+    ("memory_balanced", {"column_slice_threshold": 64 * 100}, 4, "sum"),
+    ("basic", {"row_slice_threshold": 100000 * 16, "data_parallel_threshold": 70 * 16}, 1, None),
+])
+def test_weights_update_equivalence_world2(strategy, kwargs, hotness, combiner):
+    """Post-SGD-step reassembled weights must match the undistributed twin.
+
+    mp tables receive exact full-batch grads; dp tables receive averaged
+    grads. The twin mimics this by scaling dp grads by 1/world.
+    """
+    table_sizes = [67, 130, 259, 40]
+    width = 16
+    world = 2
+    results = run_distributed(
+        _dist_forward_backward, world=world,
+        args=(table_sizes, width, strategy, kwargs, hotness, combiner))
+
+    import distributed_embeddings_amd as de
+    plan = de.DistEmbeddingStrategy(
+        [de.TableConfig(s, width, combiner) for s in table_sizes], world,
+        strategy=strategy, **kwargs)
+    dp_ids = set(plan.dp_table_ids)
+
+    weights = _ref_weights(table_sizes, width)
+    params = [w.clone().requires_grad_(True) for w in weights]
+    inputs = _make_inputs(table_sizes, 4, hotness, world=world)
+    outs = []
+    for t, (p, x) in enumerate(zip(params, inputs)):
+        if combiner == "sum":
+            outs.append(p[x].sum(1))
+        elif combiner == "mean":
+            outs.append(p[x].mean(1))
+        else:
+            outs.append(p[x])
+    loss = sum((o * o).sum() for o in outs)
+    loss.backward()
+    with torch.no_grad():
+        for t, p in enumerate(params):
+            scale = 1.0 / world if t in dp_ids else 1.0
+            p -= 0.1 * scale * p.grad
+
+    for rank in range(world):
+        got = results[rank]["weights"]
+        for t in range(len(table_sizes)):
+            assert torch.allclose(got[t], params[t].detach(), atol=1e-5), \
+                f"rank {rank} table {t}: max err {(got[t] - params[t]).abs().max()}"
+
+
+def _ragged_worker(rank, world):
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd import Ragged
+    tables = [de.TableConfig(50, 8, "sum"), de.TableConfig(60, 8, "sum")]
+    model = de.DistributedEmbedding(tables, strategy="basic")
+    weights = _ref_weights([50, 60], 8)
+    model.set_weights([w.numpy() for w in weights])
+    lists0 = [[[1, 2], [3], [4, 5, 6], [7]], [[8], [9, 10], [11], [12, 13]]]
+    lists1 = [[[20], [21, 22], [23], [24]], [[25, 26], [27], [28], [29]]]
+    my = lists0 if rank == 0 else lists1
+    inputs = [Ragged.from_lists(l) for l in my]
+    outs = model(inputs)
+    return [o.detach() for o in outs]
+
+
+def test_ragged_input_world2():
+    results = run_distributed(_ragged_worker, world=2)
+    weights = _ref_weights([50, 60], 8)
+    all_lists = [
+        [[1, 2], [3], [4, 5, 6], [7], [20], [21, 22], [23], [24]],
+        [[8], [9, 10], [11], [12, 13], [25, 26], [27], [28], [29]],
+    ]
+    for rank in range(2):
+        for t in range(2):
+            for i in range(4):
+                row = all_lists[t][rank * 4 + i]
+                ref = weights[t][torch.tensor(row)].sum(0)
+                assert torch.allclose(results[rank][t][i], ref, atol=1e-5)
+
+
+def _unequal_batch_worker(rank, world):
+    import distributed_embeddings_amd as de
+    tables = [de.TableConfig(50, 8)]
+    model = de.DistributedEmbedding(tables, strategy="basic")
+    b = 4 if rank == 0 else 6
+    try:
+        model([torch.randint(0, 50, (b,))])
+    except ValueError:
+        return "raised"
+    return "no-error"
+
+
+def test_unequal_batch_raises():
+    results = run_distributed(_unequal_batch_worker, world=2)
+    assert all(r == "raised" for r in results)
+
+
+def _mp_input_worker(rank, world):
+    import distributed_embeddings_amd as de
+    tables = [de.TableConfig(50, 8), de.TableConfig(60, 8)]
+    model = de.DistributedEmbedding(tables, strategy="basic", dp_input=False)
+    weights = _ref_weights([50, 60], 8)
+    model.set_weights([w.numpy() for w in weights])
+    my_ids = model.local_input_ids()
+    inputs = _make_inputs([50, 60], 4, world=world)  # global batch 8
+    local_inputs = [inputs[i] for i in my_ids]
+    outs = model(local_inputs)
+    return {"outs": [o.detach() for o in outs], "my_ids": my_ids}
+
+
+def test_mp_input_mode_world2():
+    results = run_distributed(_mp_input_worker, world=2)
+    weights = _ref_weights([50, 60], 8)
+    inputs = _make_inputs([50, 60], 4, world=2)
+    for rank in range(2):
+        outs = results[rank]["outs"]
+        for t in range(2):
+            ref = weights[t][inputs[t][rank * 4:(rank + 1) * 4]]
+            assert torch.allclose(outs[t], ref, atol=1e-5)
+
+
+def test_world1_passthrough_module(seed):
+    import distributed_embeddings_amd as de
+    tables = [de.Embedding(30, 4), de.Embedding(40, 4, combiner="sum")]
+    model = de.DistributedEmbedding(tables)
+    x0 = torch.randint(0, 30, (5,))
+    x1 = torch.randint(0, 40, (5, 3))
+    outs = model([x0, x1])
+    assert torch.equal(outs[0], tables[0].weight[x0])
+    assert torch.allclose(outs[1], tables[1].weight[x1].sum(1), atol=1e-6)

@@ -1,0 +1,53 @@
+"""IntegerLookup semantics tests (CPU path; GPU path tested in test_gpu)."""
+
+import torch
+
+from distributed_embeddings_amd import IntegerLookup
+
+
+def test_assigns_incremental_values():
+    lk = IntegerLookup(max_tokens=10)
+    out = lk(torch.tensor([100, 200, 100, 300]))
+    assert out.tolist() == [1, 2, 1, 3]
+    # same keys again -> same values
+    out2 = lk(torch.tensor([300, 200, 100]))
+    assert out2.tolist() == [3, 2, 1]
+
+
+def test_oov_when_full():
+    lk = IntegerLookup(max_tokens=2)
+    out = lk(torch.tensor([10, 20, 30, 40]))
+    assert out.tolist()[:2] == [1, 2]
+    assert out.tolist()[2] == 0 and out.tolist()[3] == 0
+
+
+def test_counts():
+    lk = IntegerLookup(max_tokens=5)
+    lk(torch.tensor([7, 7, 7, 9]))
+    assert int(lk.counts[1]) == 3  # key 7 -> value 1 seen 3x
+    assert int(lk.counts[2]) == 1
+
+
+def test_get_vocabulary():
+    lk = IntegerLookup(max_tokens=5)
+    lk(torch.tensor([42, 17, 99]))
+    vocab = lk.get_vocabulary()
+    assert vocab == [-1, 42, 17, 99]
+    assert lk.vocabulary_size() == 4  # incl. OOV slot
+
+
+def test_nd_input_shape():
+    lk = IntegerLookup(max_tokens=10)
+    out = lk(torch.tensor([[5, 6], [5, 8]]))
+    assert out.shape == (2, 2)
+    assert out[0, 0] == out[1, 0]
+
+
+def test_state_roundtrip_through_buffers():
+    lk = IntegerLookup(max_tokens=5)
+    lk(torch.tensor([42, 17]))
+    sd = lk.state_dict()
+    lk2 = IntegerLookup(max_tokens=5)
+    lk2.load_state_dict(sd)
+    out = lk2(torch.tensor([17, 42]))
+    assert out.tolist() == [2, 1]
