@@ -1,0 +1,160 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: DLRM / Criteo-1TB config, hybrid data+model parallel.
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W` runs one
+training step per iteration (forward + BCE loss + backward + optimizer step)
+on synthetic power-law categorical inputs of the BASELINE.json config
+(DLRM Criteo, global batch 65536 at 8 GPUs => 8192 per GPU, weak scaling),
+random-init weights, and prints ONE JSON line from rank 0.
+
+Launch for N>1: torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N ...
+(reads RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env; RCCL backend).
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch-per-gpu", type=int, default=8192)
+    p.add_argument("--model", type=str, default="dlrm-criteo",
+                   choices=["dlrm-criteo", "synthetic-tiny", "synthetic-small",
+                            "synthetic-medium", "synthetic-large"])
+    p.add_argument("--alpha", type=float, default=1.05, help="power-law skew")
+    p.add_argument("--pool", type=int, default=8, help="pre-generated batch pool size")
+    p.add_argument("--strategy", type=str, default="memory_balanced")
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    return p.parse_args()
+
+
+def setup_dist(args):
+    import torch.distributed as dist
+    if "RANK" in os.environ and int(os.environ.get("WORLD_SIZE", "1")) > 1:
+        rank = int(os.environ["RANK"])
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+        return rank, dist.get_world_size(), local_rank
+    torch.cuda.set_device(0)
+    return 0, 1, 0
+
+
+def build_model(args, device):
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd.models.config import CRITEO_1TB_TABLE_SIZES, synthetic_models
+    from distributed_embeddings_amd.models.dlrm import DLRM
+    from distributed_embeddings_amd.models.synthetic import SyntheticModel, expand_tables
+
+    if args.model == "dlrm-criteo":
+        model = DLRM(CRITEO_1TB_TABLE_SIZES, embedding_dim=128,
+                     strategy=args.strategy).to(device)
+        table_sizes = CRITEO_1TB_TABLE_SIZES
+        hotness = [1] * len(table_sizes)
+        num_numerical = 13
+        name = "DLRM-Criteo-1TB"
+    else:
+        cfg = synthetic_models[args.model.split("-", 1)[1]]
+        model = SyntheticModel(cfg, strategy=args.strategy).to(device)
+        tables, input_map, hotness = expand_tables(cfg)
+        table_sizes = [tables[t][0] for t in input_map]
+        num_numerical = cfg.num_numerical_features
+        name = f"synthetic-{cfg.name}"
+    keep_hot = args.model != "dlrm-criteo"
+    return model, table_sizes, hotness, num_numerical, name, keep_hot
+
+
+def main():
+    args = parse_args()
+    rank, world, local_rank = setup_dist(args)
+    device = torch.device("cuda", local_rank)
+    torch.manual_seed(1234 + rank)
+
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd.utils.input_gen import make_batch
+
+    model, table_sizes, hotness, num_numerical, name, keep_hot = build_model(args, device)
+    b = args.batch_per_gpu
+
+    # pre-generated input pool (parity: reference InputGenerator batch pool)
+    pool = []
+    for i in range(args.pool):
+        g = torch.Generator(device="cpu").manual_seed(1000 + 131 * rank + i)
+        cats = [c.to(device) for c in
+                make_batch(table_sizes, hotness, b, args.alpha, generator=g,
+                           keep_hot_dim=keep_hot)]
+        num = torch.rand(b, num_numerical, device=device)
+        labels = torch.randint(0, 2, (b, 1), device=device).float()
+        pool.append((num, cats, labels))
+
+    opt = de.DistributedOptimizer(
+        torch.optim.SGD(model.parameters(), lr=1e-3))
+    de.broadcast_parameters(model)
+    loss_fn = torch.nn.BCEWithLogitsLoss()
+    use_bf16 = args.dtype == "bf16"
+
+    def step(i):
+        num, cats, labels = pool[i % len(pool)]
+        opt.zero_grad(set_to_none=True)
+        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
+            logits = model(num, cats)
+            loss = loss_fn(logits.float(), labels)
+        loss.backward()
+        opt.step()
+        return loss
+
+    for i in range(args.warmup):
+        step(i)
+
+    de.comm.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    de.comm.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], device=device)
+    if world > 1:
+        import torch.distributed as dist
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    global_batch = b * world
+    samples_per_sec = global_batch * args.steps / elapsed
+    baseline = 9157869.0  # 8xA100 TF32, BASELINE.md
+    if rank == 0:
+        print(json.dumps({
+            "metric": "samples/sec (whole node) DLRM-Criteo bs=64k",
+            "value": samples_per_sec,
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": samples_per_sec / baseline,
+            "dtype": args.dtype,
+            "data": "synthetic (power-law ids alpha=1.05, random-init weights)",
+            "config": {
+                "model": name,
+                "global_batch": global_batch,
+                "batch_per_gpu": b,
+                "embedding_dim": 128,
+                "parallelism": f"hybrid dp+mp (strategy={args.strategy}) x{world}",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

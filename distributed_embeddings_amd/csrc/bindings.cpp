@@ -1,0 +1,206 @@
+// Torch bindings for the gfx950 HIP kernels (in-tree extension
+// distributed_embeddings_amd._hip_ops).  Host-side orchestration of the
+// backward pipeline lives here (the one deliberate D2H sync: num_unique —
+// parity with the reference GradFunctor, embedding_lookup_kernels.cu:663-667).
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+
+#include "ops_api.h"
+
+namespace {
+
+#define CHECK_CUDA(x) TORCH_CHECK((x).is_cuda(), #x " must be on GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+
+inline hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+int log2_ceil(int64_t v) {
+  int b = 0;
+  while ((int64_t(1) << b) < v) ++b;
+  return b;
+}
+
+torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
+                                 torch::Tensor row_splits, bool mean) {
+  CHECK_CUDA(params); CHECK_CUDA(values); CHECK_CUDA(row_splits);
+  CHECK_CONTIG(params); CHECK_CONTIG(values); CHECK_CONTIG(row_splits);
+  TORCH_CHECK(params.dtype() == torch::kFloat32, "params must be fp32");
+  TORCH_CHECK(values.dtype() == torch::kInt64, "values must be int64");
+  TORCH_CHECK(row_splits.dtype() == torch::kInt64, "row_splits must be int64");
+  const int64_t num_rows = row_splits.numel() - 1;
+  const int64_t vocab = params.size(0);
+  const int width = (int)params.size(1);
+  auto out = torch::empty({num_rows, width}, params.options());
+  if (num_rows > 0) {
+    launch_csr_lookup_forward(params.data_ptr<float>(),
+                              values.data_ptr<int64_t>(),
+                              row_splits.data_ptr<int64_t>(), nullptr,
+                              out.data_ptr<float>(), num_rows, vocab, width,
+                              mean, current_stream());
+  }
+  return out;
+}
+
+torch::Tensor row_to_split(torch::Tensor rows, int64_t num_rows) {
+  CHECK_CUDA(rows); CHECK_CONTIG(rows);
+  TORCH_CHECK(rows.dtype() == torch::kInt64);
+  auto splits = torch::empty({num_rows + 1}, rows.options());
+  launch_row_to_split(rows.data_ptr<int64_t>(), rows.numel(), num_rows,
+                      splits.data_ptr<int64_t>(), current_stream());
+  return splits;
+}
+
+std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
+                                               torch::Tensor values,
+                                               torch::Tensor row_splits,
+                                               int64_t vocab, bool mean) {
+  CHECK_CUDA(grad_out); CHECK_CUDA(values); CHECK_CUDA(row_splits);
+  CHECK_CONTIG(grad_out); CHECK_CONTIG(values); CHECK_CONTIG(row_splits);
+  TORCH_CHECK(grad_out.dtype() == torch::kFloat32, "grad_out must be fp32");
+  const int64_t nnz = values.numel();
+  const int64_t num_rows = row_splits.numel() - 1;
+  const int width = (int)grad_out.size(1);
+  auto stream = current_stream();
+  auto i64 = values.options();
+  auto i32 = values.options().dtype(torch::kInt32);
+  auto f32 = grad_out.options();
+
+  if (nnz == 0) {
+    return {torch::empty({0}, i64), torch::empty({0, width}, f32)};
+  }
+
+  // 1. mask OOB -> sentinel `vocab`; iota payload.
+  auto masked = torch::empty({nnz}, i64);
+  auto iota = torch::empty({nnz}, i32);
+  launch_mask_oob_and_iota(values.data_ptr<int64_t>(), nnz, vocab,
+                           masked.data_ptr<int64_t>(), iota.data_ptr<int32_t>(),
+                           stream);
+
+  // 2. per-id row ids (+ mean weights keyed by original position).
+  auto row_ids = torch::empty({nnz}, i32);
+  torch::Tensor w;
+  float* w_ptr = nullptr;
+  if (mean) {
+    w = torch::empty({nnz}, f32);
+    w_ptr = w.data_ptr<float>();
+  }
+  launch_expand_row_ids(row_splits.data_ptr<int64_t>(), num_rows,
+                        row_ids.data_ptr<int32_t>(), w_ptr, mean, stream);
+
+  // 3. radix sort (ids, position) — end_bit covers [0, vocab] inclusive.
+  auto sorted_ids = torch::empty({nnz}, i64);
+  auto sorted_pos = torch::empty({nnz}, i32);
+  size_t temp_bytes = csr_backward_temp_bytes(nnz, vocab);
+  auto temp = torch::empty({(int64_t)temp_bytes},
+                           f32.dtype(torch::kUInt8));
+  const int end_bit = log2_ceil(vocab + 1);
+  auto err = run_sort_pairs(temp.data_ptr(), temp_bytes,
+                            masked.data_ptr<int64_t>(),
+                            sorted_ids.data_ptr<int64_t>(),
+                            iota.data_ptr<int32_t>(),
+                            sorted_pos.data_ptr<int32_t>(), nnz, end_bit,
+                            stream);
+  TORCH_CHECK(err == hipSuccess, "radix_sort_pairs failed: ",
+              hipGetErrorString(err));
+
+  // 4. permute row ids (+ weights) into sorted order, widened to i64.
+  auto srow = torch::empty({nnz}, i64);
+  torch::Tensor sw;
+  float* sw_ptr = nullptr;
+  if (mean) {
+    sw = torch::empty({nnz}, f32);
+    sw_ptr = sw.data_ptr<float>();
+  }
+  launch_gather_sorted(sorted_pos.data_ptr<int32_t>(),
+                       row_ids.data_ptr<int32_t>(), w_ptr, nnz,
+                       srow.data_ptr<int64_t>(), sw_ptr, stream);
+
+  // 5. unique-by-key via head flags + scan.
+  auto head = torch::empty({nnz}, i32);
+  auto pos = torch::empty({nnz}, i32);
+  launch_mark_heads(sorted_ids.data_ptr<int64_t>(), nnz, vocab,
+                    head.data_ptr<int32_t>(), stream);
+  err = run_inclusive_scan_i32(temp.data_ptr(), temp_bytes,
+                               head.data_ptr<int32_t>(),
+                               pos.data_ptr<int32_t>(), nnz, stream);
+  TORCH_CHECK(err == hipSuccess, "inclusive_scan failed");
+
+  auto unique_tmp = torch::empty({nnz}, i64);
+  auto seg_tmp = torch::empty({nnz + 1}, i64);
+  auto num_unique_dev = torch::zeros({1}, i32);
+  launch_scatter_unique(sorted_ids.data_ptr<int64_t>(),
+                        head.data_ptr<int32_t>(), pos.data_ptr<int32_t>(), nnz,
+                        vocab, unique_tmp.data_ptr<int64_t>(),
+                        seg_tmp.data_ptr<int64_t>(),
+                        num_unique_dev.data_ptr<int32_t>(), stream);
+  auto bounds = torch::empty({2}, i64);
+  launch_find_valid_bounds(sorted_ids.data_ptr<int64_t>(), nnz, vocab,
+                           bounds.data_ptr<int64_t>(), stream);
+  launch_set_seg_end(seg_tmp.data_ptr<int64_t>(),
+                     num_unique_dev.data_ptr<int32_t>(),
+                     bounds.data_ptr<int64_t>(), stream);
+
+  // 6. the one host sync: number of unique ids (output allocation).
+  const int64_t nu = num_unique_dev.to(torch::kCPU).item<int32_t>();
+
+  auto unique_ids = unique_tmp.narrow(0, 0, nu).contiguous();
+  auto unique_grad = torch::empty({nu, width}, f32);
+  if (nu > 0) {
+    // 7. segmented sum == forward gather-reduce over grad_out rows.
+    launch_csr_lookup_forward(grad_out.data_ptr<float>(),
+                              srow.data_ptr<int64_t>(),
+                              seg_tmp.data_ptr<int64_t>(), sw_ptr,
+                              unique_grad.data_ptr<float>(), nu,
+                              grad_out.size(0), width, /*mean=*/false, stream);
+  }
+  return {unique_ids, unique_grad};
+}
+
+torch::Tensor integer_lookup(torch::Tensor keys, torch::Tensor table_keys,
+                             torch::Tensor table_values, torch::Tensor counts,
+                             int64_t max_tokens) {
+  CHECK_CUDA(keys); CHECK_CUDA(table_keys); CHECK_CUDA(table_values);
+  CHECK_CUDA(counts);
+  CHECK_CONTIG(keys); CHECK_CONTIG(table_keys); CHECK_CONTIG(table_values);
+  CHECK_CONTIG(counts);
+  const int64_t n = keys.numel();
+  const int64_t capacity = table_keys.numel();
+  auto i32 = keys.options().dtype(torch::kInt32);
+  auto i64 = keys.options();
+  auto out = torch::empty({n}, i64);
+  if (n == 0) return out;
+  const int64_t nvals = max_tokens + 1;
+  size_t temp_bytes = integer_lookup_temp_bytes(max_tokens);
+  auto temp = torch::empty({(int64_t)temp_bytes}, i32.dtype(torch::kUInt8));
+  auto flags = torch::empty({nvals}, i32);
+  auto pos = torch::empty({nvals}, i32);
+  auto avail = torch::empty({nvals}, i64);
+  auto navail = torch::zeros({1}, i32);
+  auto next_avail = torch::zeros({1}, i32);
+  launch_integer_lookup(keys.data_ptr<int64_t>(), n,
+                        table_keys.data_ptr<int64_t>(),
+                        table_values.data_ptr<int64_t>(), capacity,
+                        counts.data_ptr<int32_t>(), max_tokens,
+                        temp.data_ptr(), temp_bytes, flags.data_ptr<int32_t>(),
+                        pos.data_ptr<int32_t>(), avail.data_ptr<int64_t>(),
+                        navail.data_ptr<int32_t>(),
+                        next_avail.data_ptr<int32_t>(),
+                        out.data_ptr<int64_t>(), current_stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("csr_lookup_forward", &csr_lookup_forward,
+        "CSR segmented gather-reduce forward (gfx950)");
+  m.def("csr_lookup_backward", &csr_lookup_backward,
+        "sparse backward: sort + unique + segmented sum (gfx950)");
+  m.def("row_to_split", &row_to_split, "COO rows -> CSR splits (gfx950)");
+  m.def("integer_lookup", &integer_lookup,
+        "open-addressing hash vocab build + lookup (gfx950)");
+}

@@ -1,0 +1,63 @@
+// Internal launcher API between embedding_ops.hip (kernels) and bindings.cpp
+// (torch glue).  gfx950-only; no CUDA-compat paths.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstddef>
+#include <cstdint>
+
+void launch_csr_lookup_forward(const float* params, const int64_t* values,
+                               const int64_t* splits, const float* per_id_w,
+                               float* out, int64_t num_rows, int64_t vocab,
+                               int width, bool mean, hipStream_t stream);
+
+void launch_row_to_split(const int64_t* rows, int64_t nnz, int64_t num_rows,
+                         int64_t* splits, hipStream_t stream);
+
+size_t csr_backward_temp_bytes(int64_t nnz, int64_t vocab);
+
+void launch_expand_row_ids(const int64_t* splits, int64_t num_rows,
+                           int32_t* row_ids, float* w, bool mean,
+                           hipStream_t stream);
+
+hipError_t run_sort_pairs(void* temp, size_t temp_bytes, const int64_t* keys_in,
+                          int64_t* keys_out, const int32_t* vals_in,
+                          int32_t* vals_out, int64_t n, int end_bit,
+                          hipStream_t stream);
+
+hipError_t run_inclusive_scan_i32(void* temp, size_t temp_bytes,
+                                  const int32_t* in, int32_t* out, int64_t n,
+                                  hipStream_t stream);
+
+void launch_mark_heads(const int64_t* sorted_ids, int64_t n, int64_t vocab,
+                       int32_t* head, hipStream_t stream);
+
+void launch_scatter_unique(const int64_t* sorted_ids, const int32_t* head,
+                           const int32_t* pos, int64_t n, int64_t vocab,
+                           int64_t* unique_ids, int64_t* seg_offsets,
+                           int32_t* num_unique, hipStream_t stream);
+
+void launch_mask_oob_and_iota(const int64_t* ids, int64_t n, int64_t vocab,
+                              int64_t* masked, int32_t* iota,
+                              hipStream_t stream);
+
+void launch_gather_sorted(const int32_t* perm, const int32_t* row_ids,
+                          const float* w, int64_t n, int64_t* srow, float* sw,
+                          hipStream_t stream);
+
+void launch_set_seg_end(int64_t* seg_offsets, const int32_t* num_unique,
+                        const int64_t* bounds, hipStream_t stream);
+
+void launch_find_valid_bounds(const int64_t* sorted_ids, int64_t n,
+                              int64_t vocab, int64_t* bounds,
+                              hipStream_t stream);
+
+void launch_integer_lookup(const int64_t* keys, int64_t n, int64_t* tkeys,
+                           int64_t* tvals, int64_t capacity, int32_t* counts,
+                           int64_t max_tokens, void* temp, size_t temp_bytes,
+                           int32_t* scratch_flags, int32_t* scratch_pos,
+                           int64_t* avail, int32_t* navail, int32_t* next_avail,
+                           int64_t* out, hipStream_t stream);
+
+size_t integer_lookup_temp_bytes(int64_t max_tokens);

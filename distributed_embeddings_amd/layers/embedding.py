@@ -58,6 +58,7 @@ class Embedding(nn.Module):
         initializer: Optional[Callable[[torch.Tensor], None]] = None,
         dtype: torch.dtype = torch.float32,
         device=None,
+        sparse_grad: bool = True,
     ):
         super().__init__()
         if input_dim <= 0 or output_dim <= 0:
@@ -67,6 +68,10 @@ class Embedding(nn.Module):
         self.input_dim = int(input_dim)
         self.output_dim = int(output_dim)
         self.combiner = combiner
+        # sparse_grad: plain-gather lookups route through the CSR kernel so the
+        # weight grad is a coalesced sparse tensor (IndexedSlices contract) —
+        # a dense grad for a 288 GB-class table is not an option.
+        self.sparse_grad = sparse_grad
         self.weight = nn.Parameter(torch.empty(input_dim, output_dim, dtype=dtype, device=device))
         (initializer or _default_init)(self.weight)
         # Internal: row-slice shards tolerate out-of-range ids (contribute a
@@ -88,7 +93,12 @@ class Embedding(nn.Module):
 
     def _gather(self, ids: torch.Tensor) -> torch.Tensor:
         flat = ids.reshape(-1)
-        if self._oob_zero:
+        if self.sparse_grad and (self.weight.requires_grad or self._oob_zero):
+            # hotness-1 CSR: same gather, sparse (IndexedSlices-style) grad,
+            # OOB ids contribute zero rows.
+            splits = torch.arange(flat.numel() + 1, device=flat.device, dtype=torch.long)
+            out = embedding_lookup(self.weight, Ragged(flat, splits), "sum")
+        elif self._oob_zero:
             valid = (flat >= 0) & (flat < self.input_dim)
             safe = torch.where(valid, flat, torch.zeros_like(flat))
             out = self.weight.index_select(0, safe)

@@ -1,0 +1,120 @@
+"""DLRM model (MI355X-native flagship).
+
+Capability parity with the reference example
+(``/root/reference/examples/dlrm/main.py:74-147`` + ``utils.py:92-113``):
+bottom MLP over dense features, per-category embedding tables behind
+``DistributedEmbedding`` when world>1, pairwise dot-product feature
+interaction (lower triangle) re-concatenated with the bottom MLP output, and
+a top MLP producing one logit.
+
+MI355X notes: the MLPs run as plain ``nn.Linear`` (hipBLASLt GEMMs) under
+bf16 autocast; ``dot_interact`` is the fused-MFMA target (torch fallback here,
+HIP kernel in ``csrc/``; the batched 27x128 @ 128x27 pairwise dots are
+MFMA-shaped work — SURVEY.md §7 step 9).
+"""
+
+import math
+from typing import List, Optional, Sequence
+
+import torch
+from torch import nn
+
+from ..layers.embedding import Embedding, scaled_uniform_init
+from ..parallel import comm
+from ..parallel.dist_embedding import DistributedEmbedding
+
+
+def dot_interact(emb_outs: List[torch.Tensor], bottom_mlp_out: torch.Tensor) -> torch.Tensor:
+    """Pairwise-dot interaction over [bottom_out] + embeddings.
+
+    Parity: reference ``utils.py:92-113`` — lower-triangular portion of the
+    [F+1, F+1] Gram matrix, then re-concat the bottom MLP output.
+    """
+    feats = torch.stack([bottom_mlp_out] + emb_outs, dim=1)  # [B, F, D]
+    gram = torch.bmm(feats, feats.transpose(1, 2))           # [B, F, F]
+    f = gram.shape[1]
+    ii, jj = torch.tril_indices(f, f, offset=-1, device=gram.device)
+    interactions = gram[:, ii, jj]                           # [B, F(F-1)/2]
+    return torch.cat([interactions, bottom_mlp_out], dim=1)
+
+
+def _mlp(sizes: Sequence[int], in_dim: int, final_linear: bool) -> nn.Sequential:
+    layers: List[nn.Module] = []
+    d = in_dim
+    for i, s in enumerate(sizes):
+        lin = nn.Linear(d, s)
+        nn.init.xavier_normal_(lin.weight)
+        nn.init.normal_(lin.bias, std=math.sqrt(1.0 / s))
+        layers.append(lin)
+        if not (final_linear and i == len(sizes) - 1):
+            layers.append(nn.ReLU(inplace=True))
+        d = s
+    return nn.Sequential(*layers)
+
+
+class DLRM(nn.Module):
+    """DLRM with hybrid data+model parallel embeddings.
+
+    Args:
+      table_sizes: vocab size per categorical feature.
+      embedding_dim: embedding width (default 128).
+      bottom_mlp_dims / top_mlp_dims: hidden sizes (reference defaults).
+      num_numerical: dense feature count (13 for Criteo).
+      strategy: DistributedEmbedding placement strategy.
+      dp_input: data-parallel input mode (see DistributedEmbedding).
+    """
+
+    def __init__(
+        self,
+        table_sizes: Sequence[int],
+        embedding_dim: int = 128,
+        bottom_mlp_dims: Sequence[int] = (512, 256, 128),
+        top_mlp_dims: Sequence[int] = (1024, 1024, 512, 256, 1),
+        num_numerical: int = 13,
+        strategy: str = "memory_balanced",
+        dp_input: bool = True,
+        column_slice_threshold: Optional[int] = None,
+        data_parallel_threshold: Optional[int] = None,
+    ):
+        super().__init__()
+        self.table_sizes = list(table_sizes)
+        self.embedding_dim = embedding_dim
+        self.distributed = comm.world_size() > 1
+        self.dp_input = dp_input
+
+        bottom_mlp_dims = list(bottom_mlp_dims)
+        if bottom_mlp_dims[-1] != embedding_dim:
+            # the interaction stacks bottom output with embeddings: dims match
+            bottom_mlp_dims[-1] = embedding_dim
+        self.bottom_mlp = _mlp(bottom_mlp_dims, num_numerical, final_linear=False)
+        num_feats = len(table_sizes) + 1
+        interact_dim = num_feats * (num_feats - 1) // 2 + bottom_mlp_dims[-1]
+        self.top_mlp = _mlp(top_mlp_dims, interact_dim, final_linear=True)
+
+        tables = [
+            Embedding(s, embedding_dim, initializer=scaled_uniform_init)
+            for s in table_sizes
+        ]
+        if self.distributed:
+            self.embeddings = DistributedEmbedding(
+                tables, strategy=strategy, dp_input=dp_input,
+                column_slice_threshold=column_slice_threshold,
+                data_parallel_threshold=data_parallel_threshold)
+        else:
+            self.embeddings = nn.ModuleList(tables)
+
+    def local_cat_feature_ids(self) -> List[int]:
+        if self.distributed and not self.dp_input:
+            return self.embeddings.local_input_ids()
+        return list(range(len(self.table_sizes)))
+
+    def forward(self, numerical: torch.Tensor,
+                cat_features: Sequence[torch.Tensor]) -> torch.Tensor:
+        bottom = self.bottom_mlp(numerical)
+        if self.distributed:
+            emb = self.embeddings(list(cat_features))
+        else:
+            emb = [e(x) for e, x in zip(self.embeddings, cat_features)]
+        emb = [e.to(bottom.dtype) for e in emb]
+        x = dot_interact(emb, bottom)
+        return self.top_mlp(x)

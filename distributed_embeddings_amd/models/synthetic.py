@@ -1,0 +1,89 @@
+"""Synthetic benchmark models.
+
+Capability parity with the reference
+(``/root/reference/examples/benchmarks/synthetic_models/synthetic_models.py:116-176``):
+expand the ``EmbeddingConfig`` list into per-table ``Embedding(combiner='sum')``
+layers + an ``input_table_map`` (shared tables get one input per hotness),
+wrap in ``DistributedEmbedding(memory_balanced)``, emulate the interaction
+with an average-pool when ``interact_stride`` is set, finish with an MLP.
+"""
+
+from typing import List, Optional
+
+import torch
+from torch import nn
+
+from ..layers.embedding import Embedding
+from ..parallel import comm
+from ..parallel.dist_embedding import DistributedEmbedding
+from .config import ModelConfig
+
+
+def expand_tables(model_config: ModelConfig):
+    """EmbeddingConfig list -> (table configs, input_table_map, hotness list)."""
+    tables, input_table_map, hotness = [], [], []
+    for cfg in model_config.embedding_configs:
+        if len(cfg.nnz) > 1 and not cfg.shared:
+            raise NotImplementedError("non-shared multi-hot tables")
+        for _ in range(cfg.num_tables):
+            t = len(tables)
+            tables.append((cfg.num_rows, cfg.width))
+            for nnz in cfg.nnz:
+                input_table_map.append(t)
+                hotness.append(nnz)
+    return tables, input_table_map, hotness
+
+
+class SyntheticModel(nn.Module):
+    def __init__(self, model_config: ModelConfig,
+                 column_slice_threshold: Optional[int] = None,
+                 dp_input: bool = True,
+                 strategy: str = "memory_balanced",
+                 data_parallel_threshold: Optional[int] = None):
+        super().__init__()
+        self.config = model_config
+        tables, input_table_map, self.hotness = expand_tables(model_config)
+        self.input_table_map = input_table_map
+        layers = [Embedding(rows, width, combiner="sum") for rows, width in tables]
+        self.distributed = comm.world_size() > 1
+        if self.distributed:
+            self.embeddings = DistributedEmbedding(
+                layers, strategy=strategy, dp_input=dp_input,
+                input_table_map=input_table_map,
+                column_slice_threshold=column_slice_threshold,
+                data_parallel_threshold=data_parallel_threshold)
+        else:
+            self.embeddings = nn.ModuleList(layers)
+
+        self.interact_stride = model_config.interact_stride
+        if self.interact_stride:
+            self.interact = nn.AvgPool1d(self.interact_stride,
+                                         stride=self.interact_stride,
+                                         ceil_mode=True)
+        else:
+            self.interact = None
+
+        total_width = sum(tables[t][1] for t in input_table_map)
+        if self.interact_stride:
+            import math
+            total_width = math.ceil(total_width / self.interact_stride)
+        mlp_in = total_width + model_config.num_numerical_features
+        mods: List[nn.Module] = []
+        d = mlp_in
+        for s in model_config.mlp_sizes:
+            mods += [nn.Linear(d, s), nn.ReLU(inplace=True)]
+            d = s
+        mods.append(nn.Linear(d, 1))
+        self.mlp = nn.Sequential(*mods)
+
+    def forward(self, numerical: torch.Tensor, cat_features) -> torch.Tensor:
+        if self.distributed:
+            embs = self.embeddings(list(cat_features))
+        else:
+            embs = [self.embeddings[t](x)
+                    for t, x in zip(self.input_table_map, cat_features)]
+        x = torch.cat([e.to(numerical.dtype) for e in embs], dim=1)
+        if self.interact is not None:
+            x = self.interact(x.unsqueeze(1)).squeeze(1)
+        x = torch.cat([x, numerical], dim=1)
+        return self.mlp(x)

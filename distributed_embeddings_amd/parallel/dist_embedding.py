@@ -443,9 +443,15 @@ class DistributedEmbedding(nn.Module):
                     metas.append((j, ids.shape, flat.numel()))
                 allids = torch.cat(flat_parts)
                 if offload:
-                    emb = layer.weight.index_select(0, allids.cpu()).to(pair_ids[pair_js[0]].device)
+                    allids_cpu = allids.cpu()
+                    splits = torch.arange(allids_cpu.numel() + 1, dtype=torch.long)
+                    emb = embedding_lookup(layer.weight, Ragged(allids_cpu, splits),
+                                           "sum").to(allids.device)
                 else:
-                    emb = layer.weight.index_select(0, allids)
+                    # hotness-1 CSR: gather with sparse (IndexedSlices) grad.
+                    splits = torch.arange(allids.numel() + 1, device=allids.device,
+                                          dtype=torch.long)
+                    emb = embedding_lookup(layer.weight, Ragged(allids, splits), "sum")
                 pos = 0
                 for j, shape, n in metas:
                     rows = shape[0]
