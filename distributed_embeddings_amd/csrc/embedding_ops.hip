@@ -479,60 +479,61 @@ __global__ void scatter_free_values(const int32_t* __restrict__ flags,
   if (i == n - 1) *navail = pos[i];
 }
 
-// Pass 2: insert-and-find.  values[] slot protocol: writer claims key slot
-// with 64-bit CAS (EMPTY = -1), then releases the value; readers spin on the
-// value sentinel (-1) with bounded backoff.
+// Pass 2a: insert.  Claims key slots with 64-bit device-scope CAS
+// (EMPTY = -1) and assigns fresh values; NO spin-waiting — a lane that loses
+// the CAS to its own key simply stops (the winner's value write becomes
+// visible at the kernel boundary, read by pass 2b).  An intra-wave
+// winner/spinner pair would deadlock under wave64 branch serialization, so
+// the insert/find split is load-bearing, not a style choice.
 #define IL_EMPTY (-1ll)
-#define IL_VPENDING (-1ll)
 
-__global__ void hash_insert_find(const int64_t* __restrict__ keys, int64_t n,
-                                 int64_t* __restrict__ tkeys,
-                                 int64_t* __restrict__ tvals, int64_t capacity,
-                                 const int64_t* __restrict__ avail,
-                                 const int32_t* __restrict__ navail,
-                                 int32_t* __restrict__ next_avail,
-                                 int32_t* __restrict__ counts,
-                                 int64_t* __restrict__ out) {
+__global__ void hash_insert(const int64_t* __restrict__ keys, int64_t n,
+                            int64_t* __restrict__ tkeys,
+                            int64_t* __restrict__ tvals, int64_t capacity,
+                            const int64_t* __restrict__ avail,
+                            const int32_t* __restrict__ navail,
+                            int32_t* __restrict__ next_avail) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const int64_t key = keys[i];
+  uint64_t slot = mix64((uint64_t)key) % (uint64_t)capacity;
+  for (int64_t probe = 0; probe < capacity; ++probe) {
+    int64_t prev = (int64_t)atomicCAS((unsigned long long*)&tkeys[slot],
+                                      (unsigned long long)IL_EMPTY,
+                                      (unsigned long long)key);
+    if (prev == IL_EMPTY) {
+      // claimed: draw the next free value (or 0 = OOV when full)
+      const int32_t a = atomicAdd(next_avail, 1);
+      tvals[slot] = (a < *navail) ? avail[a] : 0;
+      return;
+    }
+    if (prev == key) return;  // someone else inserted this key
+    slot = (slot + 1) % (uint64_t)capacity;
+  }
+}
+
+// Pass 2b: find (all values final after the insert kernel) + frequency count.
+__global__ void hash_find(const int64_t* __restrict__ keys, int64_t n,
+                          const int64_t* __restrict__ tkeys,
+                          const int64_t* __restrict__ tvals, int64_t capacity,
+                          int32_t* __restrict__ counts,
+                          int64_t* __restrict__ out) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   const int64_t key = keys[i];
   uint64_t slot = mix64((uint64_t)key) % (uint64_t)capacity;
   int64_t value = 0;
   for (int64_t probe = 0; probe < capacity; ++probe) {
-    int64_t prev = atomicCAS((unsigned long long*)&tkeys[slot],
-                             (unsigned long long)IL_EMPTY,
-                             (unsigned long long)key);
-    if (prev == IL_EMPTY) {
-      // claimed: draw the next free value (or 0 when full)
-      const int32_t a = atomicAdd(next_avail, 1);
-      int64_t v = (a < *navail) ? avail[a] : 0;
-      __hip_atomic_store(&tvals[slot], v, __ATOMIC_RELEASE,
-                         __HIP_MEMORY_SCOPE_AGENT);
-      value = v;
+    const int64_t k = tkeys[slot];
+    if (k == key) {
+      value = tvals[slot];
       break;
     }
-    if (prev == key) {
-      int64_t v;
-      do {
-        v = __hip_atomic_load(&tvals[slot], __ATOMIC_ACQUIRE,
-                              __HIP_MEMORY_SCOPE_AGENT);
-      } while (v == IL_VPENDING);
-      value = v;
-      break;
-    }
+    if (k == IL_EMPTY) break;  // absent (insert failed: table full)
     slot = (slot + 1) % (uint64_t)capacity;
   }
   atomicAdd(&counts[value], 1);
   out[i] = value;
-}
-
-// tvals must be pre-set to IL_VPENDING wherever tkeys == EMPTY; host keeps
-// tvals == -1 for empty slots (see bindings).
-__global__ void reset_pending_values(int64_t* __restrict__ tvals,
-                                     const int64_t* __restrict__ tkeys,
-                                     int64_t capacity) {
-  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < capacity && tkeys[i] == IL_EMPTY) tvals[i] = IL_VPENDING;
 }
 
 void launch_integer_lookup(const int64_t* keys, int64_t n, int64_t* tkeys,
@@ -543,8 +544,6 @@ void launch_integer_lookup(const int64_t* keys, int64_t n, int64_t* tkeys,
                            int64_t* out, hipStream_t stream) {
   const int block = 256;
   const int64_t nvals = max_tokens + 1;
-  hipLaunchKernelGGL(reset_pending_values, dim3((int)cdiv64(capacity, block)),
-                     dim3(block), 0, stream, tvals, tkeys, capacity);
   hipLaunchKernelGGL(mark_free_values, dim3((int)cdiv64(nvals, block)),
                      dim3(block), 0, stream, counts, nvals, scratch_flags);
   rocprim::inclusive_scan(temp, temp_bytes, scratch_flags, scratch_pos,
@@ -553,9 +552,11 @@ void launch_integer_lookup(const int64_t* keys, int64_t n, int64_t* tkeys,
                      dim3(block), 0, stream, scratch_flags, scratch_pos, nvals,
                      avail, navail);
   hipMemsetAsync(next_avail, 0, sizeof(int32_t), stream);
-  hipLaunchKernelGGL(hash_insert_find, dim3((int)cdiv64(n, block)), dim3(block),
-                     0, stream, keys, n, tkeys, tvals, capacity, avail, navail,
-                     next_avail, counts, out);
+  hipLaunchKernelGGL(hash_insert, dim3((int)cdiv64(n, block)), dim3(block), 0,
+                     stream, keys, n, tkeys, tvals, capacity, avail, navail,
+                     next_avail);
+  hipLaunchKernelGGL(hash_find, dim3((int)cdiv64(n, block)), dim3(block), 0,
+                     stream, keys, n, tkeys, tvals, capacity, counts, out);
 }
 
 size_t integer_lookup_temp_bytes(int64_t max_tokens) {

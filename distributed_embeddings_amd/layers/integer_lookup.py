@@ -83,10 +83,20 @@ class IntegerLookup(nn.Module):
             flat_out[i] = v
         return out
 
+    @staticmethod
+    def _mix64(k: int) -> int:
+        """splitmix64 finalizer — must match ``mix64`` in embedding_ops.hip so
+        CPU-inserted state probes identically on GPU."""
+        m = (1 << 64) - 1
+        k = (k + 0x9E3779B97F4A7C15) & m
+        k = ((k ^ (k >> 30)) * 0xBF58476D1CE4E5B9) & m
+        k = ((k ^ (k >> 27)) * 0x94D049BB133111EB) & m
+        return (k ^ (k >> 31)) & m
+
     def _cpu_insert(self, key: int, value: int):
         cap = self.capacity
-        slot = key % cap
-        while int(self.table_keys[slot]) >= 0:
+        slot = self._mix64(key & ((1 << 64) - 1)) % cap
+        while int(self.table_keys[slot]) != -1:
             slot = (slot + 1) % cap
         self.table_keys[slot] = key
         self.table_values[slot] = value
