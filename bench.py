@@ -54,15 +54,17 @@ def build_model(args, device):
     from distributed_embeddings_amd.models.synthetic import SyntheticModel, expand_tables
 
     if args.model == "dlrm-criteo":
-        model = DLRM(CRITEO_1TB_TABLE_SIZES, embedding_dim=128,
-                     strategy=args.strategy).to(device)
+        with device:  # construct tables directly in HBM (96 GB fp32)
+            model = DLRM(CRITEO_1TB_TABLE_SIZES, embedding_dim=128,
+                         strategy=args.strategy)
         table_sizes = CRITEO_1TB_TABLE_SIZES
         hotness = [1] * len(table_sizes)
         num_numerical = 13
         name = "DLRM-Criteo-1TB"
     else:
         cfg = synthetic_models[args.model.split("-", 1)[1]]
-        model = SyntheticModel(cfg, strategy=args.strategy).to(device)
+        with device:
+            model = SyntheticModel(cfg, strategy=args.strategy)
         tables, input_map, hotness = expand_tables(cfg)
         table_sizes = [tables[t][0] for t in input_map]
         num_numerical = cfg.num_numerical_features
