@@ -39,8 +39,8 @@ torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
     launch_csr_lookup_forward(params.data_ptr<float>(),
                               values.data_ptr<int64_t>(),
                               row_splits.data_ptr<int64_t>(), nullptr,
-                              out.data_ptr<float>(), num_rows, vocab, width,
-                              mean, current_stream());
+                              out.data_ptr<float>(), num_rows, values.numel(),
+                              vocab, width, mean, current_stream());
   }
   return out;
 }
@@ -154,7 +154,7 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
     launch_csr_lookup_forward(grad_out.data_ptr<float>(),
                               srow.data_ptr<int64_t>(),
                               seg_tmp.data_ptr<int64_t>(), sw_ptr,
-                              unique_grad.data_ptr<float>(), nu,
+                              unique_grad.data_ptr<float>(), nu, nnz,
                               grad_out.size(0), width, /*mean=*/false, stream);
   }
   return {unique_ids, unique_grad};

@@ -42,7 +42,11 @@ static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
 // Narrow kernel: width <= 64.  TILE = pow2 >= width; 64/TILE rows per wave.
 // Lanes tl >= width within a tile idle on loads but stay converged.
-template <int TILE, bool MEAN, bool HAS_W>
+// SPLIT: gridDim.y waves cooperate on each row's reduction (strided over the
+// segment) and combine with atomicAdd — the load-balancing answer to
+// power-law mega-segments (reference analog: blockDim.y reduction splitting,
+// embedding_lookup_kernels.cu:195-226).
+template <int TILE, bool MEAN, bool HAS_W, bool SPLIT>
 __global__ void csr_fwd_narrow(const float* __restrict__ params,
                                const int64_t* __restrict__ values,
                                const int64_t* __restrict__ splits,
@@ -56,25 +60,31 @@ __global__ void csr_fwd_narrow(const float* __restrict__ params,
   const int64_t wave_id =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const int ystep = SPLIT ? gridDim.y : 1;
+  const int y = SPLIT ? blockIdx.y : 0;
   for (int64_t base = wave_id * RPW; base < num_rows; base += n_waves * RPW) {
     const int64_t row = base + sub;
     if (row >= num_rows || tl >= width) continue;
     const int64_t s = splits[row], e = splits[row + 1];
     float acc = 0.f;
-    for (int64_t k = s; k < e; ++k) {
+    for (int64_t k = s + y; k < e; k += ystep) {
       const int64_t idx = values[k];
       if (idx < 0 || idx >= vocab) continue;
       const float w = HAS_W ? per_id_w[k] : 1.f;
       acc += w * params[idx * width + tl];
     }
     if (MEAN && e > s) acc /= (float)(e - s);
-    out[row * width + tl] = acc;
+    if (SPLIT) {
+      atomicAdd(&out[row * width + tl], acc);
+    } else {
+      out[row * width + tl] = acc;
+    }
   }
 }
 
 // Wide kernel: width > 64.  One wave per row; VEC elements per lane
 // (vectorized loads when alignment permits); outer loop over 64*VEC chunks.
-template <int VEC, bool MEAN, bool HAS_W>
+template <int VEC, bool MEAN, bool HAS_W, bool SPLIT>
 __global__ void csr_fwd_wide(const float* __restrict__ params,
                              const int64_t* __restrict__ values,
                              const int64_t* __restrict__ splits,
@@ -86,6 +96,8 @@ __global__ void csr_fwd_wide(const float* __restrict__ params,
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
   constexpr int CHUNK = WAVE * VEC;
+  const int ystep = SPLIT ? gridDim.y : 1;
+  const int y = SPLIT ? blockIdx.y : 0;
   for (int64_t row = wave_id; row < num_rows; row += n_waves) {
     const int64_t s = splits[row], e = splits[row + 1];
     const float inv = (MEAN && e > s) ? 1.f / (float)(e - s) : 1.f;
@@ -94,7 +106,7 @@ __global__ void csr_fwd_wide(const float* __restrict__ params,
 #pragma unroll
       for (int v = 0; v < VEC; ++v) acc[v] = 0.f;
       const int col0 = cbase + lane * VEC;
-      for (int64_t k = s; k < e; ++k) {
+      for (int64_t k = s + y; k < e; k += ystep) {
         const int64_t idx = values[k];
         if (idx < 0 || idx >= vocab) continue;
         const float w = HAS_W ? per_id_w[k] : 1.f;
@@ -124,7 +136,11 @@ __global__ void csr_fwd_wide(const float* __restrict__ params,
         }
       }
       float* outp = out + row * (int64_t)width + col0;
-      if (VEC == 4 && col0 + 4 <= width) {
+      if (SPLIT) {
+#pragma unroll
+        for (int v = 0; v < VEC; ++v)
+          if (col0 + v < width) atomicAdd(&outp[v], acc[v] * inv);
+      } else if (VEC == 4 && col0 + 4 <= width) {
         float4 o4 = make_float4(acc[0] * inv, acc[1] * inv, acc[2] * inv,
                                 acc[3] * inv);
         *reinterpret_cast<float4*>(outp) = o4;
@@ -147,81 +163,96 @@ static int pick_grid(int64_t work_items, int block_waves) {
   return (int)blocks;
 }
 
+// Segment-split factor: raise parallelism when rows are few and segments
+// long (tiny-vocab backward, power-law skew).  Result 1 => no atomics.
+static int pick_split(int64_t num_rows, int64_t nnz, int64_t row_waves) {
+  if (nnz <= 0 || num_rows <= 0) return 1;
+  const int64_t ave = nnz / num_rows;
+  int split = 1;
+  // target: enough waves to fill the chip (256 CUs x ~8 waves), without
+  // splitting segments below ~64 ids per wave.
+  while (split < 64 && ave / split > 64 && row_waves * split < 16384) {
+    split *= 2;
+  }
+  return split;
+}
+
+template <int TILE>
+static void launch_narrow(const float* params, const int64_t* values,
+                          const int64_t* splits, const float* per_id_w,
+                          float* out, int64_t num_rows, int64_t nnz,
+                          int64_t vocab, int width, bool mean,
+                          hipStream_t stream) {
+  const int block = 256, bw = block / WAVE;
+  const int64_t row_waves = cdiv64(num_rows, WAVE / TILE);
+  const int split = pick_split(num_rows, nnz, row_waves);
+  const dim3 grid(pick_grid(row_waves, bw), split);
+  if (split > 1)
+    hipMemsetAsync(out, 0, sizeof(float) * num_rows * width, stream);
+#define L(MEAN, HASW, SPLIT)                                                   \
+  hipLaunchKernelGGL((csr_fwd_narrow<TILE, MEAN, HASW, SPLIT>), grid,          \
+                     dim3(block), 0, stream, params, values, splits,           \
+                     per_id_w, out, num_rows, vocab, width)
+  if (split > 1) {
+    if (mean) { if (per_id_w) L(true, true, true); else L(true, false, true); }
+    else      { if (per_id_w) L(false, true, true); else L(false, false, true); }
+  } else {
+    if (mean) { if (per_id_w) L(true, true, false); else L(true, false, false); }
+    else      { if (per_id_w) L(false, true, false); else L(false, false, false); }
+  }
+#undef L
+}
+
+template <int VEC>
+static void launch_wide(const float* params, const int64_t* values,
+                        const int64_t* splits, const float* per_id_w,
+                        float* out, int64_t num_rows, int64_t nnz,
+                        int64_t vocab, int width, bool mean,
+                        hipStream_t stream) {
+  const int block = 256, bw = block / WAVE;
+  const int split = pick_split(num_rows, nnz, num_rows);
+  const dim3 grid(pick_grid(num_rows, bw), split);
+  if (split > 1)
+    hipMemsetAsync(out, 0, sizeof(float) * num_rows * width, stream);
+#define L(MEAN, HASW, SPLIT)                                                   \
+  hipLaunchKernelGGL((csr_fwd_wide<VEC, MEAN, HASW, SPLIT>), grid,             \
+                     dim3(block), 0, stream, params, values, splits,           \
+                     per_id_w, out, num_rows, vocab, width)
+  if (split > 1) {
+    if (mean) { if (per_id_w) L(true, true, true); else L(true, false, true); }
+    else      { if (per_id_w) L(false, true, true); else L(false, false, true); }
+  } else {
+    if (mean) { if (per_id_w) L(true, true, false); else L(true, false, false); }
+    else      { if (per_id_w) L(false, true, false); else L(false, false, false); }
+  }
+#undef L
+}
+
 void launch_csr_lookup_forward(const float* params, const int64_t* values,
                                const int64_t* splits, const float* per_id_w,
-                               float* out, int64_t num_rows, int64_t vocab,
-                               int width, bool mean, hipStream_t stream) {
-  const int block = 256;
-  const int bw = block / WAVE;
-
-#define DISPATCH_NARROW(TILE)                                                 \
-  do {                                                                        \
-    const int rpw = WAVE / TILE;                                              \
-    const int grid = pick_grid(cdiv64(num_rows, rpw), bw);                    \
-    if (mean) {                                                               \
-      if (per_id_w)                                                           \
-        hipLaunchKernelGGL((csr_fwd_narrow<TILE, true, true>), dim3(grid),    \
-                           dim3(block), 0, stream, params, values, splits,    \
-                           per_id_w, out, num_rows, vocab, width);            \
-      else                                                                    \
-        hipLaunchKernelGGL((csr_fwd_narrow<TILE, true, false>), dim3(grid),   \
-                           dim3(block), 0, stream, params, values, splits,    \
-                           nullptr, out, num_rows, vocab, width);             \
-    } else {                                                                  \
-      if (per_id_w)                                                           \
-        hipLaunchKernelGGL((csr_fwd_narrow<TILE, false, true>), dim3(grid),   \
-                           dim3(block), 0, stream, params, values, splits,    \
-                           per_id_w, out, num_rows, vocab, width);            \
-      else                                                                    \
-        hipLaunchKernelGGL((csr_fwd_narrow<TILE, false, false>), dim3(grid),  \
-                           dim3(block), 0, stream, params, values, splits,    \
-                           nullptr, out, num_rows, vocab, width);             \
-    }                                                                         \
-  } while (0)
-
-#define DISPATCH_WIDE(VEC)                                                    \
-  do {                                                                        \
-    const int grid = pick_grid(num_rows, bw);                                 \
-    if (mean) {                                                               \
-      if (per_id_w)                                                           \
-        hipLaunchKernelGGL((csr_fwd_wide<VEC, true, true>), dim3(grid),       \
-                           dim3(block), 0, stream, params, values, splits,    \
-                           per_id_w, out, num_rows, vocab, width);            \
-      else                                                                    \
-        hipLaunchKernelGGL((csr_fwd_wide<VEC, true, false>), dim3(grid),      \
-                           dim3(block), 0, stream, params, values, splits,    \
-                           nullptr, out, num_rows, vocab, width);             \
-    } else {                                                                  \
-      if (per_id_w)                                                           \
-        hipLaunchKernelGGL((csr_fwd_wide<VEC, false, true>), dim3(grid),      \
-                           dim3(block), 0, stream, params, values, splits,    \
-                           per_id_w, out, num_rows, vocab, width);            \
-      else                                                                    \
-        hipLaunchKernelGGL((csr_fwd_wide<VEC, false, false>), dim3(grid),     \
-                           dim3(block), 0, stream, params, values, splits,    \
-                           nullptr, out, num_rows, vocab, width);             \
-    }                                                                         \
-  } while (0)
-
+                               float* out, int64_t num_rows, int64_t nnz,
+                               int64_t vocab, int width, bool mean,
+                               hipStream_t stream) {
+#define ARGS params, values, splits, per_id_w, out, num_rows, nnz, vocab, \
+             width, mean, stream
   if (width <= 64) {
     switch (next_pow2(width)) {
-      case 1: DISPATCH_NARROW(1); break;
-      case 2: DISPATCH_NARROW(2); break;
-      case 4: DISPATCH_NARROW(4); break;
-      case 8: DISPATCH_NARROW(8); break;
-      case 16: DISPATCH_NARROW(16); break;
-      case 32: DISPATCH_NARROW(32); break;
-      default: DISPATCH_NARROW(64); break;
+      case 1: launch_narrow<1>(ARGS); break;
+      case 2: launch_narrow<2>(ARGS); break;
+      case 4: launch_narrow<4>(ARGS); break;
+      case 8: launch_narrow<8>(ARGS); break;
+      case 16: launch_narrow<16>(ARGS); break;
+      case 32: launch_narrow<32>(ARGS); break;
+      default: launch_narrow<64>(ARGS); break;
     }
   } else if (width % 4 == 0) {
-    DISPATCH_WIDE(4);
+    launch_wide<4>(ARGS);
   } else if (width % 2 == 0) {
-    DISPATCH_WIDE(2);
+    launch_wide<2>(ARGS);
   } else {
-    DISPATCH_WIDE(1);
+    launch_wide<1>(ARGS);
   }
-#undef DISPATCH_NARROW
-#undef DISPATCH_WIDE
+#undef ARGS
 }
 
 // ---------------------------------------------------------------------------

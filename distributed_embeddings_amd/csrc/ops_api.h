@@ -9,8 +9,9 @@
 
 void launch_csr_lookup_forward(const float* params, const int64_t* values,
                                const int64_t* splits, const float* per_id_w,
-                               float* out, int64_t num_rows, int64_t vocab,
-                               int width, bool mean, hipStream_t stream);
+                               float* out, int64_t num_rows, int64_t nnz,
+                               int64_t vocab, int width, bool mean,
+                               hipStream_t stream);
 
 void launch_row_to_split(const int64_t* rows, int64_t nnz, int64_t num_rows,
                          int64_t* splits, hipStream_t stream);

@@ -91,17 +91,18 @@ class DLRM(nn.Module):
         interact_dim = num_feats * (num_feats - 1) // 2 + bottom_mlp_dims[-1]
         self.top_mlp = _mlp(top_mlp_dims, interact_dim, final_linear=True)
 
+        from ..parallel.strategy import TableConfig
         tables = [
-            Embedding(s, embedding_dim, initializer=scaled_uniform_init)
+            TableConfig(s, embedding_dim, None, initializer=scaled_uniform_init)
             for s in table_sizes
         ]
-        if self.distributed:
-            self.embeddings = DistributedEmbedding(
-                tables, strategy=strategy, dp_input=dp_input,
-                column_slice_threshold=column_slice_threshold,
-                data_parallel_threshold=data_parallel_threshold)
-        else:
-            self.embeddings = nn.ModuleList(tables)
+        # DistributedEmbedding at every world size: at world==1 it still fuses
+        # all same-width tables into one variable -> one lookup kernel + one
+        # backward pipeline per step.
+        self.embeddings = DistributedEmbedding(
+            tables, strategy=strategy, dp_input=dp_input,
+            column_slice_threshold=column_slice_threshold,
+            data_parallel_threshold=data_parallel_threshold)
 
     def local_cat_feature_ids(self) -> List[int]:
         if self.distributed and not self.dp_input:
@@ -111,10 +112,7 @@ class DLRM(nn.Module):
     def forward(self, numerical: torch.Tensor,
                 cat_features: Sequence[torch.Tensor]) -> torch.Tensor:
         bottom = self.bottom_mlp(numerical)
-        if self.distributed:
-            emb = self.embeddings(list(cat_features))
-        else:
-            emb = [e(x) for e, x in zip(self.embeddings, cat_features)]
+        emb = self.embeddings(list(cat_features))
         emb = [e.to(bottom.dtype) for e in emb]
         x = dot_interact(emb, bottom)
         return self.top_mlp(x)

@@ -42,18 +42,16 @@ class SyntheticModel(nn.Module):
                  data_parallel_threshold: Optional[int] = None):
         super().__init__()
         self.config = model_config
+        from ..parallel.strategy import TableConfig
         tables, input_table_map, self.hotness = expand_tables(model_config)
         self.input_table_map = input_table_map
-        layers = [Embedding(rows, width, combiner="sum") for rows, width in tables]
+        layers = [TableConfig(rows, width, "sum") for rows, width in tables]
         self.distributed = comm.world_size() > 1
-        if self.distributed:
-            self.embeddings = DistributedEmbedding(
-                layers, strategy=strategy, dp_input=dp_input,
-                input_table_map=input_table_map,
-                column_slice_threshold=column_slice_threshold,
-                data_parallel_threshold=data_parallel_threshold)
-        else:
-            self.embeddings = nn.ModuleList(layers)
+        self.embeddings = DistributedEmbedding(
+            layers, strategy=strategy, dp_input=dp_input,
+            input_table_map=input_table_map,
+            column_slice_threshold=column_slice_threshold,
+            data_parallel_threshold=data_parallel_threshold)
 
         self.interact_stride = model_config.interact_stride
         if self.interact_stride:
@@ -77,11 +75,7 @@ class SyntheticModel(nn.Module):
         self.mlp = nn.Sequential(*mods)
 
     def forward(self, numerical: torch.Tensor, cat_features) -> torch.Tensor:
-        if self.distributed:
-            embs = self.embeddings(list(cat_features))
-        else:
-            embs = [self.embeddings[t](x)
-                    for t, x in zip(self.input_table_map, cat_features)]
+        embs = self.embeddings(list(cat_features))
         x = torch.cat([e.to(numerical.dtype) for e in embs], dim=1)
         if self.interact is not None:
             x = self.interact(x.unsqueeze(1)).squeeze(1)

@@ -95,23 +95,6 @@ class DistributedEmbedding(nn.Module):
         self._built = False
         self._local_batch = None
 
-        if self.world_size == 1:
-            # Single process: keep (or build) plain local layers.
-            layers = []
-            for e, cfg in zip(embeddings, configs):
-                if isinstance(e, Embedding):
-                    layers.append(e)
-                elif isinstance(e, nn.Embedding):
-                    lyr = Embedding(cfg.input_dim, cfg.output_dim, cfg.combiner)
-                    with torch.no_grad():
-                        lyr.weight.copy_(e.weight)
-                    layers.append(lyr)
-                else:
-                    layers.append(Embedding(cfg.input_dim, cfg.output_dim, cfg.combiner,
-                                            initializer=cfg.initializer))
-            self.local_layers = nn.ModuleList(layers)
-            return
-
         plan = self.strategy
         # ---- data-parallel layers (replicated; grads allreduced) ----
         self.dp_layers = nn.ModuleList([
@@ -121,13 +104,38 @@ class DistributedEmbedding(nn.Module):
         ])
 
         # ---- column/table-parallel fused variables ----
+        # The fused group runs at EVERY world size (world==1 included): one
+        # lookup kernel + one backward pipeline per group per step, instead of
+        # one per table — the single biggest launch/sync saving for
+        # many-table models (MI355X departure from the reference's
+        # XLA-fusion hint, dist_model_parallel.py:650).
         col_layers = []
         for grp in plan.local_concat_groups(self.rank):
             lyr = Embedding(grp.input_dim, grp.output_dim, grp.combiner)
             for p in lyr.parameters():
-                p.de_local = True  # model-parallel: excluded from grad allreduce
+                if self.world_size > 1:
+                    p.de_local = True  # model-parallel: excluded from grad allreduce
             if grp.cpu_offload:
                 lyr._cpu_offload = True
+            # apply per-member initializers (or copy weights handed to us)
+            with torch.no_grad():
+                for m in grp.members:
+                    cfg = configs[m.table_id]
+                    dst = lyr.weight[m.concat_row_offset:
+                                     m.concat_row_offset + cfg.input_dim]
+                    src_layer = embeddings[m.table_id] if isinstance(
+                        embeddings[m.table_id], (Embedding, nn.Embedding)) else None
+                    if src_layer is not None:
+                        dst.copy_(src_layer.weight[:, m.col_offset:
+                                                   m.col_offset + m.width])
+                    elif cfg.initializer is not None:
+                        if m.width == cfg.output_dim:
+                            cfg.initializer(dst)
+                        else:
+                            full = torch.empty(cfg.input_dim, cfg.output_dim,
+                                               dtype=dst.dtype, device=dst.device)
+                            cfg.initializer(full)
+                            dst.copy_(full[:, m.col_offset:m.col_offset + m.width])
             col_layers.append(lyr)
         self.col_layers = nn.ModuleList(col_layers)
 
@@ -162,7 +170,7 @@ class DistributedEmbedding(nn.Module):
         return [col_inputs[i] for i in self.strategy.rank_input_ids[self.rank]]
 
     def _validate_batch(self, b: int):
-        if self._built:
+        if self._built or self.world_size == 1:
             return
         import torch.distributed as dist
         sizes = [None] * self.world_size
@@ -175,12 +183,6 @@ class DistributedEmbedding(nn.Module):
     # ---------------------------------------------------------------- forward
 
     def forward(self, inputs: Sequence[Union[torch.Tensor, Ragged]]) -> List[torch.Tensor]:
-        if self.world_size == 1:
-            if len(inputs) != len(self.strategy.input_table_map):
-                raise ValueError("wrong number of inputs")
-            return [self.local_layers[t](x)
-                    for x, t in zip(inputs, self.strategy.input_table_map)]
-
         plan = self.strategy
         dp_in, col_in, row_in = plan.input_groups
         if self.dp_input:
@@ -234,14 +236,6 @@ class DistributedEmbedding(nn.Module):
                     e = min(n, s + step)
                     dst[s:e].copy_(self._as_tensor(src[s:e]).to(dst.dtype))
 
-        if self.world_size == 1:
-            for lyr, w in zip(self.local_layers_per_table(), weights):
-                w = self._as_tensor(w)
-                if tuple(lyr.weight.shape) != tuple(w.shape):
-                    raise ValueError("weight shape mismatch")
-                copy_into(lyr.weight, w)
-            return
-
         plan = self.strategy
         for local_t, t in enumerate(plan.dp_table_ids):
             copy_into(self.dp_layers[local_t].weight, self._as_tensor(weights[t]))
@@ -265,10 +259,6 @@ class DistributedEmbedding(nn.Module):
         """
         import torch.distributed as dist
         cfgs = self.strategy.configs
-        if self.world_size == 1:
-            return [l.weight.detach().cpu().numpy().copy()
-                    for l in self.local_layers_per_table()]
-
         plan = self.strategy
         out: List[Optional[np.ndarray]] = [None] * len(cfgs)
         for local_t, t in enumerate(plan.dp_table_ids):
@@ -284,7 +274,8 @@ class DistributedEmbedding(nn.Module):
                 if s.rank == self.rank:
                     src = self.col_layers[s.concat_group].weight.detach()
                     buf.copy_(src[s.concat_row_offset:s.concat_row_offset + cfg.input_dim])
-                dist.broadcast(buf, src=s.rank)
+                if self.world_size > 1:
+                    dist.broadcast(buf, src=s.rank)
                 cols.append(buf.cpu())
             out[t] = torch.cat(cols, dim=1).numpy()
 
@@ -299,14 +290,12 @@ class DistributedEmbedding(nn.Module):
 
     def _comm_device(self):
         import torch.distributed as dist
-        if dist.get_backend() == "nccl":
+        if self.world_size > 1 and dist.get_backend() == "nccl":
             return torch.device("cuda", torch.cuda.current_device())
+        if self.world_size == 1:
+            return self.col_layers[0].weight.device if len(self.col_layers) else \
+                torch.device("cpu")
         return torch.device("cpu")
-
-    def local_layers_per_table(self):
-        """world_size==1 helper: local layer for each original table."""
-        # one layer per table (identity map through input_table_map not needed)
-        return list(self.local_layers)
 
     # ----------------------------------------------------- table parallel path
 
@@ -570,6 +559,9 @@ class DistributedEmbedding(nn.Module):
     def _exchange_pair_cols(self, my_pair_cols):
         """Share per-pair output column counts across ranks (cached)."""
         if getattr(self, "_all_pair_cols", None) is not None:
+            return self._all_pair_cols
+        if self.world_size == 1:
+            self._all_pair_cols = [list(my_pair_cols)]
             return self._all_pair_cols
         import torch.distributed as dist
         gathered = [None] * self.world_size
