@@ -51,3 +51,9 @@ w16 = torch.randn(1_000_000, 16, device="cuda")
 ids16 = torch.randint(0, 1_000_000, (b,), device="cuda")
 us = timeit(lambda: ext.csr_lookup_forward(w16, ids16, splits, False))
 print(f"csr_fwd w16 hot1    {us:8.1f} us")
+
+# long-segment (tiny vocab) backward — the split-kernel stress
+w3 = torch.randn(3, width, device="cuda")
+ids3 = torch.randint(0, 3, (8192,), device="cuda")
+us = timeit(lambda: ext.csr_lookup_backward(g, ids3, splits, 3, False))
+print(f"csr_bwd vocab3      {us:8.1f} us")
