@@ -96,8 +96,9 @@ def main():
         labels = torch.randint(0, 2, (b, 1), device=device).float()
         pool.append((num, cats, labels))
 
+    from distributed_embeddings_amd.parallel.optim import SparseEmbeddingOptimizer
     opt = de.DistributedOptimizer(
-        torch.optim.SGD(model.parameters(), lr=1e-3))
+        SparseEmbeddingOptimizer(model.parameters(), lr=1e-3, method="sgd"))
     de.broadcast_parameters(model)
     loss_fn = torch.nn.BCEWithLogitsLoss()
     use_bf16 = args.dtype == "bf16"

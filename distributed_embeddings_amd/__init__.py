@@ -14,6 +14,7 @@ from .layers.embedding import ConcatOneHotEmbedding, Embedding, scaled_uniform_i
 from .layers.integer_lookup import IntegerLookup
 from .parallel.strategy import DistEmbeddingStrategy, TableConfig
 from .parallel.dist_embedding import DistributedEmbedding
+from .parallel.optim import SparseEmbeddingOptimizer
 from .parallel.grad import (
     BroadcastParametersOnFirstStep,
     DistributedOptimizer,
@@ -37,6 +38,7 @@ __all__ = [
     "DistEmbeddingStrategy",
     "DistributedEmbedding",
     "DistributedOptimizer",
+    "SparseEmbeddingOptimizer",
     "BroadcastParametersOnFirstStep",
     "allreduce_gradients",
     "broadcast_parameters",

@@ -36,11 +36,15 @@ torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
   const int width = (int)params.size(1);
   auto out = torch::empty({num_rows, width}, params.options());
   if (num_rows > 0) {
+    auto long_rows = torch::empty({num_rows}, values.options());
+    auto long_count = torch::empty({1}, values.options().dtype(torch::kInt32));
     launch_csr_lookup_forward(params.data_ptr<float>(),
                               values.data_ptr<int64_t>(),
                               row_splits.data_ptr<int64_t>(), nullptr,
                               out.data_ptr<float>(), num_rows, values.numel(),
-                              vocab, width, mean, current_stream());
+                              vocab, width, mean,
+                              long_rows.data_ptr<int64_t>(),
+                              long_count.data_ptr<int32_t>(), current_stream());
   }
   return out;
 }
@@ -151,11 +155,15 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
   auto unique_grad = torch::empty({nu, width}, f32);
   if (nu > 0) {
     // 7. segmented sum == forward gather-reduce over grad_out rows.
+    auto long_rows = torch::empty({nu}, i64);
+    auto long_count = torch::empty({1}, i32);
     launch_csr_lookup_forward(grad_out.data_ptr<float>(),
                               srow.data_ptr<int64_t>(),
                               seg_tmp.data_ptr<int64_t>(), sw_ptr,
                               unique_grad.data_ptr<float>(), nu, nnz,
-                              grad_out.size(0), width, /*mean=*/false, stream);
+                              grad_out.size(0), width, /*mean=*/false,
+                              long_rows.data_ptr<int64_t>(),
+                              long_count.data_ptr<int32_t>(), stream);
   }
   return {unique_ids, unique_grad};
 }
@@ -193,6 +201,28 @@ torch::Tensor integer_lookup(torch::Tensor keys, torch::Tensor table_keys,
   return out;
 }
 
+void sparse_row_update(torch::Tensor weight, torch::Tensor state,
+                       torch::Tensor ids, torch::Tensor grad, double lr,
+                       double eps, bool adagrad) {
+  CHECK_CUDA(weight); CHECK_CUDA(ids); CHECK_CUDA(grad);
+  CHECK_CONTIG(weight); CHECK_CONTIG(ids); CHECK_CONTIG(grad);
+  TORCH_CHECK(weight.dtype() == torch::kFloat32);
+  TORCH_CHECK(grad.dtype() == torch::kFloat32);
+  TORCH_CHECK(ids.dtype() == torch::kInt64);
+  const int64_t n = ids.numel();
+  if (n == 0) return;
+  float* state_ptr = nullptr;
+  if (adagrad) {
+    CHECK_CUDA(state); CHECK_CONTIG(state);
+    TORCH_CHECK(state.sizes() == weight.sizes(), "adagrad state shape mismatch");
+    state_ptr = state.data_ptr<float>();
+  }
+  launch_sparse_row_update(weight.data_ptr<float>(), state_ptr,
+                           ids.data_ptr<int64_t>(), grad.data_ptr<float>(), n,
+                           (int)weight.size(1), (float)lr, (float)eps, adagrad,
+                           current_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -203,4 +233,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("row_to_split", &row_to_split, "COO rows -> CSR splits (gfx950)");
   m.def("integer_lookup", &integer_lookup,
         "open-addressing hash vocab build + lookup (gfx950)");
+  m.def("sparse_row_update", &sparse_row_update,
+        "fused sparse SGD/Adagrad row update (gfx950)");
 }

@@ -40,19 +40,29 @@ static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // Forward: CSR segmented gather-reduce.
 // ---------------------------------------------------------------------------
 
-// Narrow kernel: width <= 64.  TILE = pow2 >= width; 64/TILE rows per wave.
-// Lanes tl >= width within a tile idle on loads but stay converged.
-// SPLIT: gridDim.y waves cooperate on each row's reduction (strided over the
-// segment) and combine with atomicAdd — the load-balancing answer to
-// power-law mega-segments (reference analog: blockDim.y reduction splitting,
-// embedding_lookup_kernels.cu:195-226).
-template <int TILE, bool MEAN, bool HAS_W, bool SPLIT>
+// CSR segmented gather-reduce, two-kernel adaptive design:
+//   Kernel A (csr_fwd_*) owns one row per wave (or sub-wave tile).  Rows with
+//   segments <= LONG_T reduce in registers and store directly (no atomics).
+//   Longer rows (power-law mega-segments, tiny-vocab backward) are zero-
+//   filled and pushed to a device-side long-row list.
+//   Kernel B (csr_fwd_long_*) grid-strides waves over (long row, chunk)
+//   pairs, each wave reducing one LONG_T-sized chunk and combining with
+//   atomicAdd.  No host sync, no full-output memset.
+// This is the wave64 answer to the reference's blockDim.y reduction
+// splitting + round-robin step counter (embedding_lookup_kernels.cu:195-226).
+#define LONG_T 128
+#define MAX_CHUNKS 1024
+
+// Narrow kernel A: width <= 64.  TILE = pow2 >= width; 64/TILE rows per wave.
+template <int TILE, bool MEAN, bool HAS_W>
 __global__ void csr_fwd_narrow(const float* __restrict__ params,
                                const int64_t* __restrict__ values,
                                const int64_t* __restrict__ splits,
                                const float* __restrict__ per_id_w,
                                float* __restrict__ out, int64_t num_rows,
-                               int64_t vocab, int width) {
+                               int64_t vocab, int width,
+                               int64_t* __restrict__ long_rows,
+                               int32_t* __restrict__ long_count) {
   constexpr int RPW = WAVE / TILE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int sub = lane / TILE;
@@ -60,53 +70,56 @@ __global__ void csr_fwd_narrow(const float* __restrict__ params,
   const int64_t wave_id =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
-  const int ystep = SPLIT ? gridDim.y : 1;
-  const int y = SPLIT ? blockIdx.y : 0;
   for (int64_t base = wave_id * RPW; base < num_rows; base += n_waves * RPW) {
     const int64_t row = base + sub;
     if (row >= num_rows || tl >= width) continue;
     const int64_t s = splits[row], e = splits[row + 1];
+    if (e - s > LONG_T) {
+      out[row * width + tl] = 0.f;
+      if (tl == 0) long_rows[atomicAdd(long_count, 1)] = row;
+      continue;
+    }
     float acc = 0.f;
-    for (int64_t k = s + y; k < e; k += ystep) {
+    for (int64_t k = s; k < e; ++k) {
       const int64_t idx = values[k];
       if (idx < 0 || idx >= vocab) continue;
       const float w = HAS_W ? per_id_w[k] : 1.f;
       acc += w * params[idx * width + tl];
     }
     if (MEAN && e > s) acc /= (float)(e - s);
-    if (SPLIT) {
-      atomicAdd(&out[row * width + tl], acc);
-    } else {
-      out[row * width + tl] = acc;
-    }
+    out[row * width + tl] = acc;
   }
 }
 
-// Wide kernel: width > 64.  One wave per row; VEC elements per lane
-// (vectorized loads when alignment permits); outer loop over 64*VEC chunks.
-template <int VEC, bool MEAN, bool HAS_W, bool SPLIT>
+// Wide kernel A: width > 64.  One wave per row; VEC elements per lane.
+template <int VEC, bool MEAN, bool HAS_W>
 __global__ void csr_fwd_wide(const float* __restrict__ params,
                              const int64_t* __restrict__ values,
                              const int64_t* __restrict__ splits,
                              const float* __restrict__ per_id_w,
                              float* __restrict__ out, int64_t num_rows,
-                             int64_t vocab, int width) {
+                             int64_t vocab, int width,
+                             int64_t* __restrict__ long_rows,
+                             int32_t* __restrict__ long_count) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
   constexpr int CHUNK = WAVE * VEC;
-  const int ystep = SPLIT ? gridDim.y : 1;
-  const int y = SPLIT ? blockIdx.y : 0;
   for (int64_t row = wave_id; row < num_rows; row += n_waves) {
     const int64_t s = splits[row], e = splits[row + 1];
+    if (e - s > LONG_T) {
+      for (int c = lane; c < width; c += WAVE) out[row * width + c] = 0.f;
+      if (lane == 0) long_rows[atomicAdd(long_count, 1)] = row;
+      continue;
+    }
     const float inv = (MEAN && e > s) ? 1.f / (float)(e - s) : 1.f;
     for (int cbase = 0; cbase < width; cbase += CHUNK) {
       float acc[VEC];
 #pragma unroll
       for (int v = 0; v < VEC; ++v) acc[v] = 0.f;
       const int col0 = cbase + lane * VEC;
-      for (int64_t k = s + y; k < e; k += ystep) {
+      for (int64_t k = s; k < e; ++k) {
         const int64_t idx = values[k];
         if (idx < 0 || idx >= vocab) continue;
         const float w = HAS_W ? per_id_w[k] : 1.f;
@@ -136,20 +149,79 @@ __global__ void csr_fwd_wide(const float* __restrict__ params,
         }
       }
       float* outp = out + row * (int64_t)width + col0;
-      if (SPLIT) {
-#pragma unroll
-        for (int v = 0; v < VEC; ++v)
-          if (col0 + v < width) atomicAdd(&outp[v], acc[v] * inv);
-      } else if (VEC == 4 && col0 + 4 <= width) {
-        float4 o4 = make_float4(acc[0] * inv, acc[1] * inv, acc[2] * inv,
-                                acc[3] * inv);
-        *reinterpret_cast<float4*>(outp) = o4;
+      if (VEC == 4 && col0 + 4 <= width) {
+        *reinterpret_cast<float4*>(outp) = make_float4(
+            acc[0] * inv, acc[1] * inv, acc[2] * inv, acc[3] * inv);
       } else if (VEC == 2 && col0 + 2 <= width) {
         *reinterpret_cast<float2*>(outp) = make_float2(acc[0] * inv, acc[1] * inv);
       } else {
 #pragma unroll
         for (int v = 0; v < VEC; ++v)
           if (col0 + v < width) outp[v] = acc[v] * inv;
+      }
+    }
+  }
+}
+
+// Kernel B: long rows.  Work item = (long row, LONG_T-chunk); one wave each.
+// NW lanes-per-row tiling matches kernel A (TILE for narrow, full wave for
+// wide).  Partials combine with global atomicAdd (out pre-zeroed by A).
+template <int TILE, int VEC, bool MEAN, bool HAS_W>
+__global__ void csr_fwd_long(const float* __restrict__ params,
+                             const int64_t* __restrict__ values,
+                             const int64_t* __restrict__ splits,
+                             const float* __restrict__ per_id_w,
+                             float* __restrict__ out, int64_t vocab, int width,
+                             const int64_t* __restrict__ long_rows,
+                             const int32_t* __restrict__ long_count) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id =
+      ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const int64_t n_long = *long_count;
+  const int64_t n_items = n_long * MAX_CHUNKS;
+  for (int64_t item = wave_id; item < n_items; item += n_waves) {
+    const int64_t li = item / MAX_CHUNKS;
+    const int64_t chunk = item % MAX_CHUNKS;
+    const int64_t row = long_rows[li];
+    const int64_t s = splits[row], e = splits[row + 1];
+    const int64_t ks = s + chunk * LONG_T;
+    if (ks >= e) continue;  // past this row's last chunk
+    const int64_t ke = min(ks + (int64_t)LONG_T, e);
+    const float inv = (MEAN && e > s) ? 1.f / (float)(e - s) : 1.f;
+    if constexpr (TILE > 0) {
+      // narrow tiling: lane tl covers column tl; only lanes < width active
+      const int tl = lane % (TILE > 0 ? TILE : 1);
+      const int sub = lane / (TILE > 0 ? TILE : 1);
+      if (sub != 0 || tl >= width) continue;
+      float acc = 0.f;
+      for (int64_t k = ks; k < ke; ++k) {
+        const int64_t idx = values[k];
+        if (idx < 0 || idx >= vocab) continue;
+        const float w = HAS_W ? per_id_w[k] : 1.f;
+        acc += w * params[idx * width + tl];
+      }
+      atomicAdd(&out[row * width + tl], acc * inv);
+    } else {
+      constexpr int CH = WAVE * (VEC > 0 ? VEC : 1);
+      for (int cbase = 0; cbase < width; cbase += CH) {
+        float acc[VEC > 0 ? VEC : 1];
+#pragma unroll
+        for (int v = 0; v < (VEC > 0 ? VEC : 1); ++v) acc[v] = 0.f;
+        const int col0 = cbase + lane * (VEC > 0 ? VEC : 1);
+        for (int64_t k = ks; k < ke; ++k) {
+          const int64_t idx = values[k];
+          if (idx < 0 || idx >= vocab) continue;
+          const float w = HAS_W ? per_id_w[k] : 1.f;
+          const float* rowp = params + idx * (int64_t)width + col0;
+#pragma unroll
+          for (int v = 0; v < (VEC > 0 ? VEC : 1); ++v)
+            if (col0 + v < width) acc[v] += w * rowp[v];
+        }
+        float* outp = out + row * (int64_t)width + col0;
+#pragma unroll
+        for (int v = 0; v < (VEC > 0 ? VEC : 1); ++v)
+          if (col0 + v < width) atomicAdd(&outp[v], acc[v] * inv);
       }
     }
   }
@@ -163,94 +235,69 @@ static int pick_grid(int64_t work_items, int block_waves) {
   return (int)blocks;
 }
 
-// Segment-split factor: raise parallelism when rows are few and segments
-// long (tiny-vocab backward, power-law skew).  Result 1 => no atomics.
-static int pick_split(int64_t num_rows, int64_t nnz, int64_t row_waves) {
-  if (nnz <= 0 || num_rows <= 0) return 1;
-  const int64_t ave = nnz / num_rows;
-  int split = 1;
-  // target: enough waves to fill the chip (256 CUs x ~8 waves), without
-  // splitting segments below ~64 ids per wave.
-  while (split < 64 && ave / split > 64 && row_waves * split < 16384) {
-    split *= 2;
-  }
-  return split;
-}
-
-template <int TILE>
-static void launch_narrow(const float* params, const int64_t* values,
-                          const int64_t* splits, const float* per_id_w,
-                          float* out, int64_t num_rows, int64_t nnz,
-                          int64_t vocab, int width, bool mean,
-                          hipStream_t stream) {
+template <int TILE, int VEC>
+static void launch_csr_pair(const float* params, const int64_t* values,
+                            const int64_t* splits, const float* per_id_w,
+                            float* out, int64_t num_rows, int64_t nnz,
+                            int64_t vocab, int width, bool mean,
+                            int64_t* long_rows, int32_t* long_count,
+                            hipStream_t stream) {
   const int block = 256, bw = block / WAVE;
-  const int64_t row_waves = cdiv64(num_rows, WAVE / TILE);
-  const int split = pick_split(num_rows, nnz, row_waves);
-  const dim3 grid(pick_grid(row_waves, bw), split);
-  if (split > 1)
-    hipMemsetAsync(out, 0, sizeof(float) * num_rows * width, stream);
-#define L(MEAN, HASW, SPLIT)                                                   \
-  hipLaunchKernelGGL((csr_fwd_narrow<TILE, MEAN, HASW, SPLIT>), grid,          \
-                     dim3(block), 0, stream, params, values, splits,           \
-                     per_id_w, out, num_rows, vocab, width)
-  if (split > 1) {
-    if (mean) { if (per_id_w) L(true, true, true); else L(true, false, true); }
-    else      { if (per_id_w) L(false, true, true); else L(false, false, true); }
+  const int64_t row_waves = TILE > 0 ? cdiv64(num_rows, WAVE / (TILE > 0 ? TILE : 1)) : num_rows;
+  const dim3 grid(pick_grid(row_waves, bw));
+  hipMemsetAsync(long_count, 0, sizeof(int32_t), stream);
+#define LA(MEAN, HASW)                                                         \
+  do {                                                                         \
+    if constexpr (TILE > 0)                                                    \
+      hipLaunchKernelGGL((csr_fwd_narrow<(TILE > 0 ? TILE : 1), MEAN, HASW>),  \
+                         grid, dim3(block), 0, stream, params, values, splits, \
+                         per_id_w, out, num_rows, vocab, width, long_rows,     \
+                         long_count);                                          \
+    else                                                                       \
+      hipLaunchKernelGGL((csr_fwd_wide<(VEC > 0 ? VEC : 1), MEAN, HASW>),      \
+                         grid, dim3(block), 0, stream, params, values, splits, \
+                         per_id_w, out, num_rows, vocab, width, long_rows,     \
+                         long_count);                                          \
+  } while (0)
+#define LB(MEAN, HASW)                                                         \
+  hipLaunchKernelGGL((csr_fwd_long<TILE, VEC, MEAN, HASW>), dim3(2048),        \
+                     dim3(block), 0, stream, params, values, splits, per_id_w, \
+                     out, vocab, width, long_rows, long_count)
+  if (mean) {
+    if (per_id_w) { LA(true, true); LB(true, true); }
+    else          { LA(true, false); LB(true, false); }
   } else {
-    if (mean) { if (per_id_w) L(true, true, false); else L(true, false, false); }
-    else      { if (per_id_w) L(false, true, false); else L(false, false, false); }
+    if (per_id_w) { LA(false, true); LB(false, true); }
+    else          { LA(false, false); LB(false, false); }
   }
-#undef L
-}
-
-template <int VEC>
-static void launch_wide(const float* params, const int64_t* values,
-                        const int64_t* splits, const float* per_id_w,
-                        float* out, int64_t num_rows, int64_t nnz,
-                        int64_t vocab, int width, bool mean,
-                        hipStream_t stream) {
-  const int block = 256, bw = block / WAVE;
-  const int split = pick_split(num_rows, nnz, num_rows);
-  const dim3 grid(pick_grid(num_rows, bw), split);
-  if (split > 1)
-    hipMemsetAsync(out, 0, sizeof(float) * num_rows * width, stream);
-#define L(MEAN, HASW, SPLIT)                                                   \
-  hipLaunchKernelGGL((csr_fwd_wide<VEC, MEAN, HASW, SPLIT>), grid,             \
-                     dim3(block), 0, stream, params, values, splits,           \
-                     per_id_w, out, num_rows, vocab, width)
-  if (split > 1) {
-    if (mean) { if (per_id_w) L(true, true, true); else L(true, false, true); }
-    else      { if (per_id_w) L(false, true, true); else L(false, false, true); }
-  } else {
-    if (mean) { if (per_id_w) L(true, true, false); else L(true, false, false); }
-    else      { if (per_id_w) L(false, true, false); else L(false, false, false); }
-  }
-#undef L
+#undef LA
+#undef LB
 }
 
 void launch_csr_lookup_forward(const float* params, const int64_t* values,
                                const int64_t* splits, const float* per_id_w,
                                float* out, int64_t num_rows, int64_t nnz,
                                int64_t vocab, int width, bool mean,
+                               int64_t* long_rows, int32_t* long_count,
                                hipStream_t stream) {
 #define ARGS params, values, splits, per_id_w, out, num_rows, nnz, vocab, \
-             width, mean, stream
+             width, mean, long_rows, long_count, stream
   if (width <= 64) {
     switch (next_pow2(width)) {
-      case 1: launch_narrow<1>(ARGS); break;
-      case 2: launch_narrow<2>(ARGS); break;
-      case 4: launch_narrow<4>(ARGS); break;
-      case 8: launch_narrow<8>(ARGS); break;
-      case 16: launch_narrow<16>(ARGS); break;
-      case 32: launch_narrow<32>(ARGS); break;
-      default: launch_narrow<64>(ARGS); break;
+      case 1: launch_csr_pair<1, 0>(ARGS); break;
+      case 2: launch_csr_pair<2, 0>(ARGS); break;
+      case 4: launch_csr_pair<4, 0>(ARGS); break;
+      case 8: launch_csr_pair<8, 0>(ARGS); break;
+      case 16: launch_csr_pair<16, 0>(ARGS); break;
+      case 32: launch_csr_pair<32, 0>(ARGS); break;
+      default: launch_csr_pair<64, 0>(ARGS); break;
     }
   } else if (width % 4 == 0) {
-    launch_wide<4>(ARGS);
+    launch_csr_pair<0, 4>(ARGS);
   } else if (width % 2 == 0) {
-    launch_wide<2>(ARGS);
+    launch_csr_pair<0, 2>(ARGS);
   } else {
-    launch_wide<1>(ARGS);
+    launch_csr_pair<0, 1>(ARGS);
   }
 #undef ARGS
 }
@@ -479,6 +526,78 @@ void launch_find_valid_bounds(const int64_t* sorted_ids, int64_t n,
                               hipStream_t stream) {
   hipLaunchKernelGGL(find_valid_bounds, dim3(1), dim3(2), 0, stream, sorted_ids,
                      n, vocab, bounds);
+}
+
+// ---------------------------------------------------------------------------
+// Fused sparse optimizer steps: apply (unique_ids, unique_grad) rows directly
+// to the table — no torch sparse re-coalesce, no dense grad materialization.
+// One wave per row (width>64) or sub-wave tiles (narrow), same tiling as the
+// lookup kernels.
+// ---------------------------------------------------------------------------
+
+// SGD: w[id] -= lr * g.   Adagrad: s[id] += g^2; w[id] -= lr*g/(sqrt(s)+eps).
+template <bool ADAGRAD>
+__global__ void sparse_row_update(float* __restrict__ weight,
+                                  float* __restrict__ state,
+                                  const int64_t* __restrict__ ids,
+                                  const float* __restrict__ grad,
+                                  int64_t num_rows, int width, float lr,
+                                  float eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  for (int64_t r = wave_id; r < num_rows; r += n_waves) {
+    const int64_t row = ids[r];
+    float* wp = weight + row * (int64_t)width;
+    float* sp = ADAGRAD ? state + row * (int64_t)width : nullptr;
+    const float* gp = grad + r * (int64_t)width;
+    for (int c = lane * 4; c < width; c += WAVE * 4) {
+      if (c + 4 <= width) {
+        const float4 g4 = *reinterpret_cast<const float4*>(gp + c);
+        float4 w4 = *reinterpret_cast<float4*>(wp + c);
+        if (ADAGRAD) {
+          float4 s4 = *reinterpret_cast<float4*>(sp + c);
+          s4.x += g4.x * g4.x; s4.y += g4.y * g4.y;
+          s4.z += g4.z * g4.z; s4.w += g4.w * g4.w;
+          *reinterpret_cast<float4*>(sp + c) = s4;
+          w4.x -= lr * g4.x / (sqrtf(s4.x) + eps);
+          w4.y -= lr * g4.y / (sqrtf(s4.y) + eps);
+          w4.z -= lr * g4.z / (sqrtf(s4.z) + eps);
+          w4.w -= lr * g4.w / (sqrtf(s4.w) + eps);
+        } else {
+          w4.x -= lr * g4.x; w4.y -= lr * g4.y;
+          w4.z -= lr * g4.z; w4.w -= lr * g4.w;
+        }
+        *reinterpret_cast<float4*>(wp + c) = w4;
+      } else {
+        for (int cc = c; cc < width; ++cc) {
+          const float g = gp[cc];
+          if (ADAGRAD) {
+            sp[cc] += g * g;
+            wp[cc] -= lr * g / (sqrtf(sp[cc]) + eps);
+          } else {
+            wp[cc] -= lr * g;
+          }
+        }
+      }
+    }
+  }
+}
+
+void launch_sparse_row_update(float* weight, float* state, const int64_t* ids,
+                              const float* grad, int64_t num_rows, int width,
+                              float lr, float eps, bool adagrad,
+                              hipStream_t stream) {
+  const int block = 256;
+  const int grid = pick_grid(num_rows, block / WAVE);
+  if (adagrad)
+    hipLaunchKernelGGL(sparse_row_update<true>, dim3(grid), dim3(block), 0,
+                       stream, weight, state, ids, grad, num_rows, width, lr,
+                       eps);
+  else
+    hipLaunchKernelGGL(sparse_row_update<false>, dim3(grid), dim3(block), 0,
+                       stream, weight, state, ids, grad, num_rows, width, lr,
+                       eps);
 }
 
 // ---------------------------------------------------------------------------

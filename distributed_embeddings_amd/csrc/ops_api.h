@@ -11,6 +11,7 @@ void launch_csr_lookup_forward(const float* params, const int64_t* values,
                                const int64_t* splits, const float* per_id_w,
                                float* out, int64_t num_rows, int64_t nnz,
                                int64_t vocab, int width, bool mean,
+                               int64_t* long_rows, int32_t* long_count,
                                hipStream_t stream);
 
 void launch_row_to_split(const int64_t* rows, int64_t nnz, int64_t num_rows,
@@ -62,3 +63,8 @@ void launch_integer_lookup(const int64_t* keys, int64_t n, int64_t* tkeys,
                            int64_t* out, hipStream_t stream);
 
 size_t integer_lookup_temp_bytes(int64_t max_tokens);
+
+void launch_sparse_row_update(float* weight, float* state, const int64_t* ids,
+                              const float* grad, int64_t num_rows, int width,
+                              float lr, float eps, bool adagrad,
+                              hipStream_t stream);

@@ -101,3 +101,25 @@ def test_concat_onehot(seed):
     assert out.shape == (2, 3, 4)
     assert torch.equal(out[0, 1], layer.weight[5 + 2])
     assert torch.equal(out[1, 2], layer.weight[5 + 7 + 2])
+
+
+def test_sparse_embedding_optimizer_cpu_matches_torch(seed):
+    """SparseEmbeddingOptimizer (cpu path) == torch.optim on same grads."""
+    from distributed_embeddings_amd import Embedding, SparseEmbeddingOptimizer
+    for method, torch_opt in [("sgd", lambda p: torch.optim.SGD(p, lr=0.1)),
+                              ("adagrad", lambda p: torch.optim.Adagrad(p, lr=0.1, eps=1e-10))]:
+        ids = torch.randint(0, 50, (16, 4))
+        w0 = torch.randn(50, 8)
+        e1 = Embedding(50, 8, combiner="sum")
+        e2 = Embedding(50, 8, combiner="sum")
+        with torch.no_grad():
+            e1.weight.copy_(w0)
+            e2.weight.copy_(w0)
+        o1 = SparseEmbeddingOptimizer(e1.parameters(), lr=0.1, method=method)
+        o2 = torch_opt(list(e2.parameters()))
+        for _ in range(3):
+            o1.zero_grad(); o2.zero_grad()
+            e1(ids).square().sum().backward()
+            e2(ids).square().sum().backward()
+            o1.step(); o2.step()
+        assert torch.allclose(e1.weight, e2.weight, atol=1e-5), method
