@@ -234,8 +234,10 @@ def test_long_segment_forward_vs_cpu():
     r = Ragged.from_row_lengths(vals.cuda(), torch.tensor(lens).cuda())
     for combiner in ("sum", "mean"):
         out = embedding_lookup(w.cuda(), r, combiner)
-        ref = _csr_lookup_ref(w, vals, Ragged.from_row_lengths(vals, torch.tensor(lens)).row_splits, combiner)
-        assert torch.allclose(out.cpu(), ref, atol=1e-3), \
+        ref = _csr_lookup_ref(w.double(), vals, Ragged.from_row_lengths(
+            vals, torch.tensor(lens)).row_splits, combiner).float()
+        # fp32 accumulation-order roundoff bound ~ len * eps * |max|
+        assert torch.allclose(out.cpu(), ref, atol=5e-2), \
             f"{combiner}: {(out.cpu()-ref).abs().max()}"
     # narrow width too
     w16 = torch.randn(50, 16)
