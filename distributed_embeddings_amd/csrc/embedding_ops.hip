@@ -51,7 +51,7 @@ static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // This is the wave64 answer to the reference's blockDim.y reduction
 // splitting + round-robin step counter (embedding_lookup_kernels.cu:195-226).
 #define LONG_T 128
-#define MAX_CHUNKS 1024
+#define MAX_CHUNKS 32
 
 // Narrow kernel A: width <= 64.  TILE = pow2 >= width; 64/TILE rows per wave.
 template <int TILE, bool MEAN, bool HAS_W>
@@ -185,9 +185,10 @@ __global__ void csr_fwd_long(const float* __restrict__ params,
     const int64_t chunk = item % MAX_CHUNKS;
     const int64_t row = long_rows[li];
     const int64_t s = splits[row], e = splits[row + 1];
-    const int64_t ks = s + chunk * LONG_T;
-    if (ks >= e) continue;  // past this row's last chunk
-    const int64_t ke = min(ks + (int64_t)LONG_T, e);
+    // chunk-wave `chunk` strides the whole segment in MAX_CHUNKS*LONG_T hops,
+    // so arbitrarily long segments still use exactly MAX_CHUNKS waves.
+    const int64_t k0 = s + chunk * LONG_T;
+    if (k0 >= e) continue;
     const float inv = (MEAN && e > s) ? 1.f / (float)(e - s) : 1.f;
     if constexpr (TILE > 0) {
       // narrow tiling: lane tl covers column tl; only lanes < width active
@@ -195,32 +196,45 @@ __global__ void csr_fwd_long(const float* __restrict__ params,
       const int sub = lane / (TILE > 0 ? TILE : 1);
       if (sub != 0 || tl >= width) continue;
       float acc = 0.f;
-      for (int64_t k = ks; k < ke; ++k) {
-        const int64_t idx = values[k];
-        if (idx < 0 || idx >= vocab) continue;
-        const float w = HAS_W ? per_id_w[k] : 1.f;
-        acc += w * params[idx * width + tl];
-      }
-      atomicAdd(&out[row * width + tl], acc * inv);
-    } else {
-      constexpr int CH = WAVE * (VEC > 0 ? VEC : 1);
-      for (int cbase = 0; cbase < width; cbase += CH) {
-        float acc[VEC > 0 ? VEC : 1];
-#pragma unroll
-        for (int v = 0; v < (VEC > 0 ? VEC : 1); ++v) acc[v] = 0.f;
-        const int col0 = cbase + lane * (VEC > 0 ? VEC : 1);
+      for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
+        const int64_t ke = min(ks + (int64_t)LONG_T, e);
         for (int64_t k = ks; k < ke; ++k) {
           const int64_t idx = values[k];
           if (idx < 0 || idx >= vocab) continue;
           const float w = HAS_W ? per_id_w[k] : 1.f;
-          const float* rowp = params + idx * (int64_t)width + col0;
+          acc += w * params[idx * width + tl];
+        }
+      }
+      atomicAdd(&out[row * width + tl], acc * inv);
+    } else {
+      constexpr int V = VEC > 0 ? VEC : 1;
+      constexpr int CH = WAVE * V;
+      for (int cbase = 0; cbase < width; cbase += CH) {
+        float acc[V];
 #pragma unroll
-          for (int v = 0; v < (VEC > 0 ? VEC : 1); ++v)
-            if (col0 + v < width) acc[v] += w * rowp[v];
+        for (int v = 0; v < V; ++v) acc[v] = 0.f;
+        const int col0 = cbase + lane * V;
+        for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
+          const int64_t ke = min(ks + (int64_t)LONG_T, e);
+          for (int64_t k = ks; k < ke; ++k) {
+            const int64_t idx = values[k];
+            if (idx < 0 || idx >= vocab) continue;
+            const float w = HAS_W ? per_id_w[k] : 1.f;
+            const float* rowp = params + idx * (int64_t)width + col0;
+            if (V == 4 && col0 + 4 <= width) {
+              const float4 r4 = *reinterpret_cast<const float4*>(rowp);
+              acc[0] += w * r4.x; acc[1] += w * r4.y;
+              acc[2] += w * r4.z; acc[3] += w * r4.w;
+            } else {
+#pragma unroll
+              for (int v = 0; v < V; ++v)
+                if (col0 + v < width) acc[v] += w * rowp[v];
+            }
+          }
         }
         float* outp = out + row * (int64_t)width + col0;
 #pragma unroll
-        for (int v = 0; v < (VEC > 0 ? VEC : 1); ++v)
+        for (int v = 0; v < V; ++v)
           if (col0 + v < width) atomicAdd(&outp[v], acc[v] * inv);
       }
     }

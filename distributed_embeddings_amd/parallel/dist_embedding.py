@@ -441,12 +441,11 @@ class DistributedEmbedding(nn.Module):
                     splits = torch.arange(allids.numel() + 1, device=allids.device,
                                           dtype=torch.long)
                     emb = embedding_lookup(layer.weight, Ragged(allids, splits), "sum")
-                pos = 0
-                for j, shape, n in metas:
-                    rows = shape[0]
-                    out = emb[pos:pos + n].reshape(rows, -1)
-                    outs[j] = out
-                    pos += n
+                # torch.split (not manual narrow): its backward is ONE cat
+                # instead of per-slice zero-fill + accumulate.
+                parts = torch.split(emb, [n for _, _, n in metas])
+                for (j, shape, n), part in zip(metas, parts):
+                    outs[j] = part.reshape(shape[0], -1)
             else:
                 # One CSR batch over all pairs of this group.
                 val_parts, split_parts, metas = [], [], []
@@ -482,10 +481,9 @@ class DistributedEmbedding(nn.Module):
                 else:
                     out = embedding_lookup(layer.weight, Ragged(allvals, allsplits),
                                            grp.combiner)
-                pos = 0
-                for j, nrows in metas:
-                    outs[j] = out[pos:pos + nrows]
-                    pos += nrows
+                parts = torch.split(out, [nrows for _, nrows in metas])
+                for (j, nrows), part in zip(metas, parts):
+                    outs[j] = part
         return outs
 
     def _call_table_parallel(self, col_inputs):
