@@ -24,18 +24,25 @@ from ..parallel import comm
 from ..parallel.dist_embedding import DistributedEmbedding
 
 
-def dot_interact(emb_outs: List[torch.Tensor], bottom_mlp_out: torch.Tensor) -> torch.Tensor:
+def dot_interact(emb_outs: List[torch.Tensor], bottom_mlp_out: torch.Tensor,
+                 pad_to: int = 0) -> torch.Tensor:
     """Pairwise-dot interaction over [bottom_out] + embeddings.
 
     Parity: reference ``utils.py:92-113`` — lower-triangular portion of the
     [F+1, F+1] Gram matrix, then re-concat the bottom MLP output.
+    ``pad_to``: zero-pad the output width to this size so the following GEMM
+    gets an MFMA-friendly K (479 -> 512 for Criteo).
     """
     feats = torch.stack([bottom_mlp_out] + emb_outs, dim=1)  # [B, F, D]
     gram = torch.bmm(feats, feats.transpose(1, 2))           # [B, F, F]
     f = gram.shape[1]
     ii, jj = torch.tril_indices(f, f, offset=-1, device=gram.device)
     interactions = gram[:, ii, jj]                           # [B, F(F-1)/2]
-    return torch.cat([interactions, bottom_mlp_out], dim=1)
+    parts = [interactions, bottom_mlp_out]
+    width = interactions.shape[1] + bottom_mlp_out.shape[1]
+    if pad_to > width:
+        parts.append(bottom_mlp_out.new_zeros(bottom_mlp_out.shape[0], pad_to - width))
+    return torch.cat(parts, dim=1)
 
 
 def _mlp(sizes: Sequence[int], in_dim: int, final_linear: bool) -> nn.Sequential:
@@ -89,7 +96,9 @@ class DLRM(nn.Module):
         self.bottom_mlp = _mlp(bottom_mlp_dims, num_numerical, final_linear=False)
         num_feats = len(table_sizes) + 1
         interact_dim = num_feats * (num_feats - 1) // 2 + bottom_mlp_dims[-1]
-        self.top_mlp = _mlp(top_mlp_dims, interact_dim, final_linear=True)
+        # zero-pad to a multiple of 64 for MFMA-friendly GEMM K (479 -> 512)
+        self.interact_pad = ((interact_dim + 63) // 64) * 64
+        self.top_mlp = _mlp(top_mlp_dims, self.interact_pad, final_linear=True)
 
         from ..parallel.strategy import TableConfig
         tables = [
@@ -114,5 +123,5 @@ class DLRM(nn.Module):
         bottom = self.bottom_mlp(numerical)
         emb = self.embeddings(list(cat_features))
         emb = [e.to(bottom.dtype) for e in emb]
-        x = dot_interact(emb, bottom)
+        x = dot_interact(emb, bottom, pad_to=self.interact_pad)
         return self.top_mlp(x)
