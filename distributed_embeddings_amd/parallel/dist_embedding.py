@@ -423,21 +423,21 @@ class DistributedEmbedding(nn.Module):
             grp = groups[gi]
             offload = getattr(layer, "_cpu_offload", False)
             if grp.combiner is None:
-                flat_parts, metas = [], []
-                for j in pair_js:
-                    ids = pair_ids[j]
-                    off = self._pair_row_offset[j]
-                    flat = ids.reshape(-1) + off
-                    flat_parts.append(flat)
-                    metas.append((j, ids.shape, flat.numel()))
-                allids = torch.cat(flat_parts)
+                # hotness-1 CSR gather: one cat + one cached-offset add for the
+                # whole group (instead of one add per pair per step).
+                metas = [(j, pair_ids[j].shape, pair_ids[j].numel()) for j in pair_js]
+                allids = torch.cat([pair_ids[j].reshape(-1) for j in pair_js])
+                off_vec = self._offset_vector(gi, [(self._pair_row_offset[j], n)
+                                                   for j, _, n in metas],
+                                              allids.device)
+                if off_vec is not None:
+                    allids = allids + off_vec
                 if offload:
                     allids_cpu = allids.cpu()
                     splits = torch.arange(allids_cpu.numel() + 1, dtype=torch.long)
                     emb = embedding_lookup(layer.weight, Ragged(allids_cpu, splits),
                                            "sum").to(allids.device)
                 else:
-                    # hotness-1 CSR: gather with sparse (IndexedSlices) grad.
                     splits = torch.arange(allids.numel() + 1, device=allids.device,
                                           dtype=torch.long)
                     emb = embedding_lookup(layer.weight, Ragged(allids, splits), "sum")
@@ -447,33 +447,36 @@ class DistributedEmbedding(nn.Module):
                 for (j, shape, n), part in zip(metas, parts):
                     outs[j] = part.reshape(shape[0], -1)
             else:
-                # One CSR batch over all pairs of this group.
-                val_parts, split_parts, metas = [], [], []
-                row_base = 0
+                # One CSR batch over all pairs of this group; splits merged on
+                # device (no host syncs even for ragged inputs).
+                val_parts, len_parts, metas = [], [], []
+                row_offs = []
+                device = None
                 for j in pair_js:
                     ids = pair_ids[j]
                     off = self._pair_row_offset[j]
                     if isinstance(ids, Ragged):
-                        vals = ids.values + off
-                        splits = ids.row_splits
+                        val_parts.append(ids.values)
+                        len_parts.append(ids.row_lengths())
                         nrows = ids.nrows
+                        device = ids.values.device
                     else:
-                        vals = ids.reshape(-1) + off
                         h = ids.shape[1] if ids.dim() > 1 else 1
                         nrows = ids.shape[0]
-                        splits = torch.arange(0, nrows * h + 1, h,
-                                              device=vals.device, dtype=torch.long)
-                    val_parts.append(vals)
-                    split_parts.append((splits, nrows))
+                        device = ids.device
+                        val_parts.append(ids.reshape(-1))
+                        len_parts.append(torch.full((nrows,), h, dtype=torch.long,
+                                                    device=device))
+                    row_offs.append((off, nrows))
                     metas.append((j, nrows))
                 allvals = torch.cat(val_parts)
-                # merge row_splits with cumulative value offsets
-                merged = [split_parts[0][0]]
-                val_off = int(split_parts[0][0][-1])
-                for splits, _ in split_parts[1:]:
-                    merged.append(splits[1:] + val_off)
-                    val_off += int(splits[-1])
-                allsplits = torch.cat(merged)
+                all_lengths = torch.cat(len_parts)
+                row_off_vec = self._row_offset_vector(gi, row_offs, device)
+                if row_off_vec is not None:
+                    allvals = allvals + torch.repeat_interleave(row_off_vec, all_lengths)
+                allsplits = torch.zeros(all_lengths.numel() + 1, dtype=torch.long,
+                                        device=device)
+                torch.cumsum(all_lengths, 0, out=allsplits[1:])
                 if offload:
                     out = embedding_lookup(layer.weight,
                                            Ragged(allvals.cpu(), allsplits.cpu()),
@@ -485,6 +488,36 @@ class DistributedEmbedding(nn.Module):
                 for (j, nrows), part in zip(metas, parts):
                     outs[j] = part
         return outs
+
+    def _offset_vector(self, gi, spec, device):
+        """Per-element fused-table offsets (cached; None if all zero)."""
+        key = ("elem", gi, tuple(spec), str(device))
+        cache = getattr(self, "_off_cache", None)
+        if cache is None:
+            cache = self._off_cache = {}
+        if key not in cache:
+            if all(off == 0 for off, _ in spec):
+                cache[key] = None
+            else:
+                cache[key] = torch.cat([
+                    torch.full((n,), off, dtype=torch.long, device=device)
+                    for off, n in spec])
+        return cache[key]
+
+    def _row_offset_vector(self, gi, spec, device):
+        """Per-row fused-table offsets (cached; None if all zero)."""
+        key = ("row", gi, tuple(spec), str(device))
+        cache = getattr(self, "_off_cache", None)
+        if cache is None:
+            cache = self._off_cache = {}
+        if key not in cache:
+            if all(off == 0 for off, _ in spec):
+                cache[key] = None
+            else:
+                cache[key] = torch.cat([
+                    torch.full((n,), off, dtype=torch.long, device=device)
+                    for off, n in spec])
+        return cache[key]
 
     def _call_table_parallel(self, col_inputs):
         plan = self.strategy
