@@ -223,6 +223,36 @@ void sparse_row_update(torch::Tensor weight, torch::Tensor state,
                            current_stream());
 }
 
+torch::Tensor dot_interact_fwd(torch::Tensor feats, int64_t out_w) {
+  CHECK_CUDA(feats); CHECK_CONTIG(feats);
+  TORCH_CHECK(feats.dtype() == torch::kBFloat16, "feats must be bf16");
+  TORCH_CHECK(feats.dim() == 3, "feats must be [B, F, D]");
+  const int64_t B = feats.size(0);
+  const int F = (int)feats.size(1), D = (int)feats.size(2);
+  TORCH_CHECK(F <= 32 && D % 32 == 0, "F<=32 and D%32==0 required");
+  const int tri_n = F * (F - 1) / 2;
+  TORCH_CHECK(out_w >= tri_n + D, "out width too small");
+  auto out = torch::empty({B, out_w}, feats.options());
+  launch_dot_interact_fwd(feats.data_ptr(), out.data_ptr(), B, F, D,
+                          (int)out_w, tri_n, current_stream());
+  return out;
+}
+
+torch::Tensor dot_interact_bwd(torch::Tensor gout, torch::Tensor feats) {
+  CHECK_CUDA(gout); CHECK_CUDA(feats);
+  CHECK_CONTIG(gout); CHECK_CONTIG(feats);
+  TORCH_CHECK(gout.dtype() == torch::kBFloat16 &&
+              feats.dtype() == torch::kBFloat16);
+  const int64_t B = feats.size(0);
+  const int F = (int)feats.size(1), D = (int)feats.size(2);
+  const int tri_n = F * (F - 1) / 2;
+  auto gfeats = torch::empty_like(feats);
+  launch_dot_interact_bwd(gout.data_ptr(), feats.data_ptr(),
+                          gfeats.data_ptr(), B, F, D, (int)gout.size(1), tri_n,
+                          current_stream());
+  return gfeats;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -235,4 +265,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "open-addressing hash vocab build + lookup (gfx950)");
   m.def("sparse_row_update", &sparse_row_update,
         "fused sparse SGD/Adagrad row update (gfx950)");
+  m.def("dot_interact_fwd", &dot_interact_fwd,
+        "fused DLRM pairwise-dot interaction forward (MFMA bf16, gfx950)");
+  m.def("dot_interact_bwd", &dot_interact_bwd,
+        "fused DLRM pairwise-dot interaction backward (MFMA bf16, gfx950)");
 }

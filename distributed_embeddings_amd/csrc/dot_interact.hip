@@ -1,0 +1,200 @@
+// Fused DLRM pairwise-dot interaction for gfx950 (MFMA bf16).
+//
+// forward:  feats [B, F, D] bf16 (feature 0 = bottom-MLP output)
+//        -> out [B, OUT] bf16 = [tril(feats @ feats^T) | feats[:,0,:] | 0-pad]
+// backward: gout [B, OUT] bf16 -> gfeats [B, F, D] bf16
+//           (gfeats = (G + G^T) @ feats, + bottom-concat grad into row 0)
+//
+// Replaces the reference's stack + bmm + tril boolean-mask + concat chain
+// (examples/dlrm/utils.py:92-113) with one kernel each way: per sample the
+// 32x32 Gram tile is 12 v_mfma_f32_16x16x32_bf16 ops (lower-triangle tiles
+// only), staged through an LDS copy of the sample's [32, D] feature block.
+// Requirements: F <= 32, D % 32 == 0, bf16 inputs.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include "ops_api.h"
+
+#define WAVE 64
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// A/B fragment gather for v_mfma_f32_16x16x32_bf16:
+//   A[row = lane%16][k = (lane/16)*8 + r],  B[k][col = lane%16] — r in [0,8).
+// C/D: col = lane&15, row = (lane>>4)*4 + reg.
+// (cdna_hip_programming.md §3; verified on hardware by the asymmetric
+// bmm-oracle test in tests/test_gpu.py.)
+
+// Forward: one wave per sample.
+template <int FMAX>  // padded feature count (32)
+__global__ void dot_interact_fwd(const __hip_bfloat16* __restrict__ feats,
+                                 __hip_bfloat16* __restrict__ out, int64_t B,
+                                 int F, int D, int out_w, int tri_n) {
+  extern __shared__ short lds_all[];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  short* lds = lds_all + wave * FMAX * D;  // [FMAX][D] bf16 (as short)
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+
+  for (int64_t b = wave_id; b < B; b += n_waves) {
+    const short* src = reinterpret_cast<const short*>(feats) + b * (int64_t)F * D;
+    // stage sample into LDS (pad rows >= F with zeros), 8 bf16 per lane-step
+    const int total8 = FMAX * D / 8;
+    for (int i = lane; i < total8; i += WAVE) {
+      const int elem = i * 8;
+      if (elem < F * D) {
+        *reinterpret_cast<bf16x8*>(&lds[elem]) =
+            *reinterpret_cast<const bf16x8*>(&src[elem]);
+      } else {
+        bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        *reinterpret_cast<bf16x8*>(&lds[elem]) = z;
+      }
+    }
+    // per-wave LDS: hipcc inserts the lgkm waits for its own ds ops
+
+    const int r16 = lane & 15;      // fragment row/col within 16
+    const int khalf = lane >> 4;    // 0..3 -> k = khalf*8 + r
+
+    // lower-triangle 16x16 tiles: (0,0), (1,0), (1,1)
+    const int tiles_mi[3] = {0, 1, 1};
+    const int tiles_ni[3] = {0, 0, 1};
+#pragma unroll
+    for (int t = 0; t < 3; ++t) {
+      const int mi = tiles_mi[t], ni = tiles_ni[t];
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      for (int k0 = 0; k0 < D; k0 += 32) {
+        // a_frag: rows of tile mi; b_frag: rows of tile ni (B = A^T)
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &lds[(mi * 16 + r16) * D + k0 + khalf * 8]);
+        bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+            &lds[(ni * 16 + r16) * D + k0 + khalf * 8]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
+      }
+      // scatter lower-triangle entries
+      __hip_bfloat16* orow = out + b * (int64_t)out_w;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int i = mi * 16 + (lane >> 4) * 4 + reg;
+        const int j = ni * 16 + (lane & 15);
+        if (i > j && i < F && j < F) {
+          orow[i * (i - 1) / 2 + j] = __hip_bfloat16(acc[reg]);
+        }
+      }
+    }
+    // bottom-MLP re-concat + zero pad (bit copy of bf16 pattern)
+    short* orow_s = reinterpret_cast<short*>(out + b * (int64_t)out_w);
+    for (int c = lane; c < D; c += WAVE) {
+      orow_s[tri_n + c] = lds[c];
+    }
+    for (int c = tri_n + D + lane; c < out_w; c += WAVE) {
+      orow_s[c] = 0;
+    }
+  }
+}
+
+// Backward: one wave per sample.  gfeats = (G + G^T) @ feats + bottom grad.
+template <int FMAX>
+__global__ void dot_interact_bwd(const __hip_bfloat16* __restrict__ gout,
+                                 const __hip_bfloat16* __restrict__ feats,
+                                 __hip_bfloat16* __restrict__ gfeats, int64_t B,
+                                 int F, int D, int out_w, int tri_n) {
+  extern __shared__ short lds_all[];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  // layout per wave: [FMAX][D] feats  +  [FMAX][FMAX] G_sym
+  short* lds = lds_all + wave * (FMAX * D + FMAX * FMAX);
+  short* gsym = lds + FMAX * D;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+
+  for (int64_t b = wave_id; b < B; b += n_waves) {
+    const short* src = reinterpret_cast<const short*>(feats) + b * (int64_t)F * D;
+    const int total8 = FMAX * D / 8;
+    for (int i = lane; i < total8; i += WAVE) {
+      const int elem = i * 8;
+      if (elem < F * D) {
+        *reinterpret_cast<bf16x8*>(&lds[elem]) =
+            *reinterpret_cast<const bf16x8*>(&src[elem]);
+      } else {
+        bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        *reinterpret_cast<bf16x8*>(&lds[elem]) = z;
+      }
+    }
+    // build G_sym [FMAX][FMAX] from the tril grad
+    const __hip_bfloat16* grow = gout + b * (int64_t)out_w;
+    for (int idx = lane; idx < FMAX * FMAX; idx += WAVE) {
+      const int i = idx / FMAX, j = idx % FMAX;
+      float g = 0.f;
+      if (i < F && j < F && i != j) {
+        const int r = i > j ? i : j, c = i > j ? j : i;
+        g = float(grow[r * (r - 1) / 2 + c]);
+      }
+      __hip_bfloat16 hb(g);
+      gsym[idx] = *reinterpret_cast<short*>(&hb);
+    }
+
+    const int r16 = lane & 15;
+    const int khalf = lane >> 4;
+    __hip_bfloat16* gf = gfeats + b * (int64_t)F * D;
+
+    // grad_A = G_sym @ A : M=FMAX rows, N=D cols, K=FMAX (=32: one K step)
+    for (int nj = 0; nj < D / 16; ++nj) {
+#pragma unroll
+      for (int mi = 0; mi < FMAX / 16; ++mi) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int k0 = 0; k0 < FMAX; k0 += 32) {
+          // A-op = G_sym[mi*16 + r16][k0 + khalf*8 + r]
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &gsym[(mi * 16 + r16) * FMAX + k0 + khalf * 8]);
+          // B-op[k][col] = feats[k][nj*16 + r16]: k = k0 + khalf*8 + r
+          bf16x8 bfr;
+#pragma unroll
+          for (int r = 0; r < 8; ++r) {
+            bfr[r] = lds[(k0 + khalf * 8 + r) * D + nj * 16 + r16];
+          }
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int i = mi * 16 + (lane >> 4) * 4 + reg;
+          const int j = nj * 16 + (lane & 15);
+          if (i < F && j < D) {
+            float v = acc[reg];
+            if (i == 0) v += float(grow[tri_n + j]);  // bottom-concat grad
+            gf[i * D + j] = __hip_bfloat16(v);
+          }
+        }
+      }
+    }
+  }
+}
+
+void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,
+                             int D, int out_w, int tri_n, hipStream_t stream) {
+  const int block = 256;
+  const int waves = block / WAVE;
+  int64_t blocks = (B + waves - 1) / waves;
+  if (blocks > 8192) blocks = 8192;
+  const size_t lds = (size_t)waves * 32 * D * sizeof(short);
+  hipLaunchKernelGGL((dot_interact_fwd<32>), dim3((int)blocks), dim3(block),
+                     lds, stream, (const __hip_bfloat16*)feats,
+                     (__hip_bfloat16*)out, B, F, D, out_w, tri_n);
+}
+
+void launch_dot_interact_bwd(const void* gout, const void* feats, void* gfeats,
+                             int64_t B, int F, int D, int out_w, int tri_n,
+                             hipStream_t stream) {
+  const int block = 256;
+  const int waves = block / WAVE;
+  int64_t blocks = (B + waves - 1) / waves;
+  if (blocks > 8192) blocks = 8192;
+  const size_t lds = (size_t)waves * (32 * D + 32 * 32) * sizeof(short);
+  hipLaunchKernelGGL((dot_interact_bwd<32>), dim3((int)blocks), dim3(block),
+                     lds, stream, (const __hip_bfloat16*)gout,
+                     (const __hip_bfloat16*)feats, (__hip_bfloat16*)gfeats, B,
+                     F, D, out_w, tri_n);
+}

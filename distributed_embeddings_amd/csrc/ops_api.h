@@ -68,3 +68,9 @@ void launch_sparse_row_update(float* weight, float* state, const int64_t* ids,
                               const float* grad, int64_t num_rows, int width,
                               float lr, float eps, bool adagrad,
                               hipStream_t stream);
+
+void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,
+                             int D, int out_w, int tri_n, hipStream_t stream);
+void launch_dot_interact_bwd(const void* gout, const void* feats, void* gfeats,
+                             int64_t B, int F, int D, int out_w, int tri_n,
+                             hipStream_t stream);

@@ -20,6 +20,7 @@ import torch
 from torch import nn
 
 from ..layers.embedding import Embedding, scaled_uniform_init
+from ..ops.dot_interact import dot_interact as fused_dot_interact
 from ..parallel import comm
 from ..parallel.dist_embedding import DistributedEmbedding
 
@@ -123,5 +124,5 @@ class DLRM(nn.Module):
         bottom = self.bottom_mlp(numerical)
         emb = self.embeddings(list(cat_features))
         emb = [e.to(bottom.dtype) for e in emb]
-        x = dot_interact(emb, bottom, pad_to=self.interact_pad)
+        x = fused_dot_interact(emb, bottom, pad_to=self.interact_pad)
         return self.top_mlp(x)

@@ -1,0 +1,52 @@
+"""Fused DLRM pairwise-dot interaction op (MFMA bf16, gfx950).
+
+Replaces the stack + bmm + tril-mask + concat chain of the reference
+(``examples/dlrm/utils.py:92-113``) with one MFMA kernel each direction
+(``csrc/dot_interact.hip``).  Falls back to plain torch ops off-GPU or for
+unsupported shapes (F > 32 or D % 32 != 0).
+"""
+
+from typing import List
+
+import torch
+
+from . import _backend
+
+
+class _DotInteract(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, feats, out_w):
+        ctx.save_for_backward(feats)
+        return _backend.ops().dot_interact_fwd(feats, out_w)
+
+    @staticmethod
+    def backward(ctx, gout):
+        (feats,) = ctx.saved_tensors
+        gfeats = _backend.ops().dot_interact_bwd(gout.contiguous(), feats)
+        return gfeats, None
+
+
+def _torch_dot_interact(feats: torch.Tensor, pad_to: int) -> torch.Tensor:
+    bottom = feats[:, 0, :]
+    gram = torch.bmm(feats, feats.transpose(1, 2))
+    f = gram.shape[1]
+    ii, jj = torch.tril_indices(f, f, offset=-1, device=gram.device)
+    interactions = gram[:, ii, jj]
+    parts = [interactions, bottom]
+    width = interactions.shape[1] + bottom.shape[1]
+    if pad_to > width:
+        parts.append(bottom.new_zeros(bottom.shape[0], pad_to - width))
+    return torch.cat(parts, dim=1)
+
+
+def dot_interact(emb_outs: List[torch.Tensor], bottom_mlp_out: torch.Tensor,
+                 pad_to: int = 0) -> torch.Tensor:
+    """[tril(feats @ feats^T) | bottom | 0-pad], feats = [bottom] + emb_outs."""
+    feats = torch.stack([bottom_mlp_out] + emb_outs, dim=1)
+    f, d = feats.shape[1], feats.shape[2]
+    width = f * (f - 1) // 2 + d
+    out_w = max(pad_to, width)
+    if (feats.is_cuda and feats.dtype == torch.bfloat16 and f <= 32
+            and d % 32 == 0):
+        return _DotInteract.apply(feats.contiguous(), out_w)
+    return _torch_dot_interact(feats, out_w)

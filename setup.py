@@ -18,6 +18,7 @@ ext = CUDAExtension(
     sources=[
         "distributed_embeddings_amd/csrc/bindings.cpp",
         "distributed_embeddings_amd/csrc/embedding_ops.hip",
+        "distributed_embeddings_amd/csrc/dot_interact.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -244,3 +244,45 @@ def test_long_segment_forward_vs_cpu():
     out = embedding_lookup(w16.cuda(), r, "sum")
     ref = _csr_lookup_ref(w16, vals, Ragged.from_row_lengths(vals, torch.tensor(lens)).row_splits, "sum")
     assert torch.allclose(out.cpu(), ref, atol=1e-3)
+
+
+@requires_gpu
+def test_dot_interact_fused_vs_torch():
+    """MFMA dot-interact vs the torch bmm oracle — asymmetric data so an
+    operand/output transpose cannot pass (guide G9)."""
+    from distributed_embeddings_amd.ops.dot_interact import (
+        _DotInteract, _torch_dot_interact)
+    torch.manual_seed(13)
+    B, F, D = 512, 27, 128
+    pad = 512
+    feats = (torch.randn(B, F, D) * (1 + torch.arange(F).view(1, F, 1) * 0.1)
+             ).bfloat16().cuda().requires_grad_(True)
+    feats_ref = feats.detach().clone().requires_grad_(True)
+
+    out = _DotInteract.apply(feats, pad)
+    ref = _torch_dot_interact(feats_ref, pad)
+    assert out.shape == ref.shape
+    df = (out.float() - ref.float()).abs()
+    assert float(df.max()) < 0.5, f"fwd max err {df.max()} (bf16 K=128 dots)"
+    # relative check on the big entries
+    rel = df / ref.float().abs().clamp(min=1.0)
+    assert float(rel.max()) < 0.02, f"fwd rel err {rel.max()}"
+
+    gout = torch.randn_like(out)
+    out.backward(gout)
+    ref.backward(gout)
+    gd = (feats.grad.float() - feats_ref.grad.float()).abs()
+    grel = gd / feats_ref.grad.float().abs().clamp(min=1.0)
+    assert float(grel.max()) < 0.05, f"bwd rel err {grel.max()}"
+
+
+@requires_gpu
+def test_dot_interact_small_shapes():
+    from distributed_embeddings_amd.ops.dot_interact import (
+        _DotInteract, _torch_dot_interact)
+    for F, D in [(4, 32), (16, 64), (32, 128)]:
+        feats = torch.randn(8, F, D).bfloat16().cuda()
+        out = _DotInteract.apply(feats, 0 if F > 4 else F*(F-1)//2 + D)
+        ref = _torch_dot_interact(feats, out.shape[1])
+        err = (out.float() - ref.float()).abs().max()
+        assert float(err) < 0.2, f"F={F} D={D}: {err}"
