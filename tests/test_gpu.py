@@ -260,20 +260,21 @@ def test_dot_interact_fused_vs_torch():
     feats_ref = feats.detach().clone().requires_grad_(True)
 
     out = _DotInteract.apply(feats, pad)
-    ref = _torch_dot_interact(feats_ref, pad)
+    # fp32 oracle; outputs are bf16 so allow ~2 ulp relative (2^-7)
+    ref = _torch_dot_interact(feats_ref.float(), pad)
     assert out.shape == ref.shape
-    df = (out.float() - ref.float()).abs()
-    assert float(df.max()) < 0.5, f"fwd max err {df.max()} (bf16 K=128 dots)"
-    # relative check on the big entries
-    rel = df / ref.float().abs().clamp(min=1.0)
-    assert float(rel.max()) < 0.02, f"fwd rel err {rel.max()}"
+    df = (out.float() - ref).abs()
+    tol = ref.abs() * 2 ** -6 + 0.5
+    assert bool((df <= tol).all()), \
+        f"fwd err beyond bf16 ulp: max {(df - tol).max()}"
 
     gout = torch.randn_like(out)
     out.backward(gout)
-    ref.backward(gout)
+    ref.backward(gout.float())
     gd = (feats.grad.float() - feats_ref.grad.float()).abs()
-    grel = gd / feats_ref.grad.float().abs().clamp(min=1.0)
-    assert float(grel.max()) < 0.05, f"bwd rel err {grel.max()}"
+    gtol = feats_ref.grad.float().abs() * 2 ** -5 + 0.5
+    assert bool((gd <= gtol).all()), \
+        f"bwd err beyond bf16 ulp: max {(gd - gtol).max()}"
 
 
 @requires_gpu
@@ -282,7 +283,8 @@ def test_dot_interact_small_shapes():
         _DotInteract, _torch_dot_interact)
     for F, D in [(4, 32), (16, 64), (32, 128)]:
         feats = torch.randn(8, F, D).bfloat16().cuda()
-        out = _DotInteract.apply(feats, 0 if F > 4 else F*(F-1)//2 + D)
-        ref = _torch_dot_interact(feats, out.shape[1])
-        err = (out.float() - ref.float()).abs().max()
-        assert float(err) < 0.2, f"F={F} D={D}: {err}"
+        out = _DotInteract.apply(feats, F * (F - 1) // 2 + D)
+        ref = _torch_dot_interact(feats.float(), out.shape[1])
+        df = (out.float() - ref).abs()
+        tol = ref.abs() * 2 ** -6 + 0.5
+        assert bool((df <= tol).all()), f"F={F} D={D}"
