@@ -32,6 +32,8 @@ def parse_args():
     p.add_argument("--pool", type=int, default=8, help="pre-generated batch pool size")
     p.add_argument("--strategy", type=str, default="memory_balanced")
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--fused-sgd", dest="fused_sgd", action="store_true", default=True)
+    p.add_argument("--no-fused-sgd", dest="fused_sgd", action="store_false")
     return p.parse_args()
 
 
@@ -99,6 +101,10 @@ def main():
     from distributed_embeddings_amd.parallel.optim import SparseEmbeddingOptimizer
     opt = de.DistributedOptimizer(
         SparseEmbeddingOptimizer(model.parameters(), lr=1e-3, method="sgd"))
+    if args.fused_sgd and hasattr(model, "embeddings"):
+        # in-backward fused SGD for the model-parallel tables (same SGD math,
+        # applied during backward; optimizer still updates the dense params)
+        model.embeddings.enable_fused_sgd(1e-3)
     de.broadcast_parameters(model)
     loss_fn = torch.nn.BCEWithLogitsLoss()
     use_bf16 = args.dtype == "bf16"

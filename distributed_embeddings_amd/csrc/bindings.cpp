@@ -223,6 +223,25 @@ void sparse_row_update(torch::Tensor weight, torch::Tensor state,
                            current_stream());
 }
 
+void csr_fused_sgd_apply(torch::Tensor weight, torch::Tensor values,
+                         torch::Tensor row_splits, torch::Tensor grad_out,
+                         torch::Tensor lr, bool mean) {
+  CHECK_CUDA(weight); CHECK_CUDA(values); CHECK_CUDA(row_splits);
+  CHECK_CUDA(grad_out); CHECK_CUDA(lr);
+  CHECK_CONTIG(weight); CHECK_CONTIG(values); CHECK_CONTIG(row_splits);
+  CHECK_CONTIG(grad_out);
+  TORCH_CHECK(weight.dtype() == torch::kFloat32 &&
+              grad_out.dtype() == torch::kFloat32 &&
+              lr.dtype() == torch::kFloat32);
+  const int64_t num_rows = row_splits.numel() - 1;
+  if (num_rows <= 0) return;
+  launch_csr_fused_sgd(weight.data_ptr<float>(), values.data_ptr<int64_t>(),
+                       row_splits.data_ptr<int64_t>(),
+                       grad_out.data_ptr<float>(), lr.data_ptr<float>(),
+                       num_rows, weight.size(0), (int)weight.size(1), mean,
+                       current_stream());
+}
+
 torch::Tensor dot_interact_fwd(torch::Tensor feats, int64_t out_w) {
   CHECK_CUDA(feats); CHECK_CONTIG(feats);
   TORCH_CHECK(feats.dtype() == torch::kBFloat16, "feats must be bf16");
@@ -265,6 +284,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "open-addressing hash vocab build + lookup (gfx950)");
   m.def("sparse_row_update", &sparse_row_update,
         "fused sparse SGD/Adagrad row update (gfx950)");
+  m.def("csr_fused_sgd_apply", &csr_fused_sgd_apply,
+        "in-backward fused SGD scatter update (gfx950)");
   m.def("dot_interact_fwd", &dot_interact_fwd,
         "fused DLRM pairwise-dot interaction forward (MFMA bf16, gfx950)");
   m.def("dot_interact_bwd", &dot_interact_bwd,

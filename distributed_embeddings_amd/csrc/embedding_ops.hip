@@ -543,6 +543,73 @@ void launch_find_valid_bounds(const int64_t* sorted_ids, int64_t n,
 }
 
 // ---------------------------------------------------------------------------
+// Fused in-backward SGD: scatter-apply  w[id] -= lr * (mean? 1/len : 1) *
+// grad_out[row]  directly during autograd backward.  Exact for SGD (updates
+// are linear, order-free modulo fp rounding); removes the whole
+// sort/unique/segmented-sum pipeline AND its num_unique host sync, so the
+// training step becomes hipGraph-capturable.  One wave per CSR row, lanes
+// across the width, float atomicAdd into the table.
+// ---------------------------------------------------------------------------
+
+template <int VEC, bool MEAN>
+__global__ void csr_fused_sgd(float* __restrict__ weight,
+                              const int64_t* __restrict__ values,
+                              const int64_t* __restrict__ splits,
+                              const float* __restrict__ grad_out,
+                              const float* __restrict__ lr_ptr,
+                              int64_t num_rows, int64_t vocab, int width) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const float lr = *lr_ptr;
+  constexpr int CHUNK = WAVE * VEC;
+  for (int64_t row = wave_id; row < num_rows; row += n_waves) {
+    const int64_t s = splits[row], e = splits[row + 1];
+    if (e == s) continue;
+    const float scale = MEAN ? -lr / (float)(e - s) : -lr;
+    for (int cbase = 0; cbase < width; cbase += CHUNK) {
+      float g[VEC];
+      const int col0 = cbase + lane * VEC;
+#pragma unroll
+      for (int v = 0; v < VEC; ++v) {
+        g[v] = (col0 + v < width)
+                   ? scale * grad_out[row * (int64_t)width + col0 + v]
+                   : 0.f;
+      }
+      for (int64_t k = s; k < e; ++k) {
+        const int64_t idx = values[k];
+        if (idx < 0 || idx >= vocab) continue;
+        float* wp = weight + idx * (int64_t)width + col0;
+#pragma unroll
+        for (int v = 0; v < VEC; ++v) {
+          if (col0 + v < width) atomicAdd(&wp[v], g[v]);
+        }
+      }
+    }
+  }
+}
+
+void launch_csr_fused_sgd(float* weight, const int64_t* values,
+                          const int64_t* splits, const float* grad_out,
+                          const float* lr, int64_t num_rows, int64_t vocab,
+                          int width, bool mean, hipStream_t stream) {
+  const int block = 256;
+  const int grid = pick_grid(num_rows, block / WAVE);
+#define LS(VEC, MEAN)                                                       \
+  hipLaunchKernelGGL((csr_fused_sgd<VEC, MEAN>), dim3(grid), dim3(block), 0, \
+                     stream, weight, values, splits, grad_out, lr, num_rows, \
+                     vocab, width)
+  if (width % 4 == 0) {
+    if (mean) LS(4, true); else LS(4, false);
+  } else if (width % 2 == 0) {
+    if (mean) LS(2, true); else LS(2, false);
+  } else {
+    if (mean) LS(1, true); else LS(1, false);
+  }
+#undef LS
+}
+
+// ---------------------------------------------------------------------------
 // Fused sparse optimizer steps: apply (unique_ids, unique_grad) rows directly
 // to the table — no torch sparse re-coalesce, no dense grad materialization.
 // One wave per row (width>64) or sub-wave tiles (narrow), same tiling as the

@@ -64,6 +64,11 @@ void launch_integer_lookup(const int64_t* keys, int64_t n, int64_t* tkeys,
 
 size_t integer_lookup_temp_bytes(int64_t max_tokens);
 
+void launch_csr_fused_sgd(float* weight, const int64_t* values,
+                          const int64_t* splits, const float* grad_out,
+                          const float* lr, int64_t num_rows, int64_t vocab,
+                          int width, bool mean, hipStream_t stream);
+
 void launch_sparse_row_update(float* weight, float* state, const int64_t* ids,
                               const float* grad, int64_t num_rows, int width,
                               float lr, float eps, bool adagrad,

@@ -15,7 +15,7 @@ from typing import Callable, Optional, Union
 import torch
 from torch import nn
 
-from ..ops.embedding_lookup import Ragged, embedding_lookup
+from ..ops.embedding_lookup import Ragged, csr_lookup_fused_sgd, embedding_lookup
 
 
 def _default_init(weight: torch.Tensor) -> None:
@@ -83,6 +83,30 @@ class Embedding(nn.Module):
     def extra_repr(self) -> str:
         return f"input_dim={self.input_dim}, output_dim={self.output_dim}, combiner={self.combiner}"
 
+    def enable_fused_sgd(self, lr: float):
+        """In-backward SGD: the lookup's backward scatter-applies
+        ``w[id] -= lr * grad`` directly (no grad tensor, no host sync; the
+        step becomes hipGraph-capturable).  Exact for SGD.  The optimizer
+        must not also update this weight (its ``.grad`` stays ``None``)."""
+        self.register_buffer("_fused_lr",
+                             torch.tensor([float(lr)], dtype=torch.float32,
+                                          device=self.weight.device))
+        return self
+
+    def set_fused_lr(self, lr: float):
+        if getattr(self, "_fused_lr", None) is None:
+            raise RuntimeError("enable_fused_sgd() first")
+        with torch.no_grad():
+            self._fused_lr.fill_(float(lr))
+
+    def csr_lookup(self, values: torch.Tensor, row_splits: torch.Tensor,
+                   combiner: str) -> torch.Tensor:
+        """CSR lookup through this layer (fused-SGD aware)."""
+        if getattr(self, "_fused_lr", None) is not None and self.training:
+            return csr_lookup_fused_sgd(self.weight, values, row_splits,
+                                        combiner, self._fused_lr)
+        return embedding_lookup(self.weight, Ragged(values, row_splits), combiner)
+
     def get_config(self) -> dict:
         """Planner-facing config (reference uses keras ``get_config()``)."""
         return {
@@ -97,7 +121,7 @@ class Embedding(nn.Module):
             # hotness-1 CSR: same gather, sparse (IndexedSlices-style) grad,
             # OOB ids contribute zero rows.
             splits = torch.arange(flat.numel() + 1, device=flat.device, dtype=torch.long)
-            out = embedding_lookup(self.weight, Ragged(flat, splits), "sum")
+            out = self.csr_lookup(flat, splits, "sum")
         elif self._oob_zero:
             valid = (flat >= 0) & (flat < self.input_dim)
             safe = torch.where(valid, flat, torch.zeros_like(flat))
@@ -111,8 +135,7 @@ class Embedding(nn.Module):
         if isinstance(ids, Ragged):
             if self.combiner is None:
                 raise ValueError("Ragged input requires a combiner")
-            return embedding_lookup(self.weight,
-                                    Ragged(ids.values.long(), ids.row_splits.long()),
+            return self.csr_lookup(ids.values.long(), ids.row_splits.long(),
                                     self.combiner)
 
         if ids.layout == torch.sparse_coo:

@@ -161,6 +161,45 @@ def _csr_lookup_backward_ref(grad_out, values, row_splits, vocab, combiner):
     return unique_ids, unique_grad
 
 
+class _CsrLookupFusedSGD(torch.autograd.Function):
+    """CSR lookup whose backward applies the SGD update in place.
+
+    Exact for SGD (the update is linear in the grad, order-free up to fp
+    rounding).  No gradient tensor is ever materialized for the table, no
+    sort/unique pipeline runs, and there is no host sync — the whole training
+    step becomes hipGraph-capturable.  ``lr`` is a 1-element fp32 device
+    tensor so schedules can update it without touching the graph.
+    """
+
+    @staticmethod
+    def forward(ctx, weight, values, row_splits, combiner, lr):
+        ctx.save_for_backward(weight, values, row_splits, lr)
+        ctx.combiner = combiner
+        if weight.is_cuda:
+            return _backend.ops().csr_lookup_forward(weight, values, row_splits,
+                                                     combiner == "mean")
+        return _csr_lookup_ref(weight, values, row_splits, combiner)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        weight, values, row_splits, lr = ctx.saved_tensors
+        grad_out = grad_out.contiguous()
+        with torch.no_grad():
+            if weight.is_cuda:
+                _backend.ops().csr_fused_sgd_apply(weight, values, row_splits,
+                                                   grad_out,
+                                                   lr, ctx.combiner == "mean")
+            else:
+                unique_ids, unique_grad = _csr_lookup_backward_ref(
+                    grad_out, values, row_splits, weight.shape[0], ctx.combiner)
+                weight.index_add_(0, unique_ids, unique_grad * (-lr.item()))
+        return None, None, None, None, None
+
+
+def csr_lookup_fused_sgd(weight, values, row_splits, combiner, lr):
+    return _CsrLookupFusedSGD.apply(weight, values, row_splits, combiner, lr)
+
+
 def _dense_fixed_hotness(weight, ids, combiner):
     """Dense [batch, hotness] ids + combiner -> gather + reduce.
 

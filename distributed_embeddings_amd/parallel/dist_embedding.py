@@ -159,6 +159,19 @@ class DistributedEmbedding(nn.Module):
 
     # ------------------------------------------------------------------ utils
 
+    def enable_fused_sgd(self, lr: float):
+        """Enables in-backward fused SGD on all model-parallel tables (col +
+        row groups).  Data-parallel tables keep sparse grads (they need the
+        allreduce).  See Embedding.enable_fused_sgd."""
+        for lyr in list(self.col_layers) + list(self.row_layers):
+            lyr.enable_fused_sgd(lr)
+        return self
+
+    def set_fused_lr(self, lr: float):
+        for lyr in list(self.col_layers) + list(self.row_layers):
+            if getattr(lyr, "_fused_lr", None) is not None:
+                lyr.set_fused_lr(lr)
+
     def local_input_ids(self) -> List[int]:
         """Global input indices this rank serves in the table-parallel group,
         in local order (for ``dp_input=False`` callers).
@@ -440,7 +453,7 @@ class DistributedEmbedding(nn.Module):
                 else:
                     splits = torch.arange(allids.numel() + 1, device=allids.device,
                                           dtype=torch.long)
-                    emb = embedding_lookup(layer.weight, Ragged(allids, splits), "sum")
+                    emb = layer.csr_lookup(allids, splits, "sum")
                 # torch.split (not manual narrow): its backward is ONE cat
                 # instead of per-slice zero-fill + accumulate.
                 parts = torch.split(emb, [n for _, _, n in metas])
@@ -482,8 +495,7 @@ class DistributedEmbedding(nn.Module):
                                            Ragged(allvals.cpu(), allsplits.cpu()),
                                            grp.combiner).to(allvals.device)
                 else:
-                    out = embedding_lookup(layer.weight, Ragged(allvals, allsplits),
-                                           grp.combiner)
+                    out = layer.csr_lookup(allvals, allsplits, grp.combiner)
                 parts = torch.split(out, [nrows for _, nrows in metas])
                 for (j, nrows), part in zip(metas, parts):
                     outs[j] = part

@@ -123,3 +123,26 @@ def test_sparse_embedding_optimizer_cpu_matches_torch(seed):
             e2(ids).square().sum().backward()
             o1.step(); o2.step()
         assert torch.allclose(e1.weight, e2.weight, atol=1e-5), method
+
+
+def test_fused_sgd_cpu_matches_explicit(seed):
+    """In-backward fused SGD == explicit sparse-grad SGD step (CPU path)."""
+    from distributed_embeddings_amd import Embedding, Ragged, SparseEmbeddingOptimizer
+    w0 = torch.randn(40, 8)
+    lists = [[1, 2, 3], [2], [5, 5, 7], [0]]
+    up = torch.randn(4, 8)
+    for combiner in ("sum", "mean"):
+        e1 = Embedding(40, 8, combiner=combiner)
+        e2 = Embedding(40, 8, combiner=combiner)
+        with torch.no_grad():
+            e1.weight.copy_(w0); e2.weight.copy_(w0)
+        e1.enable_fused_sgd(0.1)
+        out1 = e1(Ragged.from_lists(lists))
+        out1.backward(up)
+        assert e1.weight.grad is None
+        o2 = SparseEmbeddingOptimizer(e2.parameters(), lr=0.1)
+        out2 = e2(Ragged.from_lists(lists))
+        out2.backward(up)
+        o2.step()
+        assert torch.allclose(out1, out2, atol=1e-6)
+        assert torch.allclose(e1.weight, e2.weight, atol=1e-6), combiner
