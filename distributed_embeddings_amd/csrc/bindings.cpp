@@ -226,6 +226,11 @@ void sparse_row_update(torch::Tensor weight, torch::Tensor state,
 void csr_fused_sgd_apply(torch::Tensor weight, torch::Tensor values,
                          torch::Tensor row_splits, torch::Tensor grad_out,
                          torch::Tensor lr, bool mean) {
+  // All-device sorted pipeline (no host sync; hipGraph-capturable):
+  // sort (id, pos) -> permute row-ids -> head-flag unique -> segment pad ->
+  // one direct update per unique row (long segments chunked with one atomic
+  // per partial).  Replaces both torch sparse grads and the naive atomic
+  // scatter (hot-row same-address chains).
   CHECK_CUDA(weight); CHECK_CUDA(values); CHECK_CUDA(row_splits);
   CHECK_CUDA(grad_out); CHECK_CUDA(lr);
   CHECK_CONTIG(weight); CHECK_CONTIG(values); CHECK_CONTIG(row_splits);
@@ -234,12 +239,81 @@ void csr_fused_sgd_apply(torch::Tensor weight, torch::Tensor values,
               grad_out.dtype() == torch::kFloat32 &&
               lr.dtype() == torch::kFloat32);
   const int64_t num_rows = row_splits.numel() - 1;
-  if (num_rows <= 0) return;
-  launch_csr_fused_sgd(weight.data_ptr<float>(), values.data_ptr<int64_t>(),
-                       row_splits.data_ptr<int64_t>(),
-                       grad_out.data_ptr<float>(), lr.data_ptr<float>(),
-                       num_rows, weight.size(0), (int)weight.size(1), mean,
-                       current_stream());
+  const int64_t nnz = values.numel();
+  const int64_t vocab = weight.size(0);
+  const int width = (int)weight.size(1);
+  if (num_rows <= 0 || nnz <= 0) return;
+  auto stream = current_stream();
+  auto i64 = values.options();
+  auto i32 = values.options().dtype(torch::kInt32);
+  auto f32 = grad_out.options();
+
+  auto masked = torch::empty({nnz}, i64);
+  auto iota = torch::empty({nnz}, i32);
+  launch_mask_oob_and_iota(values.data_ptr<int64_t>(), nnz, vocab,
+                           masked.data_ptr<int64_t>(), iota.data_ptr<int32_t>(),
+                           stream);
+  auto row_ids = torch::empty({nnz}, i32);
+  torch::Tensor w;
+  float* w_ptr = nullptr;
+  if (mean) {
+    w = torch::empty({nnz}, f32);
+    w_ptr = w.data_ptr<float>();
+  }
+  launch_expand_row_ids(row_splits.data_ptr<int64_t>(), num_rows,
+                        row_ids.data_ptr<int32_t>(), w_ptr, mean, stream);
+  auto sorted_ids = torch::empty({nnz}, i64);
+  auto sorted_pos = torch::empty({nnz}, i32);
+  size_t temp_bytes = csr_backward_temp_bytes(nnz, vocab);
+  auto temp = torch::empty({(int64_t)temp_bytes}, f32.dtype(torch::kUInt8));
+  auto err = run_sort_pairs(temp.data_ptr(), temp_bytes,
+                            masked.data_ptr<int64_t>(),
+                            sorted_ids.data_ptr<int64_t>(),
+                            iota.data_ptr<int32_t>(),
+                            sorted_pos.data_ptr<int32_t>(), nnz,
+                            log2_ceil(vocab + 1), stream);
+  TORCH_CHECK(err == hipSuccess, "radix_sort_pairs failed");
+  auto srow = torch::empty({nnz}, i64);
+  torch::Tensor sw;
+  float* sw_ptr = nullptr;
+  if (mean) {
+    sw = torch::empty({nnz}, f32);
+    sw_ptr = sw.data_ptr<float>();
+  }
+  launch_gather_sorted(sorted_pos.data_ptr<int32_t>(),
+                       row_ids.data_ptr<int32_t>(), w_ptr, nnz,
+                       srow.data_ptr<int64_t>(), sw_ptr, stream);
+  auto head = torch::empty({nnz}, i32);
+  auto pos = torch::empty({nnz}, i32);
+  launch_mark_heads(sorted_ids.data_ptr<int64_t>(), nnz, vocab,
+                    head.data_ptr<int32_t>(), stream);
+  err = run_inclusive_scan_i32(temp.data_ptr(), temp_bytes,
+                               head.data_ptr<int32_t>(),
+                               pos.data_ptr<int32_t>(), nnz, stream);
+  TORCH_CHECK(err == hipSuccess, "inclusive_scan failed");
+  auto unique_tmp = torch::empty({nnz}, i64);
+  auto seg_tmp = torch::empty({nnz + 1}, i64);
+  auto nu_dev = torch::zeros({1}, i32);
+  launch_scatter_unique(sorted_ids.data_ptr<int64_t>(),
+                        head.data_ptr<int32_t>(), pos.data_ptr<int32_t>(), nnz,
+                        vocab, unique_tmp.data_ptr<int64_t>(),
+                        seg_tmp.data_ptr<int64_t>(), nu_dev.data_ptr<int32_t>(),
+                        stream);
+  auto bounds = torch::empty({2}, i64);
+  launch_find_valid_bounds(sorted_ids.data_ptr<int64_t>(), nnz, vocab,
+                           bounds.data_ptr<int64_t>(), stream);
+  launch_pad_seg_offsets(seg_tmp.data_ptr<int64_t>(), nnz,
+                         nu_dev.data_ptr<int32_t>(), bounds.data_ptr<int64_t>(),
+                         stream);
+  auto long_rows = torch::empty({nnz}, i64);
+  auto long_count = torch::empty({1}, i32);
+  launch_sorted_sgd_update(weight.data_ptr<float>(),
+                           sorted_ids.data_ptr<int64_t>(),
+                           seg_tmp.data_ptr<int64_t>(), srow.data_ptr<int64_t>(),
+                           sw_ptr, grad_out.data_ptr<float>(),
+                           lr.data_ptr<float>(), nnz, width,
+                           long_rows.data_ptr<int64_t>(),
+                           long_count.data_ptr<int32_t>(), stream);
 }
 
 torch::Tensor dot_interact_fwd(torch::Tensor feats, int64_t out_w) {

@@ -610,6 +610,165 @@ void launch_csr_fused_sgd(float* weight, const int64_t* values,
 }
 
 // ---------------------------------------------------------------------------
+// Sorted fused-SGD update: after the (all-device) sort/unique pipeline, apply
+//   weight[uid] -= lr * sum_{k in segment} w_k * grad_out[srow[k]]
+// directly from the sorted segments.  No host sync (segment count lives in
+// device memory; empty padded segments no-op), no same-address atomic chains
+// (each unique row is written once; long segments use the chunk-wave + atomic
+// pattern of csr_fwd_long with one atomic per chunk partial).
+// ---------------------------------------------------------------------------
+
+__global__ void pad_seg_offsets(int64_t* __restrict__ seg, int64_t n,
+                                const int32_t* __restrict__ num_unique,
+                                const int64_t* __restrict__ bounds) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i > n) return;
+  if (i >= *num_unique) seg[i] = bounds[1];  // first OOB position
+}
+
+template <int VEC, bool HAS_W>
+__global__ void sorted_sgd_update(float* __restrict__ weight,
+                                  const int64_t* __restrict__ sorted_ids,
+                                  const int64_t* __restrict__ seg,
+                                  const int64_t* __restrict__ srow,
+                                  const float* __restrict__ sw,
+                                  const float* __restrict__ grad_out,
+                                  const float* __restrict__ lr_ptr,
+                                  int64_t max_segs, int width,
+                                  int64_t* __restrict__ long_rows,
+                                  int32_t* __restrict__ long_count) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const float lr = *lr_ptr;
+  constexpr int CHUNK = WAVE * VEC;
+  for (int64_t r = wave_id; r < max_segs; r += n_waves) {
+    const int64_t s = seg[r], e = seg[r + 1];
+    if (s >= e) continue;
+    if (e - s > LONG_T) {
+      if (lane == 0) long_rows[atomicAdd(long_count, 1)] = r;
+      continue;
+    }
+    const int64_t uid = sorted_ids[s];
+    for (int cbase = 0; cbase < width; cbase += CHUNK) {
+      float acc[VEC];
+#pragma unroll
+      for (int v = 0; v < VEC; ++v) acc[v] = 0.f;
+      const int col0 = cbase + lane * VEC;
+      for (int64_t k = s; k < e; ++k) {
+        const float w = HAS_W ? sw[k] : 1.f;
+        const float* gp = grad_out + srow[k] * (int64_t)width + col0;
+        if (VEC == 4 && col0 + 4 <= width) {
+          const float4 g4 = *reinterpret_cast<const float4*>(gp);
+          acc[0] += w * g4.x; acc[1] += w * g4.y;
+          acc[2] += w * g4.z; acc[3] += w * g4.w;
+        } else {
+#pragma unroll
+          for (int v = 0; v < VEC; ++v)
+            if (col0 + v < width) acc[v] += w * gp[v];
+        }
+      }
+      float* wp = weight + uid * (int64_t)width + col0;
+#pragma unroll
+      for (int v = 0; v < VEC; ++v)
+        if (col0 + v < width) wp[v] -= lr * acc[v];
+    }
+  }
+}
+
+template <int VEC, bool HAS_W>
+__global__ void sorted_sgd_update_long(float* __restrict__ weight,
+                                       const int64_t* __restrict__ sorted_ids,
+                                       const int64_t* __restrict__ seg,
+                                       const int64_t* __restrict__ srow,
+                                       const float* __restrict__ sw,
+                                       const float* __restrict__ grad_out,
+                                       const float* __restrict__ lr_ptr,
+                                       int width,
+                                       const int64_t* __restrict__ long_rows,
+                                       const int32_t* __restrict__ long_count) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const float lr = *lr_ptr;
+  const int64_t n_long = *long_count;
+  const int64_t n_items = n_long * MAX_CHUNKS;
+  for (int64_t item = wave_id; item < n_items; item += n_waves) {
+    const int64_t li = item / MAX_CHUNKS;
+    const int64_t chunk = item % MAX_CHUNKS;
+    const int64_t r = long_rows[li];
+    const int64_t s = seg[r], e = seg[r + 1];
+    const int64_t k0 = s + chunk * LONG_T;
+    if (k0 >= e) continue;
+    const int64_t uid = sorted_ids[s];
+    constexpr int V = VEC > 0 ? VEC : 1;
+    constexpr int CH = WAVE * V;
+    for (int cbase = 0; cbase < width; cbase += CH) {
+      float acc[V];
+#pragma unroll
+      for (int v = 0; v < V; ++v) acc[v] = 0.f;
+      const int col0 = cbase + lane * V;
+      for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
+        const int64_t ke = min(ks + (int64_t)LONG_T, e);
+        for (int64_t k = ks; k < ke; ++k) {
+          const float w = HAS_W ? sw[k] : 1.f;
+          const float* gp = grad_out + srow[k] * (int64_t)width + col0;
+          if (V == 4 && col0 + 4 <= width) {
+            const float4 g4 = *reinterpret_cast<const float4*>(gp);
+            acc[0] += w * g4.x; acc[1] += w * g4.y;
+            acc[2] += w * g4.z; acc[3] += w * g4.w;
+          } else {
+#pragma unroll
+            for (int v = 0; v < V; ++v)
+              if (col0 + v < width) acc[v] += w * gp[v];
+          }
+        }
+      }
+      float* wp = weight + uid * (int64_t)width + col0;
+#pragma unroll
+      for (int v = 0; v < V; ++v)
+        if (col0 + v < width) atomicAdd(&wp[v], -lr * acc[v]);
+    }
+  }
+}
+
+void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
+                            const int64_t* bounds, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(pad_seg_offsets, dim3((int)cdiv64(n + 1, block)),
+                     dim3(block), 0, stream, seg, n, num_unique, bounds);
+}
+
+void launch_sorted_sgd_update(float* weight, const int64_t* sorted_ids,
+                              const int64_t* seg, const int64_t* srow,
+                              const float* sw, const float* grad_out,
+                              const float* lr, int64_t max_segs, int width,
+                              int64_t* long_rows, int32_t* long_count,
+                              hipStream_t stream) {
+  const int block = 256;
+  hipMemsetAsync(long_count, 0, sizeof(int32_t), stream);
+  const int grid = pick_grid(max_segs, block / WAVE);
+#define LU(VEC, HASW)                                                          \
+  do {                                                                         \
+    hipLaunchKernelGGL((sorted_sgd_update<VEC, HASW>), dim3(grid),             \
+                       dim3(block), 0, stream, weight, sorted_ids, seg, srow,  \
+                       sw, grad_out, lr, max_segs, width, long_rows,           \
+                       long_count);                                            \
+    hipLaunchKernelGGL((sorted_sgd_update_long<VEC, HASW>), dim3(2048),        \
+                       dim3(block), 0, stream, weight, sorted_ids, seg, srow,  \
+                       sw, grad_out, lr, width, long_rows, long_count);        \
+  } while (0)
+  if (width % 4 == 0) {
+    if (sw) LU(4, true); else LU(4, false);
+  } else if (width % 2 == 0) {
+    if (sw) LU(2, true); else LU(2, false);
+  } else {
+    if (sw) LU(1, true); else LU(1, false);
+  }
+#undef LU
+}
+
+// ---------------------------------------------------------------------------
 // Fused sparse optimizer steps: apply (unique_ids, unique_grad) rows directly
 // to the table — no torch sparse re-coalesce, no dense grad materialization.
 // One wave per row (width>64) or sub-wave tiles (narrow), same tiling as the

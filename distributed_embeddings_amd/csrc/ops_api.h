@@ -69,6 +69,16 @@ void launch_csr_fused_sgd(float* weight, const int64_t* values,
                           const float* lr, int64_t num_rows, int64_t vocab,
                           int width, bool mean, hipStream_t stream);
 
+void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
+                            const int64_t* bounds, hipStream_t stream);
+
+void launch_sorted_sgd_update(float* weight, const int64_t* sorted_ids,
+                              const int64_t* seg, const int64_t* srow,
+                              const float* sw, const float* grad_out,
+                              const float* lr, int64_t max_segs, int width,
+                              int64_t* long_rows, int32_t* long_count,
+                              hipStream_t stream);
+
 void launch_sparse_row_update(float* weight, float* state, const int64_t* ids,
                               const float* grad, int64_t num_rows, int width,
                               float lr, float eps, bool adagrad,
