@@ -91,16 +91,27 @@ def main():
     model, table_sizes, hotness, num_numerical, name, keep_hot = build_model(args, device)
     b = args.batch_per_gpu
 
-    # pre-generated input pool (parity: reference InputGenerator batch pool)
+    # pre-generated input pool (parity: reference InputGenerator batch pool).
+    # cats stored as ONE flat tensor per entry so graph replay needs a single
+    # H2H copy; per-feature views are carved out once.
     pool = []
+    cat_sizes = None
     for i in range(args.pool):
         g = torch.Generator(device="cpu").manual_seed(1000 + 131 * rank + i)
         cats = [c.to(device) for c in
                 make_batch(table_sizes, hotness, b, args.alpha, generator=g,
                            keep_hot_dim=keep_hot)]
+        if cat_sizes is None:
+            cat_shapes = [c.shape for c in cats]
+            cat_sizes = [c.numel() for c in cats]
+        flat = torch.cat([c.reshape(-1) for c in cats])
         num = torch.rand(b, num_numerical, device=device)
         labels = torch.randint(0, 2, (b, 1), device=device).float()
-        pool.append((num, cats, labels))
+        pool.append((num, flat, labels))
+
+    def carve(flat):
+        return [p.view(shape) for p, shape in
+                zip(torch.split(flat, cat_sizes), cat_shapes)]
 
     from distributed_embeddings_amd.parallel.optim import SparseEmbeddingOptimizer
     opt = de.DistributedOptimizer(
@@ -124,11 +135,10 @@ def main():
 
     graph = None
     if args.graph and world == 1:
-        # hipGraph capture: static input buffers (pool entries copied in per
-        # step), grads pre-materialized, fused SGD has no host syncs.
-        s_num, s_cats, s_labels = pool[0]
-        s_cats = [c.clone() for c in s_cats]
-        s_num, s_labels = s_num.clone(), s_labels.clone()
+        # hipGraph capture: static input buffers (one flat cat copy per step),
+        # grads pre-materialized, fused SGD has no host syncs.
+        s_num, s_flat, s_labels = (t.clone() for t in pool[0])
+        s_cats = carve(s_flat)
         try:
             for i in range(max(args.warmup, 2)):
                 run_fwd_bwd_opt(s_num, s_cats, s_labels, set_to_none=False)
@@ -142,15 +152,14 @@ def main():
             graph = None
 
     def step(i):
-        num, cats, labels = pool[i % len(pool)]
+        num, flat, labels = pool[i % len(pool)]
         if graph is not None:
             s_num.copy_(num)
-            for d, s in zip(s_cats, cats):
-                d.copy_(s)
+            s_flat.copy_(flat)
             s_labels.copy_(labels)
             graph.replay()
             return None
-        return run_fwd_bwd_opt(num, cats, labels)
+        return run_fwd_bwd_opt(num, carve(flat), labels)
 
     for i in range(args.warmup):
         step(i)

@@ -122,7 +122,6 @@ class DLRM(nn.Module):
     def forward(self, numerical: torch.Tensor,
                 cat_features: Sequence[torch.Tensor]) -> torch.Tensor:
         bottom = self.bottom_mlp(numerical)
-        emb = self.embeddings(list(cat_features))
-        emb = [e.to(bottom.dtype) for e in emb]
+        emb = self.embeddings(list(cat_features), output_dtype=bottom.dtype)
         x = fused_dot_interact(emb, bottom, pad_to=self.interact_pad)
         return self.top_mlp(x)

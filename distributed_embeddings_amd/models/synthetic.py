@@ -75,8 +75,10 @@ class SyntheticModel(nn.Module):
         self.mlp = nn.Sequential(*mods)
 
     def forward(self, numerical: torch.Tensor, cat_features) -> torch.Tensor:
-        embs = self.embeddings(list(cat_features))
-        x = torch.cat([e.to(numerical.dtype) for e in embs], dim=1)
+        # under autocast the MLP runs bf16: ask for bf16 embedding outputs
+        want = torch.bfloat16 if torch.is_autocast_enabled() else numerical.dtype
+        embs = self.embeddings(list(cat_features), output_dtype=want)
+        x = torch.cat([e.to(want) for e in embs], dim=1)
         if self.interact is not None:
             x = self.interact(x.unsqueeze(1)).squeeze(1)
         x = torch.cat([x, numerical], dim=1)

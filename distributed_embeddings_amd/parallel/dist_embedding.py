@@ -195,8 +195,10 @@ class DistributedEmbedding(nn.Module):
 
     # ---------------------------------------------------------------- forward
 
-    def forward(self, inputs: Sequence[Union[torch.Tensor, Ragged]]) -> List[torch.Tensor]:
+    def forward(self, inputs: Sequence[Union[torch.Tensor, Ragged]],
+                output_dtype: Optional[torch.dtype] = None) -> List[torch.Tensor]:
         plan = self.strategy
+        self._output_dtype = output_dtype
         dp_in, col_in, row_in = plan.input_groups
         if self.dp_input:
             if len(inputs) != len(plan.input_table_map):
@@ -213,6 +215,8 @@ class DistributedEmbedding(nn.Module):
 
         dp_out = [self.dp_layers[plan.input_maps[0][j]](x)
                   for j, x in enumerate(dp_inputs)]
+        if output_dtype is not None:
+            dp_out = [o.to(output_dtype) for o in dp_out]
         col_out = self._call_table_parallel(col_inputs) if (col_inputs or plan.col_table_ids) \
             else []
         row_out = self._call_row_slice(row_inputs) if row_inputs else []
@@ -454,6 +458,8 @@ class DistributedEmbedding(nn.Module):
                     splits = torch.arange(allids.numel() + 1, device=allids.device,
                                           dtype=torch.long)
                     emb = layer.csr_lookup(allids, splits, "sum")
+                if getattr(self, "_output_dtype", None) is not None:
+                    emb = emb.to(self._output_dtype)
                 # torch.split (not manual narrow): its backward is ONE cat
                 # instead of per-slice zero-fill + accumulate.
                 parts = torch.split(emb, [n for _, _, n in metas])
@@ -496,6 +502,8 @@ class DistributedEmbedding(nn.Module):
                                            grp.combiner).to(allvals.device)
                 else:
                     out = layer.csr_lookup(allvals, allsplits, grp.combiner)
+                if getattr(self, "_output_dtype", None) is not None:
+                    out = out.to(self._output_dtype)
                 parts = torch.split(out, [nrows for _, nrows in metas])
                 for (j, nrows), part in zip(metas, parts):
                     outs[j] = part
@@ -647,7 +655,10 @@ class DistributedEmbedding(nn.Module):
             tbl_local = plan.input_maps[2][j]
             width = self.row_layers[tbl_local].output_dim
             if self.row_layers[tbl_local].combiner is None and x.dim() > 1:
-                result.append(flat.view(b, *x.shape[1:], width))
+                out_t = flat.view(b, *x.shape[1:], width)
             else:
-                result.append(flat.view(b, width))
+                out_t = flat.view(b, width)
+            if getattr(self, "_output_dtype", None) is not None:
+                out_t = out_t.to(self._output_dtype)
+            result.append(out_t)
         return result
