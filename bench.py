@@ -115,13 +115,18 @@ def main():
 
     from distributed_embeddings_amd.parallel.optim import SparseEmbeddingOptimizer
     opt = de.DistributedOptimizer(
-        SparseEmbeddingOptimizer(model.parameters(), lr=1e-3, method="sgd"))
+        SparseEmbeddingOptimizer(model.parameters(), lr=1e-3, method="sgd"),
+        average=False)  # loss is normalized by the global batch below
     if args.fused_sgd and hasattr(model, "embeddings"):
         # in-backward fused SGD for the model-parallel tables (same SGD math,
         # applied during backward; optimizer still updates the dense params)
         model.embeddings.enable_fused_sgd(1e-3)
     de.broadcast_parameters(model)
-    loss_fn = torch.nn.BCEWithLogitsLoss()
+    # per-rank loss normalized by the GLOBAL batch: mp tables get exact
+    # global-batch grads, dp params sum to the same (see allreduce_gradients)
+    _loss_sum = torch.nn.BCEWithLogitsLoss(reduction="sum")
+    global_batch_ = b * world
+    loss_fn = lambda logits, labels: _loss_sum(logits, labels) / global_batch_
     use_bf16 = args.dtype == "bf16"
 
     def run_fwd_bwd_opt(num, cats, labels, set_to_none=True):

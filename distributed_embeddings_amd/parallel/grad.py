@@ -54,8 +54,9 @@ broadcast_variables = broadcast_parameters
 
 
 def _allreduce_dense_bucketed(params: List[torch.nn.Parameter], world: int,
-                              bucket_bytes: int = 64 << 20):
-    """Flat-bucket average of dense grads (7-link xGMI wants few, large calls)."""
+                              bucket_bytes: int = 64 << 20,
+                              average: bool = True):
+    """Flat-bucket allreduce of dense grads (7-link xGMI wants few, large calls)."""
     bucket, nbytes = [], 0
     def flush():
         nonlocal bucket, nbytes
@@ -63,7 +64,8 @@ def _allreduce_dense_bucketed(params: List[torch.nn.Parameter], world: int,
             return
         flat = torch.cat([p.grad.reshape(-1) for p in bucket])
         dist.all_reduce(flat)
-        flat /= world
+        if average:
+            flat /= world
         pos = 0
         for p in bucket:
             n = p.grad.numel()
@@ -83,12 +85,22 @@ def _allreduce_dense_bucketed(params: List[torch.nn.Parameter], world: int,
         flush()
 
 
-def allreduce_gradients(module_or_params, bucket_bytes: int = 64 << 20):
-    """Averages gradients of all non-``de_local`` params across ranks.
+def allreduce_gradients(module_or_params, bucket_bytes: int = 64 << 20,
+                        average: bool = True):
+    """Allreduces gradients of all non-``de_local`` params across ranks.
 
     Sparse gradients (data-parallel embedding tables) are densified before the
     allreduce — parity with Horovod ``sparse_as_dense=True`` in the reference
     tape/optimizer (``:1260-1262``).
+
+    Hybrid-parallel loss-scaling contract: model-parallel tables always
+    receive the *sum* of per-rank loss gradients (they are never reduced), so
+    for distributed == single-process training semantics either
+    (a) normalize the per-rank loss by the GLOBAL batch and use
+        ``average=False`` (summed dp grads) — recommended, or
+    (b) use per-rank mean losses with ``average=True`` and accept mp tables
+        seeing world_size x the single-process gradient (the reference's
+        effective behavior with Horovod averaging).
     """
     world = comm.world_size()
     if world == 1:
@@ -104,7 +116,7 @@ def allreduce_gradients(module_or_params, bucket_bytes: int = 64 << 20):
         if p.grad.layout != torch.strided:
             p.grad = p.grad.to_dense()
         dense.append(p)
-    _allreduce_dense_bucketed(dense, world, bucket_bytes)
+    _allreduce_dense_bucketed(dense, world, bucket_bytes, average)
 
 
 class DistributedOptimizer:
@@ -119,9 +131,11 @@ class DistributedOptimizer:
         loss.backward(); opt.step(); opt.zero_grad()
     """
 
-    def __init__(self, optimizer: torch.optim.Optimizer, bucket_bytes: int = 64 << 20):
+    def __init__(self, optimizer: torch.optim.Optimizer, bucket_bytes: int = 64 << 20,
+                 average: bool = True):
         self.optimizer = optimizer
         self.bucket_bytes = bucket_bytes
+        self.average = average
 
     @property
     def param_groups(self):
@@ -138,7 +152,7 @@ class DistributedOptimizer:
 
     def step(self, closure=None):
         params = [p for g in self.optimizer.param_groups for p in g["params"]]
-        allreduce_gradients(params, self.bucket_bytes)
+        allreduce_gradients(params, self.bucket_bytes, self.average)
         return self.optimizer.step(closure)
 
 

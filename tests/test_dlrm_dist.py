@@ -1,0 +1,97 @@
+"""End-to-end DLRM distributed equivalence (gloo world=2) — exercises the
+exact bench.py path: hybrid dp+mp, fused group lookups, output_dtype-cast
+all-to-all, DistributedOptimizer, fused SGD."""
+
+import numpy as np
+import pytest
+import torch
+
+from conftest import run_distributed
+
+SIZES = [50, 7, 120, 33, 64, 200]
+
+
+def _dlrm_worker(rank, world, fused_sgd, output_dtype_name):
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd.models.dlrm import DLRM
+    from distributed_embeddings_amd.parallel.optim import SparseEmbeddingOptimizer
+
+    torch.manual_seed(0)
+    model = DLRM(SIZES, embedding_dim=16, bottom_mlp_dims=(32, 16),
+                 top_mlp_dims=(32, 1), num_numerical=4,
+                 strategy="memory_balanced")
+    # deterministic weights everywhere
+    gw = torch.Generator().manual_seed(42)
+    weights = [torch.randn(s, 16, generator=gw).numpy() for s in SIZES]
+    model.embeddings.set_weights(weights)
+    for p in model.bottom_mlp.parameters():
+        torch.nn.init.normal_(p, generator=gw) if p.dim() > 1 else p.data.zero_()
+    for p in model.top_mlp.parameters():
+        torch.nn.init.normal_(p, generator=gw) if p.dim() > 1 else p.data.zero_()
+    de.broadcast_parameters(model)
+
+    if fused_sgd:
+        model.embeddings.enable_fused_sgd(0.05)
+    opt = de.DistributedOptimizer(
+        SparseEmbeddingOptimizer(model.parameters(), lr=0.05), average=False)
+
+    gi = torch.Generator().manual_seed(7)
+    B = 8
+    cats = [torch.randint(0, s, (B,), generator=gi) for s in SIZES]
+    num = torch.rand(B, 4, generator=gi)
+    labels = torch.randint(0, 2, (B, 1), generator=gi).float()
+    lb = B // world
+    sl = slice(rank * lb, (rank + 1) * lb)
+
+    for _ in range(2):
+        opt.zero_grad(set_to_none=True)
+        logits = model(num[sl], [c[sl] for c in cats])
+        loss = torch.nn.functional.binary_cross_entropy_with_logits(
+            logits.float(), labels[sl], reduction="sum") / B
+        loss.backward()
+        opt.step()
+    out = model(num[sl], [c[sl] for c in cats]).detach()
+    tables = model.embeddings.get_weights(all_ranks=True)
+    return {"out": out, "tables": [torch.as_tensor(t) for t in tables]}
+
+
+@pytest.mark.parametrize("fused_sgd", [False, True])
+def test_dlrm_world2_matches_world1(fused_sgd):
+    results2 = run_distributed(_dlrm_worker, world=2, args=(fused_sgd, None))
+    results1 = run_distributed(_dlrm_worker, world=1, args=(fused_sgd, None))
+    full_out = results1[0]["out"]
+    for rank in range(2):
+        got = results2[rank]["out"]
+        ref = full_out[rank * 4:(rank + 1) * 4]
+        assert torch.allclose(got, ref, atol=1e-4), \
+            f"rank{rank} fwd err {(got - ref).abs().max()}"
+    for t in range(len(SIZES)):
+        a = results2[0]["tables"][t]
+        b = results1[0]["tables"][t]
+        assert torch.allclose(a, b, atol=1e-4), \
+            f"table {t} err {(a - b).abs().max()}"
+
+
+def _dtype_worker(rank, world):
+    import distributed_embeddings_amd as de
+    tables = [de.TableConfig(40, 8), de.TableConfig(60, 8)]
+    model = de.DistributedEmbedding(tables, strategy="basic")
+    gw = torch.Generator().manual_seed(1)
+    weights = [torch.randn(40, 8, generator=gw).numpy(),
+               torch.randn(60, 8, generator=gw).numpy()]
+    model.set_weights(weights)
+    gi = torch.Generator().manual_seed(2)
+    inputs = [torch.randint(0, 40, (world * 4,), generator=gi),
+              torch.randint(0, 60, (world * 4,), generator=gi)]
+    sl = slice(rank * 4, (rank + 1) * 4)
+    outs = model([x[sl] for x in inputs], output_dtype=torch.bfloat16)
+    refs = [torch.from_numpy(weights[t])[inputs[t][sl]].bfloat16()
+            for t in range(2)]
+    assert all(o.dtype == torch.bfloat16 for o in outs)
+    return [float((o.float() - r.float()).abs().max()) for o, r in zip(outs, refs)]
+
+
+def test_output_dtype_bf16_a2a_world2():
+    results = run_distributed(_dtype_worker, world=2)
+    for errs in results:
+        assert max(errs) < 0.05  # bf16 rounding only
