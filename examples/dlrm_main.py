@@ -98,13 +98,15 @@ def main():
                                  dp_input=args.dp_input or world == 1)
 
     opt = de.DistributedOptimizer(SparseEmbeddingOptimizer(
-        model.parameters(), lr=args.learning_rate / args.batch_size, method="sgd"))
+        model.parameters(), lr=args.learning_rate / args.batch_size, method="sgd"),
+        average=False)  # loss normalized by global batch below
     sched = WarmupPolyDecay(opt, base_lr=args.learning_rate / args.batch_size,
                             warmup_steps=args.warmup_steps,
                             decay_start=args.decay_start,
                             decay_steps=args.decay_steps)
     de.broadcast_parameters(model)
-    loss_fn = torch.nn.BCEWithLogitsLoss()
+    _loss_sum = torch.nn.BCEWithLogitsLoss(reduction="sum")
+    loss_fn = lambda lg, lb: _loss_sum(lg, lb) / args.batch_size
 
     model.train()
     t0 = time.time()

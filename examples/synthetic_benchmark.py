@@ -72,9 +72,10 @@ def main():
         pool.append((num, cats, labels))
 
     opt = de.DistributedOptimizer(SparseEmbeddingOptimizer(
-        model.parameters(), lr=0.01, method=args.optimizer))
+        model.parameters(), lr=0.01, method=args.optimizer), average=False)
     de.broadcast_parameters(model)
-    loss_fn = torch.nn.BCEWithLogitsLoss()
+    _loss_sum = torch.nn.BCEWithLogitsLoss(reduction="sum")
+    loss_fn = lambda lg, lb: _loss_sum(lg, lb) / args.batch_size
 
     def step(i):
         num, cats, labels = pool[i % len(pool)]
