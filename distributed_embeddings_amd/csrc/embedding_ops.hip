@@ -60,7 +60,7 @@ __global__ void csr_fwd_narrow(const float* __restrict__ params,
                                const int64_t* __restrict__ splits,
                                const float* __restrict__ per_id_w,
                                float* __restrict__ out, int64_t num_rows,
-                               int64_t vocab, int width,
+                               int64_t vocab, int width, int64_t long_thresh,
                                int64_t* __restrict__ long_rows,
                                int32_t* __restrict__ long_count) {
   constexpr int RPW = WAVE / TILE;
@@ -74,7 +74,7 @@ __global__ void csr_fwd_narrow(const float* __restrict__ params,
     const int64_t row = base + sub;
     if (row >= num_rows || tl >= width) continue;
     const int64_t s = splits[row], e = splits[row + 1];
-    if (e - s > LONG_T) {
+    if (e - s > long_thresh) {
       out[row * width + tl] = 0.f;
       if (tl == 0) long_rows[atomicAdd(long_count, 1)] = row;
       continue;
@@ -98,7 +98,7 @@ __global__ void csr_fwd_wide(const float* __restrict__ params,
                              const int64_t* __restrict__ splits,
                              const float* __restrict__ per_id_w,
                              float* __restrict__ out, int64_t num_rows,
-                             int64_t vocab, int width,
+                             int64_t vocab, int width, int64_t long_thresh,
                              int64_t* __restrict__ long_rows,
                              int32_t* __restrict__ long_count) {
   const int lane = threadIdx.x & (WAVE - 1);
@@ -108,7 +108,7 @@ __global__ void csr_fwd_wide(const float* __restrict__ params,
   constexpr int CHUNK = WAVE * VEC;
   for (int64_t row = wave_id; row < num_rows; row += n_waves) {
     const int64_t s = splits[row], e = splits[row + 1];
-    if (e - s > LONG_T) {
+    if (e - s > long_thresh) {
       for (int c = lane; c < width; c += WAVE) out[row * width + c] = 0.f;
       if (lane == 0) long_rows[atomicAdd(long_count, 1)] = row;
       continue;
@@ -259,19 +259,29 @@ static void launch_csr_pair(const float* params, const int64_t* values,
   const int block = 256, bw = block / WAVE;
   const int64_t row_waves = TILE > 0 ? cdiv64(num_rows, WAVE / (TILE > 0 ? TILE : 1)) : num_rows;
   const dim3 grid(pick_grid(row_waves, bw));
+  // Adaptive long threshold: when there are plenty of rows the chip is full
+  // without splitting, so only true outliers (>4x the average and >LONG_T)
+  // offload; with few rows split aggressively for parallelism.
+  const int64_t ave = num_rows > 0 ? nnz / num_rows : 0;
+  int64_t long_thresh = LONG_T;
+  if (row_waves >= 4096) {
+    long_thresh = 4 * (ave > 0 ? ave : 1);
+    if (long_thresh < LONG_T) long_thresh = LONG_T;
+    if (long_thresh > 8192) long_thresh = 8192;
+  }
   hipMemsetAsync(long_count, 0, sizeof(int32_t), stream);
 #define LA(MEAN, HASW)                                                         \
   do {                                                                         \
     if constexpr (TILE > 0)                                                    \
       hipLaunchKernelGGL((csr_fwd_narrow<(TILE > 0 ? TILE : 1), MEAN, HASW>),  \
                          grid, dim3(block), 0, stream, params, values, splits, \
-                         per_id_w, out, num_rows, vocab, width, long_rows,     \
-                         long_count);                                          \
+                         per_id_w, out, num_rows, vocab, width, long_thresh,   \
+                         long_rows, long_count);                               \
     else                                                                       \
       hipLaunchKernelGGL((csr_fwd_wide<(VEC > 0 ? VEC : 1), MEAN, HASW>),      \
                          grid, dim3(block), 0, stream, params, values, splits, \
-                         per_id_w, out, num_rows, vocab, width, long_rows,     \
-                         long_count);                                          \
+                         per_id_w, out, num_rows, vocab, width, long_thresh,   \
+                         long_rows, long_count);                               \
   } while (0)
 #define LB(MEAN, HASW)                                                         \
   hipLaunchKernelGGL((csr_fwd_long<TILE, VEC, MEAN, HASW>), dim3(2048),        \
