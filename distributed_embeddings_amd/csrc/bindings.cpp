@@ -24,6 +24,52 @@ int log2_ceil(int64_t v) {
   return b;
 }
 
+
+// Sort dispatch: hand-written LSD radix sort (csrc/radix_sort.hip) by
+// default; DE_USE_ROCPRIM_SORT=1 switches to the rocPRIM bring-up path for
+// comparison.
+bool use_rocprim_sort() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("DE_USE_ROCPRIM_SORT");
+    v = (e && e[0] == '1') ? 1 : 0;
+  }
+  return v == 1;
+}
+
+void sort_pairs_dispatch(torch::Tensor masked, torch::Tensor iota,
+                         torch::Tensor sorted_ids, torch::Tensor sorted_pos,
+                         int64_t nnz, int end_bit, hipStream_t stream) {
+  if (use_rocprim_sort()) {
+    size_t temp_bytes = csr_backward_temp_bytes(nnz, 0);
+    auto temp = torch::empty({(int64_t)temp_bytes},
+                             masked.options().dtype(torch::kUInt8));
+    auto err = run_sort_pairs(temp.data_ptr(), temp_bytes,
+                              masked.data_ptr<int64_t>(),
+                              sorted_ids.data_ptr<int64_t>(),
+                              iota.data_ptr<int32_t>(),
+                              sorted_pos.data_ptr<int32_t>(), nnz, end_bit,
+                              stream);
+    TORCH_CHECK(err == hipSuccess, "radix_sort_pairs failed");
+    return;
+  }
+  auto i64 = masked.options();
+  auto i32 = masked.options().dtype(torch::kInt32);
+  auto keys_tmp = torch::empty({nnz}, i64);
+  auto vals_tmp = torch::empty({nnz}, i32);
+  const int64_t hist_elems = (int64_t)custom_radix_sort_hist_elems(nnz);
+  auto hist = torch::empty({hist_elems}, i32);
+  auto scan_sums = torch::empty({4096}, i32);
+  custom_radix_sort_pairs(masked.data_ptr<int64_t>(),
+                          sorted_ids.data_ptr<int64_t>(),
+                          iota.data_ptr<int32_t>(),
+                          sorted_pos.data_ptr<int32_t>(),
+                          keys_tmp.data_ptr<int64_t>(),
+                          vals_tmp.data_ptr<int32_t>(),
+                          hist.data_ptr<int32_t>(),
+                          scan_sums.data_ptr<int32_t>(), nnz, end_bit, stream);
+}
+
 torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
                                  torch::Tensor row_splits, bool mean) {
   CHECK_CUDA(params); CHECK_CUDA(values); CHECK_CUDA(row_splits);
@@ -98,18 +144,11 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
   // 3. radix sort (ids, position) — end_bit covers [0, vocab] inclusive.
   auto sorted_ids = torch::empty({nnz}, i64);
   auto sorted_pos = torch::empty({nnz}, i32);
+  sort_pairs_dispatch(masked, iota, sorted_ids, sorted_pos, nnz,
+                      log2_ceil(vocab + 1), stream);
   size_t temp_bytes = csr_backward_temp_bytes(nnz, vocab);
   auto temp = torch::empty({(int64_t)temp_bytes},
                            f32.dtype(torch::kUInt8));
-  const int end_bit = log2_ceil(vocab + 1);
-  auto err = run_sort_pairs(temp.data_ptr(), temp_bytes,
-                            masked.data_ptr<int64_t>(),
-                            sorted_ids.data_ptr<int64_t>(),
-                            iota.data_ptr<int32_t>(),
-                            sorted_pos.data_ptr<int32_t>(), nnz, end_bit,
-                            stream);
-  TORCH_CHECK(err == hipSuccess, "radix_sort_pairs failed: ",
-              hipGetErrorString(err));
 
   // 4. permute row ids (+ weights) into sorted order, widened to i64.
   auto srow = torch::empty({nnz}, i64);
@@ -128,7 +167,7 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
   auto pos = torch::empty({nnz}, i32);
   launch_mark_heads(sorted_ids.data_ptr<int64_t>(), nnz, vocab,
                     head.data_ptr<int32_t>(), stream);
-  err = run_inclusive_scan_i32(temp.data_ptr(), temp_bytes,
+  auto err = run_inclusive_scan_i32(temp.data_ptr(), temp_bytes,
                                head.data_ptr<int32_t>(),
                                pos.data_ptr<int32_t>(), nnz, stream);
   TORCH_CHECK(err == hipSuccess, "inclusive_scan failed");
@@ -264,15 +303,10 @@ void csr_fused_sgd_apply(torch::Tensor weight, torch::Tensor values,
                         row_ids.data_ptr<int32_t>(), w_ptr, mean, stream);
   auto sorted_ids = torch::empty({nnz}, i64);
   auto sorted_pos = torch::empty({nnz}, i32);
+  sort_pairs_dispatch(masked, iota, sorted_ids, sorted_pos, nnz,
+                      log2_ceil(vocab + 1), stream);
   size_t temp_bytes = csr_backward_temp_bytes(nnz, vocab);
   auto temp = torch::empty({(int64_t)temp_bytes}, f32.dtype(torch::kUInt8));
-  auto err = run_sort_pairs(temp.data_ptr(), temp_bytes,
-                            masked.data_ptr<int64_t>(),
-                            sorted_ids.data_ptr<int64_t>(),
-                            iota.data_ptr<int32_t>(),
-                            sorted_pos.data_ptr<int32_t>(), nnz,
-                            log2_ceil(vocab + 1), stream);
-  TORCH_CHECK(err == hipSuccess, "radix_sort_pairs failed");
   auto srow = torch::empty({nnz}, i64);
   torch::Tensor sw;
   float* sw_ptr = nullptr;
@@ -287,7 +321,7 @@ void csr_fused_sgd_apply(torch::Tensor weight, torch::Tensor values,
   auto pos = torch::empty({nnz}, i32);
   launch_mark_heads(sorted_ids.data_ptr<int64_t>(), nnz, vocab,
                     head.data_ptr<int32_t>(), stream);
-  err = run_inclusive_scan_i32(temp.data_ptr(), temp_bytes,
+  auto err = run_inclusive_scan_i32(temp.data_ptr(), temp_bytes,
                                head.data_ptr<int32_t>(),
                                pos.data_ptr<int32_t>(), nnz, stream);
   TORCH_CHECK(err == hipSuccess, "inclusive_scan failed");

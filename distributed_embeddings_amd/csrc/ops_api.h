@@ -89,3 +89,10 @@ void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,
 void launch_dot_interact_bwd(const void* gout, const void* feats, void* gfeats,
                              int64_t B, int F, int D, int out_w, int tri_n,
                              hipStream_t stream);
+
+void custom_radix_sort_pairs(const int64_t* keys_in, int64_t* keys_out,
+                             const int32_t* vals_in, int32_t* vals_out,
+                             int64_t* keys_tmp, int32_t* vals_tmp,
+                             int32_t* hist, int32_t* scan_sums, int64_t n,
+                             int end_bit, hipStream_t stream);
+size_t custom_radix_sort_hist_elems(int64_t n);

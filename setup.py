@@ -19,6 +19,7 @@ ext = CUDAExtension(
         "distributed_embeddings_amd/csrc/bindings.cpp",
         "distributed_embeddings_amd/csrc/embedding_ops.hip",
         "distributed_embeddings_amd/csrc/dot_interact.hip",
+        "distributed_embeddings_amd/csrc/radix_sort.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
