@@ -247,3 +247,48 @@ def test_world1_passthrough_module(seed):
     outs = model([x0, x1])
     assert torch.equal(outs[0], tables[0].weight[x0])
     assert torch.allclose(outs[1], tables[1].weight[x1].sum(1), atol=1e-6)
+
+
+def _large_testcase_worker(rank, world):
+    """Randomized many-table stress (reference large_testcase analog:
+    dist_model_parallel_test.py:46-52,513-531): 40 tables spanning 2..1e5
+    rows, mixed widths/combiners, all four modes at once."""
+    import distributed_embeddings_amd as de
+    g = torch.Generator().manual_seed(99)
+    sizes = [int(torch.randint(2, 100000, (1,), generator=g)) for _ in range(40)]
+    widths = [int(torch.randint(1, 5, (1,), generator=g)) * 8 for _ in range(40)]
+    combiners = [None, "sum", "mean"]
+    tables = [de.TableConfig(s, w, combiners[i % 3])
+              for i, (s, w) in enumerate(zip(sizes, widths))]
+    model = de.DistributedEmbedding(
+        tables, strategy="memory_balanced",
+        data_parallel_threshold=1000,
+        row_slice_threshold=50000 * 8,
+        column_slice_threshold=20000 * 8)
+    weights = []
+    gw = torch.Generator().manual_seed(123)
+    for s, w in zip(sizes, widths):
+        weights.append(torch.randn(s, w, generator=gw))
+    model.set_weights([w.numpy() for w in weights])
+    gi = torch.Generator().manual_seed(55)
+    B = world * 4
+    inputs, refs = [], []
+    for t, (s, w) in enumerate(zip(sizes, widths)):
+        c = tables[t].combiner
+        if c is None:
+            ids = torch.randint(0, s, (B,), generator=gi)
+            ref = weights[t][ids]
+        else:
+            ids = torch.randint(0, s, (B, 3), generator=gi)
+            ref = weights[t][ids].sum(1) if c == "sum" else weights[t][ids].mean(1)
+        inputs.append(ids)
+        refs.append(ref)
+    sl = slice(rank * 4, (rank + 1) * 4)
+    outs = model([x[sl] for x in inputs])
+    errs = [float((o - r[sl]).abs().max()) for o, r in zip(outs, refs)]
+    return max(errs)
+
+
+def test_large_testcase_world2():
+    results = run_distributed(_large_testcase_worker, world=2)
+    assert max(results) < 1e-4
