@@ -80,7 +80,23 @@ __global__ void csr_fwd_narrow(const float* __restrict__ params,
       continue;
     }
     float acc = 0.f;
-    for (int64_t k = s; k < e; ++k) {
+    int64_t k = s;
+    // 4-way unroll: 4 independent idx->row load chains in flight
+    for (; k + 4 <= e; k += 4) {
+      const int64_t i0 = values[k], i1 = values[k + 1];
+      const int64_t i2 = values[k + 2], i3 = values[k + 3];
+      float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+      if (i0 >= 0 && i0 < vocab)
+        a0 = (HAS_W ? per_id_w[k] : 1.f) * params[i0 * width + tl];
+      if (i1 >= 0 && i1 < vocab)
+        a1 = (HAS_W ? per_id_w[k + 1] : 1.f) * params[i1 * width + tl];
+      if (i2 >= 0 && i2 < vocab)
+        a2 = (HAS_W ? per_id_w[k + 2] : 1.f) * params[i2 * width + tl];
+      if (i3 >= 0 && i3 < vocab)
+        a3 = (HAS_W ? per_id_w[k + 3] : 1.f) * params[i3 * width + tl];
+      acc += (a0 + a1) + (a2 + a3);
+    }
+    for (; k < e; ++k) {
       const int64_t idx = values[k];
       if (idx < 0 || idx >= vocab) continue;
       const float w = HAS_W ? per_id_w[k] : 1.f;
@@ -191,21 +207,32 @@ __global__ void csr_fwd_long(const float* __restrict__ params,
     if (k0 >= e) continue;
     const float inv = (MEAN && e > s) ? 1.f / (float)(e - s) : 1.f;
     if constexpr (TILE > 0) {
-      // narrow tiling: lane tl covers column tl; only lanes < width active
-      const int tl = lane % (TILE > 0 ? TILE : 1);
-      const int sub = lane / (TILE > 0 ? TILE : 1);
-      if (sub != 0 || tl >= width) continue;
+      // narrow tiling: all 64 lanes active — each of the 64/TILE sub-tiles
+      // reduces a strided share of the chunk, then cross-sub shuffles fold
+      // the partials into sub 0 before one atomic per column.
+      constexpr int T = TILE > 0 ? TILE : 1;
+      constexpr int NSUB = WAVE / T;
+      const int tl = lane % T;
+      const int sub = lane / T;
       float acc = 0.f;
-      for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
-        const int64_t ke = min(ks + (int64_t)LONG_T, e);
-        for (int64_t k = ks; k < ke; ++k) {
-          const int64_t idx = values[k];
-          if (idx < 0 || idx >= vocab) continue;
-          const float w = HAS_W ? per_id_w[k] : 1.f;
-          acc += w * params[idx * width + tl];
+      if (tl < width) {
+        for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
+          const int64_t ke = min(ks + (int64_t)LONG_T, e);
+          for (int64_t k = ks + sub; k < ke; k += NSUB) {
+            const int64_t idx = values[k];
+            if (idx < 0 || idx >= vocab) continue;
+            const float w = HAS_W ? per_id_w[k] : 1.f;
+            acc += w * params[idx * width + tl];
+          }
         }
       }
-      atomicAdd(&out[row * width + tl], acc * inv);
+#pragma unroll
+      for (int off = WAVE / 2; off >= T; off >>= 1) {
+        acc += __shfl_down(acc, off);
+      }
+      if (sub == 0 && tl < width) {
+        atomicAdd(&out[row * width + tl], acc * inv);
+      }
     } else {
       constexpr int V = VEC > 0 ? VEC : 1;
       constexpr int CH = WAVE * V;

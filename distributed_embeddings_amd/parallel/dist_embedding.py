@@ -489,13 +489,36 @@ class DistributedEmbedding(nn.Module):
                     row_offs.append((off, nrows))
                     metas.append((j, nrows))
                 allvals = torch.cat(val_parts)
-                all_lengths = torch.cat(len_parts)
-                row_off_vec = self._row_offset_vector(gi, row_offs, device)
-                if row_off_vec is not None:
-                    allvals = allvals + torch.repeat_interleave(row_off_vec, all_lengths)
-                allsplits = torch.zeros(all_lengths.numel() + 1, dtype=torch.long,
-                                        device=device)
-                torch.cumsum(all_lengths, 0, out=allsplits[1:])
+                all_dense = all(not isinstance(pair_ids[j], Ragged) for j in pair_js)
+                if all_dense:
+                    # static shapes: per-element offsets + splits fully cached
+                    espec = []
+                    for j in pair_js:
+                        ids = pair_ids[j]
+                        h = ids.shape[1] if ids.dim() > 1 else 1
+                        espec.append((self._pair_row_offset[j], ids.shape[0] * h))
+                    off_vec = self._offset_vector(gi, espec, device)
+                    if off_vec is not None:
+                        allvals = allvals + off_vec
+                    skey = ("splits", gi, tuple(espec), str(device))
+                    cache = getattr(self, "_off_cache", None) or {}
+                    self._off_cache = cache
+                    if skey not in cache:
+                        all_lengths = torch.cat(len_parts)
+                        sp = torch.zeros(all_lengths.numel() + 1, dtype=torch.long,
+                                         device=device)
+                        torch.cumsum(all_lengths, 0, out=sp[1:])
+                        cache[skey] = sp
+                    allsplits = cache[skey]
+                else:
+                    all_lengths = torch.cat(len_parts)
+                    row_off_vec = self._row_offset_vector(gi, row_offs, device)
+                    if row_off_vec is not None:
+                        allvals = allvals + torch.repeat_interleave(row_off_vec,
+                                                                    all_lengths)
+                    allsplits = torch.zeros(all_lengths.numel() + 1, dtype=torch.long,
+                                            device=device)
+                    torch.cumsum(all_lengths, 0, out=allsplits[1:])
                 if offload:
                     out = embedding_lookup(layer.weight,
                                            Ragged(allvals.cpu(), allsplits.cpu()),
