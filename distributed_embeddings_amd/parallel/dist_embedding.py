@@ -37,6 +37,30 @@ from . import comm
 from .strategy import ConcatGroup, DistEmbeddingStrategy, TableConfig
 
 
+
+def _cat_or_view(parts):
+    """torch.cat that returns a zero-copy flat view when the parts are
+    already adjacent slices of ONE storage (the world==1 / post-a2a layout)."""
+    if len(parts) == 1:
+        return parts[0].reshape(-1)
+    base = parts[0]
+    try:
+        storage = base.untyped_storage()
+        sptr = storage.data_ptr()
+        off = base.storage_offset()
+        total = 0
+        for p in parts:
+            if (p.untyped_storage().data_ptr() != sptr
+                    or not p.is_contiguous()
+                    or p.storage_offset() != off + total):
+                raise StopIteration
+            total += p.numel()
+        return torch.empty(0, dtype=base.dtype, device=base.device).set_(
+            storage, off, (total,), (1,))
+    except StopIteration:
+        return torch.cat([q.reshape(-1) for q in parts])
+
+
 def _layer_to_config(layer) -> TableConfig:
     if isinstance(layer, Embedding):
         return TableConfig(layer.input_dim, layer.output_dim, layer.combiner)
@@ -443,7 +467,7 @@ class DistributedEmbedding(nn.Module):
                 # hotness-1 CSR gather: one cat + one cached-offset add for the
                 # whole group (instead of one add per pair per step).
                 metas = [(j, pair_ids[j].shape, pair_ids[j].numel()) for j in pair_js]
-                allids = torch.cat([pair_ids[j].reshape(-1) for j in pair_js])
+                allids = _cat_or_view([pair_ids[j] for j in pair_js])
                 off_vec = self._offset_vector(gi, [(self._pair_row_offset[j], n)
                                                    for j, _, n in metas],
                                               allids.device)
@@ -468,28 +492,36 @@ class DistributedEmbedding(nn.Module):
             else:
                 # One CSR batch over all pairs of this group; splits merged on
                 # device (no host syncs even for ragged inputs).
-                val_parts, len_parts, metas = [], [], []
-                row_offs = []
+                val_parts, metas, row_offs = [], [], []
                 device = None
+                all_dense = all(not isinstance(pair_ids[j], Ragged) for j in pair_js)
                 for j in pair_js:
                     ids = pair_ids[j]
                     off = self._pair_row_offset[j]
                     if isinstance(ids, Ragged):
                         val_parts.append(ids.values)
-                        len_parts.append(ids.row_lengths())
                         nrows = ids.nrows
                         device = ids.values.device
                     else:
-                        h = ids.shape[1] if ids.dim() > 1 else 1
                         nrows = ids.shape[0]
                         device = ids.device
-                        val_parts.append(ids.reshape(-1))
-                        len_parts.append(torch.full((nrows,), h, dtype=torch.long,
-                                                    device=device))
+                        val_parts.append(ids)
                     row_offs.append((off, nrows))
                     metas.append((j, nrows))
-                allvals = torch.cat(val_parts)
-                all_dense = all(not isinstance(pair_ids[j], Ragged) for j in pair_js)
+
+                def make_len_parts():
+                    lp = []
+                    for j in pair_js:
+                        ids = pair_ids[j]
+                        if isinstance(ids, Ragged):
+                            lp.append(ids.row_lengths())
+                        else:
+                            h = ids.shape[1] if ids.dim() > 1 else 1
+                            lp.append(torch.full((ids.shape[0],), h,
+                                                 dtype=torch.long, device=device))
+                    return lp
+
+                allvals = _cat_or_view(val_parts)
                 if all_dense:
                     # static shapes: per-element offsets + splits fully cached
                     espec = []
@@ -504,14 +536,14 @@ class DistributedEmbedding(nn.Module):
                     cache = getattr(self, "_off_cache", None) or {}
                     self._off_cache = cache
                     if skey not in cache:
-                        all_lengths = torch.cat(len_parts)
+                        all_lengths = torch.cat(make_len_parts())
                         sp = torch.zeros(all_lengths.numel() + 1, dtype=torch.long,
                                          device=device)
                         torch.cumsum(all_lengths, 0, out=sp[1:])
                         cache[skey] = sp
                     allsplits = cache[skey]
                 else:
-                    all_lengths = torch.cat(len_parts)
+                    all_lengths = torch.cat(make_len_parts())
                     row_off_vec = self._row_offset_vector(gi, row_offs, device)
                     if row_off_vec is not None:
                         allvals = allvals + torch.repeat_interleave(row_off_vec,
