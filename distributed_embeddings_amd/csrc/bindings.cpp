@@ -262,9 +262,10 @@ void sparse_row_update(torch::Tensor weight, torch::Tensor state,
                            current_stream());
 }
 
-void csr_fused_sgd_apply(torch::Tensor weight, torch::Tensor values,
-                         torch::Tensor row_splits, torch::Tensor grad_out,
-                         torch::Tensor lr, bool mean) {
+void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
+                               torch::Tensor values, torch::Tensor row_splits,
+                               torch::Tensor grad_out, torch::Tensor lr,
+                               bool mean, bool adagrad, double eps) {
   // All-device sorted pipeline (no host sync; hipGraph-capturable):
   // sort (id, pos) -> permute row-ids -> head-flag unique -> segment pad ->
   // one direct update per unique row (long segments chunked with one atomic
@@ -341,13 +342,27 @@ void csr_fused_sgd_apply(torch::Tensor weight, torch::Tensor values,
                          stream);
   auto long_rows = torch::empty({nnz}, i64);
   auto long_count = torch::empty({1}, i32);
-  launch_sorted_sgd_update(weight.data_ptr<float>(),
-                           sorted_ids.data_ptr<int64_t>(),
-                           seg_tmp.data_ptr<int64_t>(), srow.data_ptr<int64_t>(),
-                           sw_ptr, grad_out.data_ptr<float>(),
-                           lr.data_ptr<float>(), nnz, width,
-                           long_rows.data_ptr<int64_t>(),
-                           long_count.data_ptr<int32_t>(), stream);
+  float* state_ptr = nullptr;
+  torch::Tensor scratch;
+  float* scratch_ptr = nullptr;
+  int64_t scratch_rows = 0;
+  if (adagrad) {
+    CHECK_CUDA(state); CHECK_CONTIG(state);
+    TORCH_CHECK(state.sizes() == weight.sizes(), "adagrad state shape mismatch");
+    state_ptr = state.data_ptr<float>();
+    scratch_rows = nnz / (128 + 1) + 1;  // max possible long segments
+    scratch = torch::empty({scratch_rows, (int64_t)width}, f32);
+    scratch_ptr = scratch.data_ptr<float>();
+  }
+  launch_sorted_optimizer_update(weight.data_ptr<float>(), state_ptr,
+                                 (float)eps, sorted_ids.data_ptr<int64_t>(),
+                                 seg_tmp.data_ptr<int64_t>(),
+                                 srow.data_ptr<int64_t>(), sw_ptr,
+                                 grad_out.data_ptr<float>(),
+                                 lr.data_ptr<float>(), nnz, width,
+                                 long_rows.data_ptr<int64_t>(),
+                                 long_count.data_ptr<int32_t>(), scratch_ptr,
+                                 scratch_rows, adagrad, stream);
 }
 
 torch::Tensor dot_interact_fwd(torch::Tensor feats, int64_t out_w) {
@@ -392,8 +407,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "open-addressing hash vocab build + lookup (gfx950)");
   m.def("sparse_row_update", &sparse_row_update,
         "fused sparse SGD/Adagrad row update (gfx950)");
-  m.def("csr_fused_sgd_apply", &csr_fused_sgd_apply,
-        "in-backward fused SGD scatter update (gfx950)");
+  m.def("csr_fused_optimizer_apply", &csr_fused_optimizer_apply,
+        "in-backward fused SGD/Adagrad update (gfx950)");
   m.def("dot_interact_fwd", &dot_interact_fwd,
         "fused DLRM pairwise-dot interaction forward (MFMA bf16, gfx950)");
   m.def("dot_interact_bwd", &dot_interact_bwd,

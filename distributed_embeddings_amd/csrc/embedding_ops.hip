@@ -51,7 +51,7 @@ static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // This is the wave64 answer to the reference's blockDim.y reduction
 // splitting + round-robin step counter (embedding_lookup_kernels.cu:195-226).
 #define LONG_T 128
-#define MAX_CHUNKS 32
+#define MAX_CHUNKS 128
 
 // Narrow kernel A: width <= 64.  TILE = pow2 >= width; 64/TILE rows per wave.
 template <int TILE, bool MEAN, bool HAS_W>
@@ -218,7 +218,22 @@ __global__ void csr_fwd_long(const float* __restrict__ params,
       if (tl < width) {
         for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
           const int64_t ke = min(ks + (int64_t)LONG_T, e);
-          for (int64_t k = ks + sub; k < ke; k += NSUB) {
+          int64_t k = ks + sub;
+          for (; k + 3 * NSUB < ke; k += 4 * NSUB) {
+            const int64_t i0 = values[k], i1 = values[k + NSUB];
+            const int64_t i2 = values[k + 2 * NSUB], i3 = values[k + 3 * NSUB];
+            float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+            if (i0 >= 0 && i0 < vocab)
+              a0 = (HAS_W ? per_id_w[k] : 1.f) * params[i0 * width + tl];
+            if (i1 >= 0 && i1 < vocab)
+              a1 = (HAS_W ? per_id_w[k + NSUB] : 1.f) * params[i1 * width + tl];
+            if (i2 >= 0 && i2 < vocab)
+              a2 = (HAS_W ? per_id_w[k + 2 * NSUB] : 1.f) * params[i2 * width + tl];
+            if (i3 >= 0 && i3 < vocab)
+              a3 = (HAS_W ? per_id_w[k + 3 * NSUB] : 1.f) * params[i3 * width + tl];
+            acc += (a0 + a1) + (a2 + a3);
+          }
+          for (; k < ke; k += NSUB) {
             const int64_t idx = values[k];
             if (idx < 0 || idx >= vocab) continue;
             const float w = HAS_W ? per_id_w[k] : 1.f;
@@ -663,8 +678,9 @@ __global__ void pad_seg_offsets(int64_t* __restrict__ seg, int64_t n,
   if (i >= *num_unique) seg[i] = bounds[1];  // first OOB position
 }
 
-template <int VEC, bool HAS_W>
+template <int VEC, bool HAS_W, bool ADAGRAD>
 __global__ void sorted_sgd_update(float* __restrict__ weight,
+                                  float* __restrict__ state, float eps,
                                   const int64_t* __restrict__ sorted_ids,
                                   const int64_t* __restrict__ seg,
                                   const int64_t* __restrict__ srow,
@@ -706,13 +722,27 @@ __global__ void sorted_sgd_update(float* __restrict__ weight,
         }
       }
       float* wp = weight + uid * (int64_t)width + col0;
+      if (ADAGRAD) {
+        float* sp = state + uid * (int64_t)width + col0;
 #pragma unroll
-      for (int v = 0; v < VEC; ++v)
-        if (col0 + v < width) wp[v] -= lr * acc[v];
+        for (int v = 0; v < VEC; ++v) {
+          if (col0 + v < width) {
+            const float g = acc[v];
+            const float s = sp[v] + g * g;
+            sp[v] = s;
+            wp[v] -= lr * g / (sqrtf(s) + eps);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int v = 0; v < VEC; ++v)
+          if (col0 + v < width) wp[v] -= lr * acc[v];
+      }
     }
   }
 }
 
+// SGD long segments: atomic partials straight into the weight (linear).
 template <int VEC, bool HAS_W>
 __global__ void sorted_sgd_update_long(float* __restrict__ weight,
                                        const int64_t* __restrict__ sorted_ids,
@@ -769,6 +799,85 @@ __global__ void sorted_sgd_update_long(float* __restrict__ weight,
   }
 }
 
+// Adagrad long segments: the update is non-linear, so chunk partials
+// accumulate into a scratch row per long segment (atomicAdd), then a
+// finalize kernel applies state += g^2; w -= lr*g/(sqrt(state)+eps).
+template <int VEC, bool HAS_W>
+__global__ void sorted_adagrad_long_accum(const int64_t* __restrict__ seg,
+                                          const int64_t* __restrict__ srow,
+                                          const float* __restrict__ sw,
+                                          const float* __restrict__ grad_out,
+                                          int width,
+                                          const int64_t* __restrict__ long_rows,
+                                          const int32_t* __restrict__ long_count,
+                                          float* __restrict__ scratch) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const int64_t n_long = *long_count;
+  const int64_t n_items = n_long * MAX_CHUNKS;
+  for (int64_t item = wave_id; item < n_items; item += n_waves) {
+    const int64_t li = item / MAX_CHUNKS;
+    const int64_t chunk = item % MAX_CHUNKS;
+    const int64_t r = long_rows[li];
+    const int64_t s = seg[r], e = seg[r + 1];
+    const int64_t k0 = s + chunk * LONG_T;
+    if (k0 >= e) continue;
+    constexpr int V = VEC > 0 ? VEC : 1;
+    constexpr int CH = WAVE * V;
+    for (int cbase = 0; cbase < width; cbase += CH) {
+      float acc[V];
+#pragma unroll
+      for (int v = 0; v < V; ++v) acc[v] = 0.f;
+      const int col0 = cbase + lane * V;
+      for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
+        const int64_t ke = min(ks + (int64_t)LONG_T, e);
+        for (int64_t k = ks; k < ke; ++k) {
+          const float w = HAS_W ? sw[k] : 1.f;
+          const float* gp = grad_out + srow[k] * (int64_t)width + col0;
+          if (V == 4 && col0 + 4 <= width) {
+            const float4 g4 = *reinterpret_cast<const float4*>(gp);
+            acc[0] += w * g4.x; acc[1] += w * g4.y;
+            acc[2] += w * g4.z; acc[3] += w * g4.w;
+          } else {
+#pragma unroll
+            for (int v = 0; v < V; ++v)
+              if (col0 + v < width) acc[v] += w * gp[v];
+          }
+        }
+      }
+      float* sp = scratch + li * (int64_t)width + col0;
+#pragma unroll
+      for (int v = 0; v < V; ++v)
+        if (col0 + v < width) atomicAdd(&sp[v], acc[v]);
+    }
+  }
+}
+
+__global__ void sorted_adagrad_long_finalize(
+    float* __restrict__ weight, float* __restrict__ state, float eps,
+    const int64_t* __restrict__ sorted_ids, const int64_t* __restrict__ seg,
+    const float* __restrict__ lr_ptr, int width,
+    const int64_t* __restrict__ long_rows,
+    const int32_t* __restrict__ long_count, const float* __restrict__ scratch) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const float lr = *lr_ptr;
+  const int64_t n_long = *long_count;
+  for (int64_t li = wave_id; li < n_long; li += n_waves) {
+    const int64_t r = long_rows[li];
+    const int64_t uid = sorted_ids[seg[r]];
+    for (int c = lane; c < width; c += WAVE) {
+      const float g = scratch[li * (int64_t)width + c];
+      const int64_t o = uid * (int64_t)width + c;
+      const float s = state[o] + g * g;
+      state[o] = s;
+      weight[o] -= lr * g / (sqrtf(s) + eps);
+    }
+  }
+}
+
 void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
                             const int64_t* bounds, hipStream_t stream) {
   const int block = 256;
@@ -776,24 +885,45 @@ void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
                      dim3(block), 0, stream, seg, n, num_unique, bounds);
 }
 
-void launch_sorted_sgd_update(float* weight, const int64_t* sorted_ids,
-                              const int64_t* seg, const int64_t* srow,
-                              const float* sw, const float* grad_out,
-                              const float* lr, int64_t max_segs, int width,
-                              int64_t* long_rows, int32_t* long_count,
-                              hipStream_t stream) {
+void launch_sorted_optimizer_update(float* weight, float* state, float eps,
+                                    const int64_t* sorted_ids,
+                                    const int64_t* seg, const int64_t* srow,
+                                    const float* sw, const float* grad_out,
+                                    const float* lr, int64_t max_segs,
+                                    int width, int64_t* long_rows,
+                                    int32_t* long_count, float* long_scratch,
+                                    int64_t scratch_rows, bool adagrad,
+                                    hipStream_t stream) {
   const int block = 256;
   hipMemsetAsync(long_count, 0, sizeof(int32_t), stream);
+  if (adagrad) {
+    hipMemsetAsync(long_scratch, 0,
+                   sizeof(float) * scratch_rows * (int64_t)width, stream);
+  }
   const int grid = pick_grid(max_segs, block / WAVE);
 #define LU(VEC, HASW)                                                          \
   do {                                                                         \
-    hipLaunchKernelGGL((sorted_sgd_update<VEC, HASW>), dim3(grid),             \
-                       dim3(block), 0, stream, weight, sorted_ids, seg, srow,  \
-                       sw, grad_out, lr, max_segs, width, long_rows,           \
-                       long_count);                                            \
-    hipLaunchKernelGGL((sorted_sgd_update_long<VEC, HASW>), dim3(2048),        \
-                       dim3(block), 0, stream, weight, sorted_ids, seg, srow,  \
-                       sw, grad_out, lr, width, long_rows, long_count);        \
+    if (adagrad) {                                                             \
+      hipLaunchKernelGGL((sorted_sgd_update<VEC, HASW, true>), dim3(grid),     \
+                         dim3(block), 0, stream, weight, state, eps,           \
+                         sorted_ids, seg, srow, sw, grad_out, lr, max_segs,    \
+                         width, long_rows, long_count);                        \
+      hipLaunchKernelGGL((sorted_adagrad_long_accum<VEC, HASW>), dim3(2048),   \
+                         dim3(block), 0, stream, seg, srow, sw, grad_out,      \
+                         width, long_rows, long_count, long_scratch);          \
+      hipLaunchKernelGGL(sorted_adagrad_long_finalize, dim3(256), dim3(block), \
+                         0, stream, weight, state, eps, sorted_ids, seg, lr,   \
+                         width, long_rows, long_count, long_scratch);          \
+    } else {                                                                   \
+      hipLaunchKernelGGL((sorted_sgd_update<VEC, HASW, false>), dim3(grid),    \
+                         dim3(block), 0, stream, weight, nullptr, 0.f,         \
+                         sorted_ids, seg, srow, sw, grad_out, lr, max_segs,    \
+                         width, long_rows, long_count);                        \
+      hipLaunchKernelGGL((sorted_sgd_update_long<VEC, HASW>), dim3(2048),      \
+                         dim3(block), 0, stream, weight, sorted_ids, seg,      \
+                         srow, sw, grad_out, lr, width, long_rows,             \
+                         long_count);                                          \
+    }                                                                          \
   } while (0)
   if (width % 4 == 0) {
     if (sw) LU(4, true); else LU(4, false);

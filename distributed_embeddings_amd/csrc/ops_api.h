@@ -72,12 +72,15 @@ void launch_csr_fused_sgd(float* weight, const int64_t* values,
 void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
                             const int64_t* bounds, hipStream_t stream);
 
-void launch_sorted_sgd_update(float* weight, const int64_t* sorted_ids,
-                              const int64_t* seg, const int64_t* srow,
-                              const float* sw, const float* grad_out,
-                              const float* lr, int64_t max_segs, int width,
-                              int64_t* long_rows, int32_t* long_count,
-                              hipStream_t stream);
+void launch_sorted_optimizer_update(float* weight, float* state, float eps,
+                                    const int64_t* sorted_ids,
+                                    const int64_t* seg, const int64_t* srow,
+                                    const float* sw, const float* grad_out,
+                                    const float* lr, int64_t max_segs,
+                                    int width, int64_t* long_rows,
+                                    int32_t* long_count, float* long_scratch,
+                                    int64_t scratch_rows, bool adagrad,
+                                    hipStream_t stream);
 
 void launch_sparse_row_update(float* weight, float* state, const int64_t* ids,
                               const float* grad, int64_t num_rows, int width,

@@ -15,7 +15,8 @@ from typing import Callable, Optional, Union
 import torch
 from torch import nn
 
-from ..ops.embedding_lookup import Ragged, csr_lookup_fused_sgd, embedding_lookup
+from ..ops.embedding_lookup import (Ragged, csr_lookup_fused_optimizer,
+                                    csr_lookup_fused_sgd, embedding_lookup)
 
 
 def _default_init(weight: torch.Tensor) -> None:
@@ -84,13 +85,27 @@ class Embedding(nn.Module):
         return f"input_dim={self.input_dim}, output_dim={self.output_dim}, combiner={self.combiner}"
 
     def enable_fused_sgd(self, lr: float):
-        """In-backward SGD: the lookup's backward scatter-applies
-        ``w[id] -= lr * grad`` directly (no grad tensor, no host sync; the
-        step becomes hipGraph-capturable).  Exact for SGD.  The optimizer
-        must not also update this weight (its ``.grad`` stays ``None``)."""
+        """In-backward SGD (see enable_fused_optimizer)."""
+        return self.enable_fused_optimizer("sgd", lr)
+
+    def enable_fused_optimizer(self, method: str, lr: float, eps: float = 1e-10):
+        """In-backward fused optimizer: the lookup's backward applies the
+        SGD/Adagrad update directly (no grad tensor, no host sync; the step
+        becomes hipGraph-capturable).  The training optimizer must not also
+        update this weight (its ``.grad`` stays ``None``)."""
+        if method not in ("sgd", "adagrad"):
+            raise ValueError(f"unknown fused optimizer {method!r}")
         self.register_buffer("_fused_lr",
                              torch.tensor([float(lr)], dtype=torch.float32,
                                           device=self.weight.device))
+        self._fused_method = method
+        self._fused_eps = float(eps)
+        if method == "adagrad":
+            self.register_buffer("_fused_state", torch.zeros_like(self.weight))
+        else:
+            self.register_buffer("_fused_state",
+                                 torch.empty(0, dtype=torch.float32,
+                                             device=self.weight.device))
         return self
 
     def set_fused_lr(self, lr: float):
@@ -103,8 +118,10 @@ class Embedding(nn.Module):
                    combiner: str) -> torch.Tensor:
         """CSR lookup through this layer (fused-SGD aware)."""
         if getattr(self, "_fused_lr", None) is not None and self.training:
-            return csr_lookup_fused_sgd(self.weight, values, row_splits,
-                                        combiner, self._fused_lr)
+            return csr_lookup_fused_optimizer(
+                self.weight, values, row_splits, combiner, self._fused_lr,
+                self._fused_state, self._fused_method == "adagrad",
+                self._fused_eps)
         return embedding_lookup(self.weight, Ragged(values, row_splits), combiner)
 
     def get_config(self) -> dict:

@@ -161,20 +161,25 @@ def _csr_lookup_backward_ref(grad_out, values, row_splits, vocab, combiner):
     return unique_ids, unique_grad
 
 
-class _CsrLookupFusedSGD(torch.autograd.Function):
-    """CSR lookup whose backward applies the SGD update in place.
+class _CsrLookupFusedOptimizer(torch.autograd.Function):
+    """CSR lookup whose backward applies the optimizer update in place.
 
-    Exact for SGD (the update is linear in the grad, order-free up to fp
-    rounding).  No gradient tensor is ever materialized for the table, no
-    sort/unique pipeline runs, and there is no host sync — the whole training
-    step becomes hipGraph-capturable.  ``lr`` is a 1-element fp32 device
-    tensor so schedules can update it without touching the graph.
+    SGD is linear in the grad (order-free up to fp rounding); Adagrad uses
+    the same all-device sorted-segment pipeline so per-unique-row sums are
+    exact.  No gradient tensor is materialized for the table, there is no
+    host sync, and the whole training step becomes hipGraph-capturable.
+    ``lr`` is a 1-element fp32 device tensor so schedules can update it
+    without touching the graph; ``state`` is the Adagrad accumulator (empty
+    tensor for SGD).
     """
 
     @staticmethod
-    def forward(ctx, weight, values, row_splits, combiner, lr):
-        ctx.save_for_backward(weight, values, row_splits, lr)
+    def forward(ctx, weight, values, row_splits, combiner, lr, state, adagrad,
+                eps):
+        ctx.save_for_backward(weight, values, row_splits, lr, state)
         ctx.combiner = combiner
+        ctx.adagrad = adagrad
+        ctx.eps = eps
         if weight.is_cuda:
             return _backend.ops().csr_lookup_forward(weight, values, row_splits,
                                                      combiner == "mean")
@@ -182,22 +187,36 @@ class _CsrLookupFusedSGD(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_out):
-        weight, values, row_splits, lr = ctx.saved_tensors
+        weight, values, row_splits, lr, state = ctx.saved_tensors
         grad_out = grad_out.contiguous()
         with torch.no_grad():
             if weight.is_cuda:
-                _backend.ops().csr_fused_sgd_apply(weight, values, row_splits,
-                                                   grad_out,
-                                                   lr, ctx.combiner == "mean")
+                _backend.ops().csr_fused_optimizer_apply(
+                    weight, state, values, row_splits, grad_out, lr,
+                    ctx.combiner == "mean", ctx.adagrad, ctx.eps)
             else:
                 unique_ids, unique_grad = _csr_lookup_backward_ref(
                     grad_out, values, row_splits, weight.shape[0], ctx.combiner)
-                weight.index_add_(0, unique_ids, unique_grad * (-lr.item()))
-        return None, None, None, None, None
+                lr_v = lr.item()
+                if ctx.adagrad:
+                    state.index_add_(0, unique_ids, unique_grad * unique_grad)
+                    denom = state.index_select(0, unique_ids).sqrt_().add_(ctx.eps)
+                    weight.index_add_(0, unique_ids, -lr_v * unique_grad / denom)
+                else:
+                    weight.index_add_(0, unique_ids, unique_grad * (-lr_v))
+        return None, None, None, None, None, None, None, None
 
 
 def csr_lookup_fused_sgd(weight, values, row_splits, combiner, lr):
-    return _CsrLookupFusedSGD.apply(weight, values, row_splits, combiner, lr)
+    empty = torch.empty(0, dtype=torch.float32, device=weight.device)
+    return _CsrLookupFusedOptimizer.apply(weight, values, row_splits, combiner,
+                                          lr, empty, False, 0.0)
+
+
+def csr_lookup_fused_optimizer(weight, values, row_splits, combiner, lr, state,
+                               adagrad, eps):
+    return _CsrLookupFusedOptimizer.apply(weight, values, row_splits, combiner,
+                                          lr, state, adagrad, eps)
 
 
 def _dense_fixed_hotness(weight, ids, combiner):

@@ -184,11 +184,14 @@ class DistributedEmbedding(nn.Module):
     # ------------------------------------------------------------------ utils
 
     def enable_fused_sgd(self, lr: float):
-        """Enables in-backward fused SGD on all model-parallel tables (col +
-        row groups).  Data-parallel tables keep sparse grads (they need the
-        allreduce).  See Embedding.enable_fused_sgd."""
+        return self.enable_fused_optimizer("sgd", lr)
+
+    def enable_fused_optimizer(self, method: str, lr: float, eps: float = 1e-10):
+        """Enables in-backward fused SGD/Adagrad on all model-parallel tables
+        (col + row groups).  Data-parallel tables keep sparse grads (they
+        need the allreduce).  See Embedding.enable_fused_optimizer."""
         for lyr in list(self.col_layers) + list(self.row_layers):
-            lyr.enable_fused_sgd(lr)
+            lyr.enable_fused_optimizer(method, lr, eps)
         return self
 
     def set_fused_lr(self, lr: float):

@@ -73,6 +73,7 @@ def main():
 
     opt = de.DistributedOptimizer(SparseEmbeddingOptimizer(
         model.parameters(), lr=0.01, method=args.optimizer), average=False)
+    model.embeddings.enable_fused_optimizer(args.optimizer, 0.01)
     de.broadcast_parameters(model)
     _loss_sum = torch.nn.BCEWithLogitsLoss(reduction="sum")
     loss_fn = lambda lg, lb: _loss_sum(lg, lb) / args.batch_size

@@ -146,3 +146,26 @@ def test_fused_sgd_cpu_matches_explicit(seed):
         o2.step()
         assert torch.allclose(out1, out2, atol=1e-6)
         assert torch.allclose(e1.weight, e2.weight, atol=1e-6), combiner
+
+
+def test_fused_adagrad_cpu_matches_explicit(seed):
+    from distributed_embeddings_amd import Embedding, Ragged, SparseEmbeddingOptimizer
+    w0 = torch.randn(40, 8)
+    lists = [[1, 2, 3], [2], [5, 5, 7], [0]]
+    up = torch.randn(4, 8)
+    e1 = Embedding(40, 8, combiner="sum")
+    e2 = Embedding(40, 8, combiner="sum")
+    with torch.no_grad():
+        e1.weight.copy_(w0); e2.weight.copy_(w0)
+    e1.enable_fused_optimizer("adagrad", 0.1)
+    o2 = SparseEmbeddingOptimizer(e2.parameters(), lr=0.1, method="adagrad")
+    for _ in range(3):
+        o2.zero_grad()
+        out1 = e1(Ragged.from_lists(lists))
+        out1.backward(up)
+        out2 = e2(Ragged.from_lists(lists))
+        out2.backward(up)
+        o2.step()
+    assert e1.weight.grad is None
+    assert torch.allclose(e1.weight, e2.weight, atol=1e-5), \
+        float((e1.weight - e2.weight).abs().max())

@@ -314,3 +314,31 @@ def test_fused_sgd_gpu_matches_explicit():
     # atomic scatter order differs from segmented sum: fp32 roundoff only
     assert torch.allclose(e1.weight, e2.weight, atol=1e-3), \
         float((e1.weight - e2.weight).abs().max())
+
+
+@requires_gpu
+def test_fused_adagrad_gpu_matches_explicit():
+    from distributed_embeddings_amd import Embedding, Ragged, SparseEmbeddingOptimizer
+    torch.manual_seed(31)
+    w0 = torch.randn(500, 32)
+    # heavy dups incl. tiny-vocab-style hot rows (long segments)
+    ids = torch.cat([torch.randint(0, 500, (4000,)),
+                     torch.zeros(2000, dtype=torch.long),
+                     torch.full((2000,), 3)]).cuda()
+    perm = torch.randperm(8000, device="cuda")
+    ids = ids[perm]
+    splits = torch.arange(0, 8001, 4, device="cuda")
+    up = torch.randn(2000, 32, device="cuda")
+    e1 = Embedding(500, 32, combiner="mean").cuda()
+    e2 = Embedding(500, 32, combiner="mean").cuda()
+    with torch.no_grad():
+        e1.weight.copy_(w0); e2.weight.copy_(w0)
+    e1.enable_fused_optimizer("adagrad", 0.1)
+    o2 = SparseEmbeddingOptimizer(e2.parameters(), lr=0.1, method="adagrad")
+    for _ in range(2):
+        o2.zero_grad()
+        e1(Ragged(ids, splits)).backward(up)
+        e2(Ragged(ids, splits)).backward(up)
+        o2.step()
+    assert torch.allclose(e1.weight, e2.weight, atol=1e-3), \
+        float((e1.weight - e2.weight).abs().max())
