@@ -82,15 +82,20 @@ torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
   const int width = (int)params.size(1);
   auto out = torch::empty({num_rows, width}, params.options());
   if (num_rows > 0) {
+    const int64_t nnz_in = values.numel();
     auto long_rows = torch::empty({num_rows}, values.options());
     auto long_count = torch::empty({1}, values.options().dtype(torch::kInt32));
+    auto work_items = torch::empty({nnz_in / 64 + 64}, values.options());
+    auto n_work = torch::empty({1}, values.options().dtype(torch::kInt32));
     launch_csr_lookup_forward(params.data_ptr<float>(),
                               values.data_ptr<int64_t>(),
                               row_splits.data_ptr<int64_t>(), nullptr,
-                              out.data_ptr<float>(), num_rows, values.numel(),
+                              out.data_ptr<float>(), num_rows, nnz_in,
                               vocab, width, mean,
                               long_rows.data_ptr<int64_t>(),
-                              long_count.data_ptr<int32_t>(), current_stream());
+                              long_count.data_ptr<int32_t>(),
+                              work_items.data_ptr<int64_t>(),
+                              n_work.data_ptr<int32_t>(), current_stream());
   }
   return out;
 }
@@ -196,13 +201,17 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
     // 7. segmented sum == forward gather-reduce over grad_out rows.
     auto long_rows = torch::empty({nu}, i64);
     auto long_count = torch::empty({1}, i32);
+    auto work_items = torch::empty({nnz / 64 + 64}, i64);
+    auto n_work = torch::empty({1}, i32);
     launch_csr_lookup_forward(grad_out.data_ptr<float>(),
                               srow.data_ptr<int64_t>(),
                               seg_tmp.data_ptr<int64_t>(), sw_ptr,
                               unique_grad.data_ptr<float>(), nu, nnz,
                               grad_out.size(0), width, /*mean=*/false,
                               long_rows.data_ptr<int64_t>(),
-                              long_count.data_ptr<int32_t>(), stream);
+                              long_count.data_ptr<int32_t>(),
+                              work_items.data_ptr<int64_t>(),
+                              n_work.data_ptr<int32_t>(), stream);
   }
   return {unique_ids, unique_grad};
 }
@@ -342,6 +351,8 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
                          stream);
   auto long_rows = torch::empty({nnz}, i64);
   auto long_count = torch::empty({1}, i32);
+  auto work_items = torch::empty({nnz / 64 + 64}, i64);
+  auto n_work = torch::empty({1}, i32);
   float* state_ptr = nullptr;
   torch::Tensor scratch;
   float* scratch_ptr = nullptr;
@@ -361,7 +372,9 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
                                  grad_out.data_ptr<float>(),
                                  lr.data_ptr<float>(), nnz, width,
                                  long_rows.data_ptr<int64_t>(),
-                                 long_count.data_ptr<int32_t>(), scratch_ptr,
+                                 long_count.data_ptr<int32_t>(),
+                                 work_items.data_ptr<int64_t>(),
+                                 n_work.data_ptr<int32_t>(), scratch_ptr,
                                  scratch_rows, adagrad, stream);
 }
 

@@ -51,7 +51,6 @@ static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 // This is the wave64 answer to the reference's blockDim.y reduction
 // splitting + round-robin step counter (embedding_lookup_kernels.cu:195-226).
 #define LONG_T 128
-#define MAX_CHUNKS 128
 
 // Narrow kernel A: width <= 64.  TILE = pow2 >= width; 64/TILE rows per wave.
 template <int TILE, bool MEAN, bool HAS_W>
@@ -179,6 +178,30 @@ __global__ void csr_fwd_wide(const float* __restrict__ params,
   }
 }
 
+// Builds the exact (long-row, chunk) work list so consumer kernels never
+// probe empty chunk slots.  Item encoding: (li << 24) | chunk.
+__global__ void expand_long_work(const int64_t* __restrict__ long_rows,
+                                 const int32_t* __restrict__ long_count,
+                                 const int64_t* __restrict__ splits,
+                                 int64_t* __restrict__ work_items,
+                                 int32_t* __restrict__ n_work) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const int64_t n_long = *long_count;
+  for (int64_t li = wave_id; li < n_long; li += n_waves) {
+    const int64_t row = long_rows[li];
+    const int64_t len = splits[row + 1] - splits[row];
+    const int64_t chunks = (len + LONG_T - 1) / LONG_T;
+    int base = 0;
+    if (lane == 0) base = atomicAdd(n_work, (int32_t)chunks);
+    base = __shfl(base, 0);
+    for (int64_t c = lane; c < chunks; c += WAVE) {
+      work_items[base + c] = (li << 24) | c;
+    }
+  }
+}
+
 // Kernel B: long rows.  Work item = (long row, LONG_T-chunk); one wave each.
 // NW lanes-per-row tiling matches kernel A (TILE for narrow, full wave for
 // wide).  Partials combine with global atomicAdd (out pre-zeroed by A).
@@ -189,20 +212,19 @@ __global__ void csr_fwd_long(const float* __restrict__ params,
                              const float* __restrict__ per_id_w,
                              float* __restrict__ out, int64_t vocab, int width,
                              const int64_t* __restrict__ long_rows,
-                             const int32_t* __restrict__ long_count) {
+                             const int64_t* __restrict__ work_items,
+                             const int32_t* __restrict__ n_work_ptr) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
-  const int64_t n_long = *long_count;
-  const int64_t n_items = n_long * MAX_CHUNKS;
+  const int64_t n_items = *n_work_ptr;
   for (int64_t item = wave_id; item < n_items; item += n_waves) {
-    const int64_t li = item / MAX_CHUNKS;
-    const int64_t chunk = item % MAX_CHUNKS;
+    const int64_t w_it = work_items[item];
+    const int64_t li = w_it >> 24;
+    const int64_t chunk = w_it & 0xffffff;
     const int64_t row = long_rows[li];
     const int64_t s = splits[row], e = splits[row + 1];
-    // chunk-wave `chunk` strides the whole segment in MAX_CHUNKS*LONG_T hops,
-    // so arbitrarily long segments still use exactly MAX_CHUNKS waves.
     const int64_t k0 = s + chunk * LONG_T;
     if (k0 >= e) continue;
     const float inv = (MEAN && e > s) ? 1.f / (float)(e - s) : 1.f;
@@ -216,7 +238,8 @@ __global__ void csr_fwd_long(const float* __restrict__ params,
       const int sub = lane / T;
       float acc = 0.f;
       if (tl < width) {
-        for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
+        {
+          const int64_t ks = k0;
           const int64_t ke = min(ks + (int64_t)LONG_T, e);
           int64_t k = ks + sub;
           for (; k + 3 * NSUB < ke; k += 4 * NSUB) {
@@ -256,7 +279,8 @@ __global__ void csr_fwd_long(const float* __restrict__ params,
 #pragma unroll
         for (int v = 0; v < V; ++v) acc[v] = 0.f;
         const int col0 = cbase + lane * V;
-        for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
+        {
+          const int64_t ks = k0;
           const int64_t ke = min(ks + (int64_t)LONG_T, e);
           for (int64_t k = ks; k < ke; ++k) {
             const int64_t idx = values[k];
@@ -297,6 +321,7 @@ static void launch_csr_pair(const float* params, const int64_t* values,
                             float* out, int64_t num_rows, int64_t nnz,
                             int64_t vocab, int width, bool mean,
                             int64_t* long_rows, int32_t* long_count,
+                            int64_t* work_items, int32_t* n_work,
                             hipStream_t stream) {
   const int block = 256, bw = block / WAVE;
   const int64_t row_waves = TILE > 0 ? cdiv64(num_rows, WAVE / (TILE > 0 ? TILE : 1)) : num_rows;
@@ -312,6 +337,7 @@ static void launch_csr_pair(const float* params, const int64_t* values,
     if (long_thresh > 8192) long_thresh = 8192;
   }
   hipMemsetAsync(long_count, 0, sizeof(int32_t), stream);
+  hipMemsetAsync(n_work, 0, sizeof(int32_t), stream);
 #define LA(MEAN, HASW)                                                         \
   do {                                                                         \
     if constexpr (TILE > 0)                                                    \
@@ -326,9 +352,14 @@ static void launch_csr_pair(const float* params, const int64_t* values,
                          long_rows, long_count);                               \
   } while (0)
 #define LB(MEAN, HASW)                                                         \
-  hipLaunchKernelGGL((csr_fwd_long<TILE, VEC, MEAN, HASW>), dim3(2048),        \
-                     dim3(block), 0, stream, params, values, splits, per_id_w, \
-                     out, vocab, width, long_rows, long_count)
+  do {                                                                         \
+    hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,    \
+                       long_rows, long_count, splits, work_items, n_work);     \
+    hipLaunchKernelGGL((csr_fwd_long<TILE, VEC, MEAN, HASW>), dim3(2048),      \
+                       dim3(block), 0, stream, params, values, splits,         \
+                       per_id_w, out, vocab, width, long_rows, work_items,     \
+                       n_work);                                                \
+  } while (0)
   if (mean) {
     if (per_id_w) { LA(true, true); LB(true, true); }
     else          { LA(true, false); LB(true, false); }
@@ -345,9 +376,10 @@ void launch_csr_lookup_forward(const float* params, const int64_t* values,
                                float* out, int64_t num_rows, int64_t nnz,
                                int64_t vocab, int width, bool mean,
                                int64_t* long_rows, int32_t* long_count,
+                               int64_t* work_items, int32_t* n_work,
                                hipStream_t stream) {
 #define ARGS params, values, splits, per_id_w, out, num_rows, nnz, vocab, \
-             width, mean, long_rows, long_count, stream
+             width, mean, long_rows, long_count, work_items, n_work, stream
   if (width <= 64) {
     switch (next_pow2(width)) {
       case 1: launch_csr_pair<1, 0>(ARGS); break;
@@ -753,16 +785,17 @@ __global__ void sorted_sgd_update_long(float* __restrict__ weight,
                                        const float* __restrict__ lr_ptr,
                                        int width,
                                        const int64_t* __restrict__ long_rows,
-                                       const int32_t* __restrict__ long_count) {
+                                       const int64_t* __restrict__ work_items,
+                                       const int32_t* __restrict__ n_work_ptr) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
   const float lr = *lr_ptr;
-  const int64_t n_long = *long_count;
-  const int64_t n_items = n_long * MAX_CHUNKS;
+  const int64_t n_items = *n_work_ptr;
   for (int64_t item = wave_id; item < n_items; item += n_waves) {
-    const int64_t li = item / MAX_CHUNKS;
-    const int64_t chunk = item % MAX_CHUNKS;
+    const int64_t w_it = work_items[item];
+    const int64_t li = w_it >> 24;
+    const int64_t chunk = w_it & 0xffffff;
     const int64_t r = long_rows[li];
     const int64_t s = seg[r], e = seg[r + 1];
     const int64_t k0 = s + chunk * LONG_T;
@@ -775,7 +808,8 @@ __global__ void sorted_sgd_update_long(float* __restrict__ weight,
 #pragma unroll
       for (int v = 0; v < V; ++v) acc[v] = 0.f;
       const int col0 = cbase + lane * V;
-      for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
+      {
+        const int64_t ks = k0;
         const int64_t ke = min(ks + (int64_t)LONG_T, e);
         for (int64_t k = ks; k < ke; ++k) {
           const float w = HAS_W ? sw[k] : 1.f;
@@ -809,16 +843,17 @@ __global__ void sorted_adagrad_long_accum(const int64_t* __restrict__ seg,
                                           const float* __restrict__ grad_out,
                                           int width,
                                           const int64_t* __restrict__ long_rows,
-                                          const int32_t* __restrict__ long_count,
+                                          const int64_t* __restrict__ work_items,
+                                          const int32_t* __restrict__ n_work_ptr,
                                           float* __restrict__ scratch) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
-  const int64_t n_long = *long_count;
-  const int64_t n_items = n_long * MAX_CHUNKS;
+  const int64_t n_items = *n_work_ptr;
   for (int64_t item = wave_id; item < n_items; item += n_waves) {
-    const int64_t li = item / MAX_CHUNKS;
-    const int64_t chunk = item % MAX_CHUNKS;
+    const int64_t w_it = work_items[item];
+    const int64_t li = w_it >> 24;
+    const int64_t chunk = w_it & 0xffffff;
     const int64_t r = long_rows[li];
     const int64_t s = seg[r], e = seg[r + 1];
     const int64_t k0 = s + chunk * LONG_T;
@@ -830,7 +865,8 @@ __global__ void sorted_adagrad_long_accum(const int64_t* __restrict__ seg,
 #pragma unroll
       for (int v = 0; v < V; ++v) acc[v] = 0.f;
       const int col0 = cbase + lane * V;
-      for (int64_t ks = k0; ks < e; ks += (int64_t)MAX_CHUNKS * LONG_T) {
+      {
+        const int64_t ks = k0;
         const int64_t ke = min(ks + (int64_t)LONG_T, e);
         for (int64_t k = ks; k < ke; ++k) {
           const float w = HAS_W ? sw[k] : 1.f;
@@ -891,11 +927,13 @@ void launch_sorted_optimizer_update(float* weight, float* state, float eps,
                                     const float* sw, const float* grad_out,
                                     const float* lr, int64_t max_segs,
                                     int width, int64_t* long_rows,
-                                    int32_t* long_count, float* long_scratch,
+                                    int32_t* long_count, int64_t* work_items,
+                                    int32_t* n_work, float* long_scratch,
                                     int64_t scratch_rows, bool adagrad,
                                     hipStream_t stream) {
   const int block = 256;
   hipMemsetAsync(long_count, 0, sizeof(int32_t), stream);
+  hipMemsetAsync(n_work, 0, sizeof(int32_t), stream);
   if (adagrad) {
     hipMemsetAsync(long_scratch, 0,
                    sizeof(float) * scratch_rows * (int64_t)width, stream);
@@ -908,9 +946,11 @@ void launch_sorted_optimizer_update(float* weight, float* state, float eps,
                          dim3(block), 0, stream, weight, state, eps,           \
                          sorted_ids, seg, srow, sw, grad_out, lr, max_segs,    \
                          width, long_rows, long_count);                        \
+      hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,  \
+                         long_rows, long_count, seg, work_items, n_work);      \
       hipLaunchKernelGGL((sorted_adagrad_long_accum<VEC, HASW>), dim3(2048),   \
                          dim3(block), 0, stream, seg, srow, sw, grad_out,      \
-                         width, long_rows, long_count, long_scratch);          \
+                         width, long_rows, work_items, n_work, long_scratch);  \
       hipLaunchKernelGGL(sorted_adagrad_long_finalize, dim3(256), dim3(block), \
                          0, stream, weight, state, eps, sorted_ids, seg, lr,   \
                          width, long_rows, long_count, long_scratch);          \
@@ -919,10 +959,12 @@ void launch_sorted_optimizer_update(float* weight, float* state, float eps,
                          dim3(block), 0, stream, weight, nullptr, 0.f,         \
                          sorted_ids, seg, srow, sw, grad_out, lr, max_segs,    \
                          width, long_rows, long_count);                        \
+      hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,  \
+                         long_rows, long_count, seg, work_items, n_work);      \
       hipLaunchKernelGGL((sorted_sgd_update_long<VEC, HASW>), dim3(2048),      \
                          dim3(block), 0, stream, weight, sorted_ids, seg,      \
                          srow, sw, grad_out, lr, width, long_rows,             \
-                         long_count);                                          \
+                         work_items, n_work);                                  \
     }                                                                          \
   } while (0)
   if (width % 4 == 0) {

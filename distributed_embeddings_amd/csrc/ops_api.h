@@ -12,6 +12,7 @@ void launch_csr_lookup_forward(const float* params, const int64_t* values,
                                float* out, int64_t num_rows, int64_t nnz,
                                int64_t vocab, int width, bool mean,
                                int64_t* long_rows, int32_t* long_count,
+                               int64_t* work_items, int32_t* n_work,
                                hipStream_t stream);
 
 void launch_row_to_split(const int64_t* rows, int64_t nnz, int64_t num_rows,
@@ -78,7 +79,8 @@ void launch_sorted_optimizer_update(float* weight, float* state, float eps,
                                     const float* sw, const float* grad_out,
                                     const float* lr, int64_t max_segs,
                                     int width, int64_t* long_rows,
-                                    int32_t* long_count, float* long_scratch,
+                                    int32_t* long_count, int64_t* work_items,
+                                    int32_t* n_work, float* long_scratch,
                                     int64_t scratch_rows, bool adagrad,
                                     hipStream_t stream);
 
