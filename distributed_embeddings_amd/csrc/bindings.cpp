@@ -370,7 +370,8 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
                                  seg_tmp.data_ptr<int64_t>(),
                                  srow.data_ptr<int64_t>(), sw_ptr,
                                  grad_out.data_ptr<float>(),
-                                 lr.data_ptr<float>(), nnz, width,
+                                 lr.data_ptr<float>(),
+                                 nu_dev.data_ptr<int32_t>(), nnz, width,
                                  long_rows.data_ptr<int64_t>(),
                                  long_count.data_ptr<int32_t>(),
                                  work_items.data_ptr<int64_t>(),

@@ -710,8 +710,48 @@ __global__ void pad_seg_offsets(int64_t* __restrict__ seg, int64_t n,
   if (i >= *num_unique) seg[i] = bounds[1];  // first OOB position
 }
 
-template <int VEC, bool HAS_W, bool ADAGRAD>
-__global__ void sorted_sgd_update(float* __restrict__ weight,
+// Segment-sum helper: reduces grad_out rows of segment [ks, ke) into
+// acc[V] per lane; NARROW (TILE>0): NSUB=64/TILE sub-tiles split k with a
+// cross-sub shuffle fold afterwards; WIDE: V floats per lane, k unrolled x2.
+template <int TILE, int VEC, bool HAS_W>
+__device__ __forceinline__ void seg_grad_reduce(
+    const int64_t* __restrict__ srow, const float* __restrict__ sw,
+    const float* __restrict__ grad_out, int width, int64_t ks, int64_t ke,
+    int lane, float* acc /* size V */) {
+  if constexpr (TILE > 0) {
+    constexpr int T = TILE > 0 ? TILE : 1;
+    constexpr int NSUB = WAVE / T;
+    const int tl = lane % T;
+    const int sub = lane / T;
+    float a = 0.f;
+    if (tl < width) {
+      int64_t k = ks + sub;
+      for (; k + NSUB < ke; k += 2 * NSUB) {
+        const int64_t r0 = srow[k], r1 = srow[k + NSUB];
+        const float w0 = HAS_W ? sw[k] : 1.f;
+        const float w1 = HAS_W ? sw[k + NSUB] : 1.f;
+        a += w0 * grad_out[r0 * (int64_t)width + tl] +
+             w1 * grad_out[r1 * (int64_t)width + tl];
+      }
+      if (k < ke) {
+        a += (HAS_W ? sw[k] : 1.f) * grad_out[srow[k] * (int64_t)width + tl];
+      }
+    }
+#pragma unroll
+    for (int off = WAVE / 2; off >= T; off >>= 1) a += __shfl_down(a, off);
+    acc[0] = a;  // valid on sub 0 lanes
+  } else {
+    constexpr int V = VEC > 0 ? VEC : 1;
+    const int col0 = lane * V;  // caller loops width chunks externally
+    (void)col0;
+  }
+}
+
+// Short segments: one wave per segment, direct (non-atomic) update.
+// Grid-strides only over the REAL segment count (*nu_ptr), not the padded
+// nnz-sized buffer.
+template <int TILE, int VEC, bool HAS_W, bool ADAGRAD>
+__global__ void sorted_opt_update(float* __restrict__ weight,
                                   float* __restrict__ state, float eps,
                                   const int64_t* __restrict__ sorted_ids,
                                   const int64_t* __restrict__ seg,
@@ -719,78 +759,100 @@ __global__ void sorted_sgd_update(float* __restrict__ weight,
                                   const float* __restrict__ sw,
                                   const float* __restrict__ grad_out,
                                   const float* __restrict__ lr_ptr,
-                                  int64_t max_segs, int width,
+                                  const int32_t* __restrict__ nu_ptr,
+                                  int64_t long_thresh, int width,
                                   int64_t* __restrict__ long_rows,
                                   int32_t* __restrict__ long_count) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
   const float lr = *lr_ptr;
-  constexpr int CHUNK = WAVE * VEC;
-  for (int64_t r = wave_id; r < max_segs; r += n_waves) {
+  const int64_t n_segs = *nu_ptr;
+  for (int64_t r = wave_id; r < n_segs; r += n_waves) {
     const int64_t s = seg[r], e = seg[r + 1];
     if (s >= e) continue;
-    if (e - s > LONG_T) {
+    if (e - s > long_thresh) {
       if (lane == 0) long_rows[atomicAdd(long_count, 1)] = r;
       continue;
     }
     const int64_t uid = sorted_ids[s];
-    for (int cbase = 0; cbase < width; cbase += CHUNK) {
-      float acc[VEC];
-#pragma unroll
-      for (int v = 0; v < VEC; ++v) acc[v] = 0.f;
-      const int col0 = cbase + lane * VEC;
-      for (int64_t k = s; k < e; ++k) {
-        const float w = HAS_W ? sw[k] : 1.f;
-        const float* gp = grad_out + srow[k] * (int64_t)width + col0;
-        if (VEC == 4 && col0 + 4 <= width) {
-          const float4 g4 = *reinterpret_cast<const float4*>(gp);
-          acc[0] += w * g4.x; acc[1] += w * g4.y;
-          acc[2] += w * g4.z; acc[3] += w * g4.w;
+    if constexpr (TILE > 0) {
+      constexpr int T = TILE > 0 ? TILE : 1;
+      float acc[1];
+      seg_grad_reduce<T, 0, HAS_W>(srow, sw, grad_out, width, s, e, lane, acc);
+      const int tl = lane % T;
+      if (lane / T == 0 && tl < width) {
+        const int64_t o = uid * (int64_t)width + tl;
+        if (ADAGRAD) {
+          const float g = acc[0];
+          const float st = state[o] + g * g;
+          state[o] = st;
+          weight[o] -= lr * g / (sqrtf(st) + eps);
         } else {
-#pragma unroll
-          for (int v = 0; v < VEC; ++v)
-            if (col0 + v < width) acc[v] += w * gp[v];
+          weight[o] -= lr * acc[0];
         }
       }
-      float* wp = weight + uid * (int64_t)width + col0;
-      if (ADAGRAD) {
-        float* sp = state + uid * (int64_t)width + col0;
+    } else {
+      constexpr int V = VEC > 0 ? VEC : 1;
+      constexpr int CH = WAVE * V;
+      for (int cbase = 0; cbase < width; cbase += CH) {
+        float acc[V];
 #pragma unroll
-        for (int v = 0; v < VEC; ++v) {
-          if (col0 + v < width) {
-            const float g = acc[v];
-            const float s = sp[v] + g * g;
-            sp[v] = s;
-            wp[v] -= lr * g / (sqrtf(s) + eps);
+        for (int v = 0; v < V; ++v) acc[v] = 0.f;
+        const int col0 = cbase + lane * V;
+        for (int64_t k = s; k < e; ++k) {
+          const float w = HAS_W ? sw[k] : 1.f;
+          const float* gp = grad_out + srow[k] * (int64_t)width + col0;
+          if (V == 4 && col0 + 4 <= width) {
+            const float4 g4 = *reinterpret_cast<const float4*>(gp);
+            acc[0] += w * g4.x; acc[1] += w * g4.y;
+            acc[2] += w * g4.z; acc[3] += w * g4.w;
+          } else {
+#pragma unroll
+            for (int v = 0; v < V; ++v)
+              if (col0 + v < width) acc[v] += w * gp[v];
           }
         }
-      } else {
+        if (ADAGRAD) {
+          float* sp = state + uid * (int64_t)width;
+          float* wp = weight + uid * (int64_t)width;
 #pragma unroll
-        for (int v = 0; v < VEC; ++v)
-          if (col0 + v < width) wp[v] -= lr * acc[v];
+          for (int v = 0; v < V; ++v) {
+            if (col0 + v < width) {
+              const float g = acc[v];
+              const float st = sp[col0 + v] + g * g;
+              sp[col0 + v] = st;
+              wp[col0 + v] -= lr * g / (sqrtf(st) + eps);
+            }
+          }
+        } else {
+          float* wp = weight + uid * (int64_t)width;
+#pragma unroll
+          for (int v = 0; v < V; ++v)
+            if (col0 + v < width) wp[v + col0] -= lr * acc[v];
+        }
       }
     }
   }
 }
 
-// SGD long segments: atomic partials straight into the weight (linear).
-template <int VEC, bool HAS_W>
-__global__ void sorted_sgd_update_long(float* __restrict__ weight,
-                                       const int64_t* __restrict__ sorted_ids,
-                                       const int64_t* __restrict__ seg,
-                                       const int64_t* __restrict__ srow,
-                                       const float* __restrict__ sw,
-                                       const float* __restrict__ grad_out,
-                                       const float* __restrict__ lr_ptr,
-                                       int width,
-                                       const int64_t* __restrict__ long_rows,
-                                       const int64_t* __restrict__ work_items,
-                                       const int32_t* __restrict__ n_work_ptr) {
+// Long segments, SGD: chunk partials atomically into the weight (linear).
+// Long segments, Adagrad: chunk partials into scratch rows, then finalize.
+template <int TILE, int VEC, bool HAS_W, bool TO_SCRATCH>
+__global__ void sorted_opt_long(float* __restrict__ target,  // weight or scratch
+                                const int64_t* __restrict__ sorted_ids,
+                                const int64_t* __restrict__ seg,
+                                const int64_t* __restrict__ srow,
+                                const float* __restrict__ sw,
+                                const float* __restrict__ grad_out,
+                                const float* __restrict__ lr_ptr, int width,
+                                const int64_t* __restrict__ long_rows,
+                                const int64_t* __restrict__ work_items,
+                                const int32_t* __restrict__ n_work_ptr) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
-  const float lr = *lr_ptr;
+  const float lr = TO_SCRATCH ? 1.f : *lr_ptr;
   const int64_t n_items = *n_work_ptr;
   for (int64_t item = wave_id; item < n_items; item += n_waves) {
     const int64_t w_it = work_items[item];
@@ -798,19 +860,27 @@ __global__ void sorted_sgd_update_long(float* __restrict__ weight,
     const int64_t chunk = w_it & 0xffffff;
     const int64_t r = long_rows[li];
     const int64_t s = seg[r], e = seg[r + 1];
-    const int64_t k0 = s + chunk * LONG_T;
-    if (k0 >= e) continue;
-    const int64_t uid = sorted_ids[s];
-    constexpr int V = VEC > 0 ? VEC : 1;
-    constexpr int CH = WAVE * V;
-    for (int cbase = 0; cbase < width; cbase += CH) {
-      float acc[V];
+    const int64_t ks = s + chunk * LONG_T;
+    if (ks >= e) continue;
+    const int64_t ke = min(ks + (int64_t)LONG_T, e);
+    const int64_t trow = TO_SCRATCH ? li : sorted_ids[s];
+    if constexpr (TILE > 0) {
+      constexpr int T = TILE > 0 ? TILE : 1;
+      float acc[1];
+      seg_grad_reduce<T, 0, HAS_W>(srow, sw, grad_out, width, ks, ke, lane, acc);
+      const int tl = lane % T;
+      if (lane / T == 0 && tl < width) {
+        atomicAdd(&target[trow * (int64_t)width + tl],
+                  TO_SCRATCH ? acc[0] : -lr * acc[0]);
+      }
+    } else {
+      constexpr int V = VEC > 0 ? VEC : 1;
+      constexpr int CH = WAVE * V;
+      for (int cbase = 0; cbase < width; cbase += CH) {
+        float acc[V];
 #pragma unroll
-      for (int v = 0; v < V; ++v) acc[v] = 0.f;
-      const int col0 = cbase + lane * V;
-      {
-        const int64_t ks = k0;
-        const int64_t ke = min(ks + (int64_t)LONG_T, e);
+        for (int v = 0; v < V; ++v) acc[v] = 0.f;
+        const int col0 = cbase + lane * V;
         for (int64_t k = ks; k < ke; ++k) {
           const float w = HAS_W ? sw[k] : 1.f;
           const float* gp = grad_out + srow[k] * (int64_t)width + col0;
@@ -824,68 +894,12 @@ __global__ void sorted_sgd_update_long(float* __restrict__ weight,
               if (col0 + v < width) acc[v] += w * gp[v];
           }
         }
+        float* tp = target + trow * (int64_t)width + col0;
+#pragma unroll
+        for (int v = 0; v < V; ++v)
+          if (col0 + v < width)
+            atomicAdd(&tp[v], TO_SCRATCH ? acc[v] : -lr * acc[v]);
       }
-      float* wp = weight + uid * (int64_t)width + col0;
-#pragma unroll
-      for (int v = 0; v < V; ++v)
-        if (col0 + v < width) atomicAdd(&wp[v], -lr * acc[v]);
-    }
-  }
-}
-
-// Adagrad long segments: the update is non-linear, so chunk partials
-// accumulate into a scratch row per long segment (atomicAdd), then a
-// finalize kernel applies state += g^2; w -= lr*g/(sqrt(state)+eps).
-template <int VEC, bool HAS_W>
-__global__ void sorted_adagrad_long_accum(const int64_t* __restrict__ seg,
-                                          const int64_t* __restrict__ srow,
-                                          const float* __restrict__ sw,
-                                          const float* __restrict__ grad_out,
-                                          int width,
-                                          const int64_t* __restrict__ long_rows,
-                                          const int64_t* __restrict__ work_items,
-                                          const int32_t* __restrict__ n_work_ptr,
-                                          float* __restrict__ scratch) {
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
-  const int64_t n_items = *n_work_ptr;
-  for (int64_t item = wave_id; item < n_items; item += n_waves) {
-    const int64_t w_it = work_items[item];
-    const int64_t li = w_it >> 24;
-    const int64_t chunk = w_it & 0xffffff;
-    const int64_t r = long_rows[li];
-    const int64_t s = seg[r], e = seg[r + 1];
-    const int64_t k0 = s + chunk * LONG_T;
-    if (k0 >= e) continue;
-    constexpr int V = VEC > 0 ? VEC : 1;
-    constexpr int CH = WAVE * V;
-    for (int cbase = 0; cbase < width; cbase += CH) {
-      float acc[V];
-#pragma unroll
-      for (int v = 0; v < V; ++v) acc[v] = 0.f;
-      const int col0 = cbase + lane * V;
-      {
-        const int64_t ks = k0;
-        const int64_t ke = min(ks + (int64_t)LONG_T, e);
-        for (int64_t k = ks; k < ke; ++k) {
-          const float w = HAS_W ? sw[k] : 1.f;
-          const float* gp = grad_out + srow[k] * (int64_t)width + col0;
-          if (V == 4 && col0 + 4 <= width) {
-            const float4 g4 = *reinterpret_cast<const float4*>(gp);
-            acc[0] += w * g4.x; acc[1] += w * g4.y;
-            acc[2] += w * g4.z; acc[3] += w * g4.w;
-          } else {
-#pragma unroll
-            for (int v = 0; v < V; ++v)
-              if (col0 + v < width) acc[v] += w * gp[v];
-          }
-        }
-      }
-      float* sp = scratch + li * (int64_t)width + col0;
-#pragma unroll
-      for (int v = 0; v < V; ++v)
-        if (col0 + v < width) atomicAdd(&sp[v], acc[v]);
     }
   }
 }
@@ -921,60 +935,87 @@ void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
                      dim3(block), 0, stream, seg, n, num_unique, bounds);
 }
 
+template <int TILE, int VEC>
+static void launch_sorted_opt_pair(float* weight, float* state, float eps,
+                                   const int64_t* sorted_ids,
+                                   const int64_t* seg, const int64_t* srow,
+                                   const float* sw, const float* grad_out,
+                                   const float* lr, const int32_t* nu_ptr,
+                                   int64_t max_segs, int width,
+                                   int64_t* long_rows, int32_t* long_count,
+                                   int64_t* work_items, int32_t* n_work,
+                                   float* long_scratch, bool adagrad,
+                                   hipStream_t stream) {
+  const int block = 256;
+  const int64_t row_waves =
+      TILE > 0 ? cdiv64(max_segs, WAVE / (TILE > 0 ? TILE : 1)) : max_segs;
+  const int grid = pick_grid(row_waves, block / WAVE);
+#define SU(HASW, ADA)                                                          \
+  hipLaunchKernelGGL((sorted_opt_update<TILE, VEC, HASW, ADA>), dim3(grid),    \
+                     dim3(block), 0, stream, weight, state, eps, sorted_ids,   \
+                     seg, srow, sw, grad_out, lr, nu_ptr, (int64_t)LONG_T,     \
+                     width, long_rows, long_count)
+#define SL(HASW, SCR)                                                          \
+  hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, SCR>), dim3(2048),      \
+                     dim3(block), 0, stream, SCR ? long_scratch : weight,      \
+                     sorted_ids, seg, srow, sw, grad_out, lr, width,           \
+                     long_rows, work_items, n_work)
+  if (adagrad) {
+    if (sw) SU(true, true); else SU(false, true);
+  } else {
+    if (sw) SU(true, false); else SU(false, false);
+  }
+  hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,
+                     long_rows, long_count, seg, work_items, n_work);
+  if (adagrad) {
+    if (sw) SL(true, true); else SL(false, true);
+    hipLaunchKernelGGL(sorted_adagrad_long_finalize, dim3(256), dim3(block), 0,
+                       stream, weight, state, eps, sorted_ids, seg, lr, width,
+                       long_rows, long_count, long_scratch);
+  } else {
+    if (sw) SL(true, false); else SL(false, false);
+  }
+#undef SU
+#undef SL
+}
+
 void launch_sorted_optimizer_update(float* weight, float* state, float eps,
                                     const int64_t* sorted_ids,
                                     const int64_t* seg, const int64_t* srow,
                                     const float* sw, const float* grad_out,
-                                    const float* lr, int64_t max_segs,
-                                    int width, int64_t* long_rows,
-                                    int32_t* long_count, int64_t* work_items,
-                                    int32_t* n_work, float* long_scratch,
-                                    int64_t scratch_rows, bool adagrad,
-                                    hipStream_t stream) {
-  const int block = 256;
+                                    const float* lr, const int32_t* nu_ptr,
+                                    int64_t max_segs, int width,
+                                    int64_t* long_rows, int32_t* long_count,
+                                    int64_t* work_items, int32_t* n_work,
+                                    float* long_scratch, int64_t scratch_rows,
+                                    bool adagrad, hipStream_t stream) {
   hipMemsetAsync(long_count, 0, sizeof(int32_t), stream);
   hipMemsetAsync(n_work, 0, sizeof(int32_t), stream);
   if (adagrad) {
     hipMemsetAsync(long_scratch, 0,
                    sizeof(float) * scratch_rows * (int64_t)width, stream);
   }
-  const int grid = pick_grid(max_segs, block / WAVE);
-#define LU(VEC, HASW)                                                          \
-  do {                                                                         \
-    if (adagrad) {                                                             \
-      hipLaunchKernelGGL((sorted_sgd_update<VEC, HASW, true>), dim3(grid),     \
-                         dim3(block), 0, stream, weight, state, eps,           \
-                         sorted_ids, seg, srow, sw, grad_out, lr, max_segs,    \
-                         width, long_rows, long_count);                        \
-      hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,  \
-                         long_rows, long_count, seg, work_items, n_work);      \
-      hipLaunchKernelGGL((sorted_adagrad_long_accum<VEC, HASW>), dim3(2048),   \
-                         dim3(block), 0, stream, seg, srow, sw, grad_out,      \
-                         width, long_rows, work_items, n_work, long_scratch);  \
-      hipLaunchKernelGGL(sorted_adagrad_long_finalize, dim3(256), dim3(block), \
-                         0, stream, weight, state, eps, sorted_ids, seg, lr,   \
-                         width, long_rows, long_count, long_scratch);          \
-    } else {                                                                   \
-      hipLaunchKernelGGL((sorted_sgd_update<VEC, HASW, false>), dim3(grid),    \
-                         dim3(block), 0, stream, weight, nullptr, 0.f,         \
-                         sorted_ids, seg, srow, sw, grad_out, lr, max_segs,    \
-                         width, long_rows, long_count);                        \
-      hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,  \
-                         long_rows, long_count, seg, work_items, n_work);      \
-      hipLaunchKernelGGL((sorted_sgd_update_long<VEC, HASW>), dim3(2048),      \
-                         dim3(block), 0, stream, weight, sorted_ids, seg,      \
-                         srow, sw, grad_out, lr, width, long_rows,             \
-                         work_items, n_work);                                  \
-    }                                                                          \
-  } while (0)
-  if (width % 4 == 0) {
-    if (sw) LU(4, true); else LU(4, false);
+#define ARGS weight, state, eps, sorted_ids, seg, srow, sw, grad_out, lr,    \
+             nu_ptr, max_segs, width, long_rows, long_count, work_items,     \
+             n_work, long_scratch, adagrad, stream
+  if (width <= 64) {
+    switch (next_pow2(width)) {
+      case 1: launch_sorted_opt_pair<1, 0>(ARGS); break;
+      case 2: launch_sorted_opt_pair<2, 0>(ARGS); break;
+      case 4: launch_sorted_opt_pair<4, 0>(ARGS); break;
+      case 8: launch_sorted_opt_pair<8, 0>(ARGS); break;
+      case 16: launch_sorted_opt_pair<16, 0>(ARGS); break;
+      case 32: launch_sorted_opt_pair<32, 0>(ARGS); break;
+      default: launch_sorted_opt_pair<64, 0>(ARGS); break;
+    }
+  } else if (width % 4 == 0) {
+    launch_sorted_opt_pair<0, 4>(ARGS);
   } else if (width % 2 == 0) {
-    if (sw) LU(2, true); else LU(2, false);
+    launch_sorted_opt_pair<0, 2>(ARGS);
   } else {
-    if (sw) LU(1, true); else LU(1, false);
+    launch_sorted_opt_pair<0, 1>(ARGS);
   }
-#undef LU
+#undef ARGS
 }
 
 // ---------------------------------------------------------------------------

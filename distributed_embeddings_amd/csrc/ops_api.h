@@ -77,12 +77,12 @@ void launch_sorted_optimizer_update(float* weight, float* state, float eps,
                                     const int64_t* sorted_ids,
                                     const int64_t* seg, const int64_t* srow,
                                     const float* sw, const float* grad_out,
-                                    const float* lr, int64_t max_segs,
-                                    int width, int64_t* long_rows,
-                                    int32_t* long_count, int64_t* work_items,
-                                    int32_t* n_work, float* long_scratch,
-                                    int64_t scratch_rows, bool adagrad,
-                                    hipStream_t stream);
+                                    const float* lr, const int32_t* nu_ptr,
+                                    int64_t max_segs, int width,
+                                    int64_t* long_rows, int32_t* long_count,
+                                    int64_t* work_items, int32_t* n_work,
+                                    float* long_scratch, int64_t scratch_rows,
+                                    bool adagrad, hipStream_t stream);
 
 void launch_sparse_row_update(float* weight, float* state, const int64_t* ids,
                               const float* grad, int64_t num_rows, int width,
