@@ -37,37 +37,43 @@ bool use_rocprim_sort() {
   return v == 1;
 }
 
-void sort_pairs_dispatch(torch::Tensor masked, torch::Tensor iota,
-                         torch::Tensor sorted_ids, torch::Tensor sorted_pos,
-                         int64_t nnz, int end_bit, hipStream_t stream) {
+// Sorts ids (masking OOB to the `vocab` sentinel) by packing
+// (id << 32 | position) into one u64 array; returns sorted_ids + the sorted
+// original positions.  Hand-written sort by default, rocPRIM via env.
+void sort_ids_dispatch(torch::Tensor values, int64_t vocab, int64_t nnz,
+                       int end_bit, torch::Tensor sorted_ids,
+                       torch::Tensor sorted_pos, hipStream_t stream) {
+  TORCH_CHECK(end_bit <= 32, "fused table vocab must fit 32 bits");
+  auto u64 = values.options().dtype(torch::kUInt64);
+  auto i32 = values.options().dtype(torch::kInt32);
+  auto packed = torch::empty({nnz}, u64);
+  launch_mask_oob_pack(values.data_ptr<int64_t>(), nnz, vocab,
+                       (uint64_t*)packed.data_ptr(), stream);
+  auto packed_out = torch::empty({nnz}, u64);
   if (use_rocprim_sort()) {
-    size_t temp_bytes = csr_backward_temp_bytes(nnz, 0);
+    size_t temp_bytes = rocprim_sort_keys_temp_bytes(nnz);
     auto temp = torch::empty({(int64_t)temp_bytes},
-                             masked.options().dtype(torch::kUInt8));
-    auto err = run_sort_pairs(temp.data_ptr(), temp_bytes,
-                              masked.data_ptr<int64_t>(),
-                              sorted_ids.data_ptr<int64_t>(),
-                              iota.data_ptr<int32_t>(),
-                              sorted_pos.data_ptr<int32_t>(), nnz, end_bit,
-                              stream);
-    TORCH_CHECK(err == hipSuccess, "radix_sort_pairs failed");
-    return;
+                             values.options().dtype(torch::kUInt8));
+    auto err = run_sort_keys_u64(temp.data_ptr(), temp_bytes,
+                                 (const uint64_t*)packed.data_ptr(),
+                                 (uint64_t*)packed_out.data_ptr(), nnz, 32,
+                                 32 + end_bit, stream);
+    TORCH_CHECK(err == hipSuccess, "radix_sort_keys failed");
+  } else {
+    auto keys_tmp = torch::empty({nnz}, u64);
+    const int64_t hist_elems = (int64_t)custom_radix_sort_hist_elems(nnz);
+    auto hist = torch::empty({hist_elems}, i32);
+    auto scan_sums = torch::empty({4096}, i32);
+    custom_radix_sort_keys((const uint64_t*)packed.data_ptr(),
+                           (uint64_t*)packed_out.data_ptr(),
+                           (uint64_t*)keys_tmp.data_ptr(),
+                           hist.data_ptr<int32_t>(),
+                           scan_sums.data_ptr<int32_t>(), nnz, 32,
+                           32 + end_bit, stream);
   }
-  auto i64 = masked.options();
-  auto i32 = masked.options().dtype(torch::kInt32);
-  auto keys_tmp = torch::empty({nnz}, i64);
-  auto vals_tmp = torch::empty({nnz}, i32);
-  const int64_t hist_elems = (int64_t)custom_radix_sort_hist_elems(nnz);
-  auto hist = torch::empty({hist_elems}, i32);
-  auto scan_sums = torch::empty({4096}, i32);
-  custom_radix_sort_pairs(masked.data_ptr<int64_t>(),
-                          sorted_ids.data_ptr<int64_t>(),
-                          iota.data_ptr<int32_t>(),
-                          sorted_pos.data_ptr<int32_t>(),
-                          keys_tmp.data_ptr<int64_t>(),
-                          vals_tmp.data_ptr<int32_t>(),
-                          hist.data_ptr<int32_t>(),
-                          scan_sums.data_ptr<int32_t>(), nnz, end_bit, stream);
+  launch_unpack_sorted((const uint64_t*)packed_out.data_ptr(), nnz,
+                       sorted_ids.data_ptr<int64_t>(),
+                       sorted_pos.data_ptr<int32_t>(), stream);
 }
 
 torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
@@ -128,12 +134,7 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
     return {torch::empty({0}, i64), torch::empty({0, width}, f32)};
   }
 
-  // 1. mask OOB -> sentinel `vocab`; iota payload.
-  auto masked = torch::empty({nnz}, i64);
-  auto iota = torch::empty({nnz}, i32);
-  launch_mask_oob_and_iota(values.data_ptr<int64_t>(), nnz, vocab,
-                           masked.data_ptr<int64_t>(), iota.data_ptr<int32_t>(),
-                           stream);
+  // 1+3 fused below: mask OOB -> sentinel and sort packed (id, pos).
 
   // 2. per-id row ids (+ mean weights keyed by original position).
   auto row_ids = torch::empty({nnz}, i32);
@@ -149,8 +150,8 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
   // 3. radix sort (ids, position) — end_bit covers [0, vocab] inclusive.
   auto sorted_ids = torch::empty({nnz}, i64);
   auto sorted_pos = torch::empty({nnz}, i32);
-  sort_pairs_dispatch(masked, iota, sorted_ids, sorted_pos, nnz,
-                      log2_ceil(vocab + 1), stream);
+  sort_ids_dispatch(values, vocab, nnz, log2_ceil(vocab + 1), sorted_ids,
+                    sorted_pos, stream);
   size_t temp_bytes = csr_backward_temp_bytes(nnz, vocab);
   auto temp = torch::empty({(int64_t)temp_bytes},
                            f32.dtype(torch::kUInt8));
@@ -297,11 +298,6 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
   auto i32 = values.options().dtype(torch::kInt32);
   auto f32 = grad_out.options();
 
-  auto masked = torch::empty({nnz}, i64);
-  auto iota = torch::empty({nnz}, i32);
-  launch_mask_oob_and_iota(values.data_ptr<int64_t>(), nnz, vocab,
-                           masked.data_ptr<int64_t>(), iota.data_ptr<int32_t>(),
-                           stream);
   auto row_ids = torch::empty({nnz}, i32);
   torch::Tensor w;
   float* w_ptr = nullptr;
@@ -313,8 +309,8 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
                         row_ids.data_ptr<int32_t>(), w_ptr, mean, stream);
   auto sorted_ids = torch::empty({nnz}, i64);
   auto sorted_pos = torch::empty({nnz}, i32);
-  sort_pairs_dispatch(masked, iota, sorted_ids, sorted_pos, nnz,
-                      log2_ceil(vocab + 1), stream);
+  sort_ids_dispatch(values, vocab, nnz, log2_ceil(vocab + 1), sorted_ids,
+                    sorted_pos, stream);
   size_t temp_bytes = csr_backward_temp_bytes(nnz, vocab);
   auto temp = torch::empty({(int64_t)temp_bytes}, f32.dtype(torch::kUInt8));
   auto srow = torch::empty({nnz}, i64);

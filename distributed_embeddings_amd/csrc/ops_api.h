@@ -101,3 +101,19 @@ void custom_radix_sort_pairs(const int64_t* keys_in, int64_t* keys_out,
                              int32_t* hist, int32_t* scan_sums, int64_t n,
                              int end_bit, hipStream_t stream);
 size_t custom_radix_sort_hist_elems(int64_t n);
+
+void custom_radix_sort_keys(const uint64_t* keys_in, uint64_t* keys_out,
+                            uint64_t* keys_tmp, int32_t* hist,
+                            int32_t* scan_sums, int64_t n, int begin_bit,
+                            int end_bit, hipStream_t stream);
+
+void launch_mask_oob_pack(const int64_t* ids, int64_t n, int64_t vocab,
+                          uint64_t* packed, hipStream_t stream);
+void launch_unpack_sorted(const uint64_t* packed, int64_t n,
+                          int64_t* sorted_ids, int32_t* sorted_pos,
+                          hipStream_t stream);
+hipError_t run_sort_keys_u64(void* temp, size_t temp_bytes,
+                             const uint64_t* keys_in, uint64_t* keys_out,
+                             int64_t n, int begin_bit, int end_bit,
+                             hipStream_t stream);
+size_t rocprim_sort_keys_temp_bytes(int64_t n);
