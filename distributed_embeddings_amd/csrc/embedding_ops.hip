@@ -529,17 +529,6 @@ void launch_expand_row_ids(const int64_t* splits, int64_t num_rows,
                        stream, splits, num_rows, row_ids, nullptr);
 }
 
-hipError_t run_sort_pairs(void* temp, size_t temp_bytes, const int64_t* keys_in,
-                          int64_t* keys_out, const int32_t* vals_in,
-                          int32_t* vals_out, int64_t n, int end_bit,
-                          hipStream_t stream) {
-  // begin_bit 0, end_bit covers vocab bits + sign handling: ids are
-  // non-negative after host-side masking of OOB (OOB mapped to vocab).
-  return rocprim::radix_sort_pairs(temp, temp_bytes, keys_in, keys_out,
-                                   vals_in, vals_out, (size_t)n, 0u,
-                                   (unsigned)end_bit, stream);
-}
-
 hipError_t run_inclusive_scan_i32(void* temp, size_t temp_bytes,
                                   const int32_t* in, int32_t* out, int64_t n,
                                   hipStream_t stream) {
@@ -564,77 +553,6 @@ void launch_scatter_unique(const int64_t* sorted_ids, const int32_t* head,
   hipLaunchKernelGGL(scatter_unique, dim3((int)grid), dim3(block), 0, stream,
                      sorted_ids, head, pos, n, vocab, unique_ids, seg_offsets,
                      num_unique);
-}
-
-// Map OOB ids to the sentinel `vocab` (sorts last) + iota payload.
-__global__ void mask_oob_and_iota(const int64_t* __restrict__ ids, int64_t n,
-                                  int64_t vocab, int64_t* __restrict__ masked,
-                                  int32_t* __restrict__ iota) {
-  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n) return;
-  const int64_t id = ids[i];
-  masked[i] = (id < 0 || id >= vocab) ? vocab : id;
-  iota[i] = (int32_t)i;
-}
-
-// Packed variant: (masked_id << 32) | position — the whole sort pipeline
-// moves ONE u64 array (digits live in bits [32, 32+log2(vocab+1))).
-__global__ void mask_oob_pack(const int64_t* __restrict__ ids, int64_t n,
-                              int64_t vocab, uint64_t* __restrict__ packed) {
-  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n) return;
-  const int64_t id = ids[i];
-  const uint64_t m = (id < 0 || id >= vocab) ? (uint64_t)vocab : (uint64_t)id;
-  packed[i] = (m << 32) | (uint32_t)i;
-}
-
-__global__ void unpack_sorted(const uint64_t* __restrict__ packed, int64_t n,
-                              int64_t* __restrict__ sorted_ids,
-                              int32_t* __restrict__ sorted_pos) {
-  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n) return;
-  const uint64_t p = packed[i];
-  sorted_ids[i] = (int64_t)(p >> 32);
-  sorted_pos[i] = (int32_t)(uint32_t)(p & 0xffffffffu);
-}
-
-void launch_mask_oob_pack(const int64_t* ids, int64_t n, int64_t vocab,
-                          uint64_t* packed, hipStream_t stream) {
-  const int block = 256;
-  hipLaunchKernelGGL(mask_oob_pack, dim3((int)cdiv64(n, block)), dim3(block),
-                     0, stream, ids, n, vocab, packed);
-}
-
-void launch_unpack_sorted(const uint64_t* packed, int64_t n,
-                          int64_t* sorted_ids, int32_t* sorted_pos,
-                          hipStream_t stream) {
-  const int block = 256;
-  hipLaunchKernelGGL(unpack_sorted, dim3((int)cdiv64(n, block)), dim3(block),
-                     0, stream, packed, n, sorted_ids, sorted_pos);
-}
-
-hipError_t run_sort_keys_u64(void* temp, size_t temp_bytes,
-                             const uint64_t* keys_in, uint64_t* keys_out,
-                             int64_t n, int begin_bit, int end_bit,
-                             hipStream_t stream) {
-  return rocprim::radix_sort_keys(temp, temp_bytes, keys_in, keys_out,
-                                  (size_t)n, (unsigned)begin_bit,
-                                  (unsigned)end_bit, stream);
-}
-
-size_t rocprim_sort_keys_temp_bytes(int64_t n) {
-  size_t bytes = 0;
-  rocprim::radix_sort_keys(nullptr, bytes, (const uint64_t*)nullptr,
-                           (uint64_t*)nullptr, (size_t)n);
-  return bytes;
-}
-
-void launch_mask_oob_and_iota(const int64_t* ids, int64_t n, int64_t vocab,
-                              int64_t* masked, int32_t* iota,
-                              hipStream_t stream) {
-  const int block = 256;
-  hipLaunchKernelGGL(mask_oob_and_iota, dim3((int)cdiv64(n, block)),
-                     dim3(block), 0, stream, ids, n, vocab, masked, iota);
 }
 
 // Permute row ids (+ mean weights) by the sorted payload, widening to i64 so
@@ -678,88 +596,21 @@ void launch_find_valid_bounds(const int64_t* sorted_ids, int64_t n,
                      n, vocab, bounds);
 }
 
-// ---------------------------------------------------------------------------
-// Fused in-backward SGD: scatter-apply  w[id] -= lr * (mean? 1/len : 1) *
-// grad_out[row]  directly during autograd backward.  Exact for SGD (updates
-// are linear, order-free modulo fp rounding); removes the whole
-// sort/unique/segmented-sum pipeline AND its num_unique host sync, so the
-// training step becomes hipGraph-capturable.  One wave per CSR row, lanes
-// across the width, float atomicAdd into the table.
-// ---------------------------------------------------------------------------
-
-template <int VEC, bool MEAN>
-__global__ void csr_fused_sgd(float* __restrict__ weight,
-                              const int64_t* __restrict__ values,
-                              const int64_t* __restrict__ splits,
-                              const float* __restrict__ grad_out,
-                              const float* __restrict__ lr_ptr,
-                              int64_t num_rows, int64_t vocab, int width) {
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
-  const float lr = *lr_ptr;
-  constexpr int CHUNK = WAVE * VEC;
-  for (int64_t row = wave_id; row < num_rows; row += n_waves) {
-    const int64_t s = splits[row], e = splits[row + 1];
-    if (e == s) continue;
-    const float scale = MEAN ? -lr / (float)(e - s) : -lr;
-    for (int cbase = 0; cbase < width; cbase += CHUNK) {
-      float g[VEC];
-      const int col0 = cbase + lane * VEC;
-#pragma unroll
-      for (int v = 0; v < VEC; ++v) {
-        g[v] = (col0 + v < width)
-                   ? scale * grad_out[row * (int64_t)width + col0 + v]
-                   : 0.f;
-      }
-      for (int64_t k = s; k < e; ++k) {
-        const int64_t idx = values[k];
-        if (idx < 0 || idx >= vocab) continue;
-        float* wp = weight + idx * (int64_t)width + col0;
-#pragma unroll
-        for (int v = 0; v < VEC; ++v) {
-          if (col0 + v < width) atomicAdd(&wp[v], g[v]);
-        }
-      }
-    }
-  }
-}
-
-void launch_csr_fused_sgd(float* weight, const int64_t* values,
-                          const int64_t* splits, const float* grad_out,
-                          const float* lr, int64_t num_rows, int64_t vocab,
-                          int width, bool mean, hipStream_t stream) {
-  const int block = 256;
-  const int grid = pick_grid(num_rows, block / WAVE);
-#define LS(VEC, MEAN)                                                       \
-  hipLaunchKernelGGL((csr_fused_sgd<VEC, MEAN>), dim3(grid), dim3(block), 0, \
-                     stream, weight, values, splits, grad_out, lr, num_rows, \
-                     vocab, width)
-  if (width % 4 == 0) {
-    if (mean) LS(4, true); else LS(4, false);
-  } else if (width % 2 == 0) {
-    if (mean) LS(2, true); else LS(2, false);
-  } else {
-    if (mean) LS(1, true); else LS(1, false);
-  }
-#undef LS
-}
-
-// ---------------------------------------------------------------------------
-// Sorted fused-SGD update: after the (all-device) sort/unique pipeline, apply
-//   weight[uid] -= lr * sum_{k in segment} w_k * grad_out[srow[k]]
-// directly from the sorted segments.  No host sync (segment count lives in
-// device memory; empty padded segments no-op), no same-address atomic chains
-// (each unique row is written once; long segments use the chunk-wave + atomic
-// pattern of csr_fwd_long with one atomic per chunk partial).
-// ---------------------------------------------------------------------------
-
+// Pads seg_offsets[i >= num_unique] to the first OOB position so padded
+// segments are empty and the update kernels need no host-side count.
 __global__ void pad_seg_offsets(int64_t* __restrict__ seg, int64_t n,
                                 const int32_t* __restrict__ num_unique,
                                 const int64_t* __restrict__ bounds) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i > n) return;
   if (i >= *num_unique) seg[i] = bounds[1];  // first OOB position
+}
+
+void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
+                            const int64_t* bounds, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(pad_seg_offsets, dim3((int)cdiv64(n + 1, block)),
+                     dim3(block), 0, stream, seg, n, num_unique, bounds);
 }
 
 // Segment-sum helper: reduces grad_out rows of segment [ks, ke) into
@@ -980,12 +831,7 @@ __global__ void sorted_adagrad_long_finalize(
   }
 }
 
-void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
-                            const int64_t* bounds, hipStream_t stream) {
-  const int block = 256;
-  hipLaunchKernelGGL(pad_seg_offsets, dim3((int)cdiv64(n + 1, block)),
-                     dim3(block), 0, stream, seg, n, num_unique, bounds);
-}
+
 
 template <int TILE, int VEC>
 static void launch_sorted_opt_pair(float* weight, float* state, float eps,

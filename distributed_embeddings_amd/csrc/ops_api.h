@@ -24,11 +24,6 @@ void launch_expand_row_ids(const int64_t* splits, int64_t num_rows,
                            int32_t* row_ids, float* w, bool mean,
                            hipStream_t stream);
 
-hipError_t run_sort_pairs(void* temp, size_t temp_bytes, const int64_t* keys_in,
-                          int64_t* keys_out, const int32_t* vals_in,
-                          int32_t* vals_out, int64_t n, int end_bit,
-                          hipStream_t stream);
-
 hipError_t run_inclusive_scan_i32(void* temp, size_t temp_bytes,
                                   const int32_t* in, int32_t* out, int64_t n,
                                   hipStream_t stream);
@@ -40,10 +35,6 @@ void launch_scatter_unique(const int64_t* sorted_ids, const int32_t* head,
                            const int32_t* pos, int64_t n, int64_t vocab,
                            int64_t* unique_ids, int64_t* seg_offsets,
                            int32_t* num_unique, hipStream_t stream);
-
-void launch_mask_oob_and_iota(const int64_t* ids, int64_t n, int64_t vocab,
-                              int64_t* masked, int32_t* iota,
-                              hipStream_t stream);
 
 void launch_gather_sorted(const int32_t* perm, const int32_t* row_ids,
                           const float* w, int64_t n, int64_t* srow, float* sw,
@@ -64,11 +55,6 @@ void launch_integer_lookup(const int64_t* keys, int64_t n, int64_t* tkeys,
                            int64_t* out, hipStream_t stream);
 
 size_t integer_lookup_temp_bytes(int64_t max_tokens);
-
-void launch_csr_fused_sgd(float* weight, const int64_t* values,
-                          const int64_t* splits, const float* grad_out,
-                          const float* lr, int64_t num_rows, int64_t vocab,
-                          int width, bool mean, hipStream_t stream);
 
 void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
                             const int64_t* bounds, hipStream_t stream);
@@ -95,11 +81,6 @@ void launch_dot_interact_bwd(const void* gout, const void* feats, void* gfeats,
                              int64_t B, int F, int D, int out_w, int tri_n,
                              hipStream_t stream);
 
-void custom_radix_sort_pairs(const int64_t* keys_in, int64_t* keys_out,
-                             const int32_t* vals_in, int32_t* vals_out,
-                             int64_t* keys_tmp, int32_t* vals_tmp,
-                             int32_t* hist, int32_t* scan_sums, int64_t n,
-                             int end_bit, hipStream_t stream);
 size_t custom_radix_sort_hist_elems(int64_t n);
 
 void custom_radix_sort_keys(const uint64_t* keys_in, uint64_t* keys_out,

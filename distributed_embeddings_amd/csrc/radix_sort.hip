@@ -26,26 +26,6 @@
 
 static inline int64_t rs_cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
-__global__ void rs_histogram(const int64_t* __restrict__ keys, int64_t n,
-                             int shift, int32_t* __restrict__ hist,
-                             int64_t nblocks) {
-  __shared__ int32_t lhist[RS_RADIX];
-  const int lane = threadIdx.x;
-  for (int i = lane; i < RS_RADIX; i += WAVE) lhist[i] = 0;
-  __syncthreads();
-  const int64_t start = (int64_t)blockIdx.x * RS_TILE;
-  const int64_t end = min(start + (int64_t)RS_TILE, n);
-  for (int64_t i = start + lane; i < end; i += WAVE) {
-    const int b = (int)((((uint64_t)keys[i]) >> shift) & 0xff);
-    atomicAdd(&lhist[b], 1);
-  }
-  __syncthreads();
-  // digit-major layout: hist[bucket * nblocks + block]
-  for (int b = lane; b < RS_RADIX; b += WAVE) {
-    hist[(int64_t)b * nblocks + blockIdx.x] = lhist[b];
-  }
-}
-
 // two-level exclusive scan over m = 256*nblocks entries
 __global__ void rs_scan_partials(const int32_t* __restrict__ in, int64_t m,
                                  int32_t* __restrict__ block_sums,
@@ -106,94 +86,6 @@ __global__ void rs_scan_addback(int32_t* __restrict__ out, int64_t m,
   }
 }
 
-__global__ void rs_scatter(const int64_t* __restrict__ keys_in,
-                           const int32_t* __restrict__ vals_in, int64_t n,
-                           int shift, const int32_t* __restrict__ offsets,
-                           int64_t nblocks, int64_t* __restrict__ keys_out,
-                           int32_t* __restrict__ vals_out) {
-  __shared__ int32_t running[RS_RADIX];  // per-bucket offset within this tile
-  const int lane = threadIdx.x;
-  for (int i = lane; i < RS_RADIX; i += WAVE) {
-    running[i] = offsets[(int64_t)i * nblocks + blockIdx.x];
-  }
-  __syncthreads();
-  const int64_t start = (int64_t)blockIdx.x * RS_TILE;
-  const int64_t end = min(start + (int64_t)RS_TILE, n);
-  for (int64_t base = start; base < end; base += WAVE) {
-    const int64_t i = base + lane;
-    const bool active = i < end;
-    const int64_t key = active ? keys_in[i] : 0;
-    const int b = (int)((((uint64_t)key) >> shift) & 0xff);
-    // same-bucket mask across the wave (8 ballot rounds)
-    uint64_t same = ~0ull;
-    #pragma unroll
-    for (int bit = 0; bit < 8; ++bit) {
-      const uint64_t setmask = __ballot((b >> bit) & 1);
-      same &= ((b >> bit) & 1) ? setmask : ~setmask;
-    }
-    const uint64_t act = __ballot(active);
-    same &= act;
-    const uint64_t before = same & ((1ull << lane) - 1ull);
-    const int rank = __popcll(before);
-    const int leader = __ffsll((unsigned long long)same) - 1;
-    int base_off = 0;
-    if (active && lane == leader) {
-      base_off = running[b];
-      running[b] += __popcll(same);
-    }
-    __syncthreads();  // running[] updates visible; also orders groups
-    if (active) {
-      // broadcast leader's base through shuffle
-      const int from = leader;
-      const int bo = __shfl(base_off, from);
-      const int64_t pos = bo + rank;
-      keys_out[pos] = key;
-      vals_out[pos] = vals_in[i];
-    }
-    __syncthreads();
-  }
-}
-
-void custom_radix_sort_pairs(const int64_t* keys_in, int64_t* keys_out,
-                             const int32_t* vals_in, int32_t* vals_out,
-                             int64_t* keys_tmp, int32_t* vals_tmp,
-                             int32_t* hist, int32_t* scan_sums, int64_t n,
-                             int end_bit, hipStream_t stream) {
-  const int64_t nblocks = rs_cdiv(n, RS_TILE);
-  const int passes = (end_bit + 7) / 8;
-  const int64_t m = (int64_t)RS_RADIX * nblocks;
-  const int chunk = (int)rs_cdiv(m, 2048) < 64 ? 64 : (int)rs_cdiv(m, 2048);
-  const int nb_scan = (int)rs_cdiv(m, chunk);
-
-  // ping-pong so the final pass lands in keys_out/vals_out
-  const int64_t* kin = keys_in;
-  const int32_t* vin = vals_in;
-  int64_t* kout;
-  int32_t* vout;
-  bool to_tmp = (passes % 2) == 0;  // even passes: first write to tmp
-  for (int p = 0; p < passes; ++p) {
-    kout = to_tmp ? keys_tmp : keys_out;
-    vout = to_tmp ? vals_tmp : vals_out;
-    const int shift = p * 8;
-    hipLaunchKernelGGL(rs_histogram, dim3((int)nblocks), dim3(WAVE), 0, stream,
-                       kin, n, shift, hist, nblocks);
-    hipLaunchKernelGGL(rs_scan_partials, dim3(nb_scan), dim3(WAVE), 0, stream,
-                       hist, m, scan_sums, hist, chunk);
-    hipLaunchKernelGGL(rs_scan_top, dim3(1), dim3(WAVE), 0, stream, scan_sums,
-                       nb_scan);
-    hipLaunchKernelGGL(rs_scan_addback, dim3(nb_scan), dim3(256), 0, stream,
-                       hist, m, scan_sums, chunk);
-    hipLaunchKernelGGL(rs_scatter, dim3((int)nblocks), dim3(WAVE), 0, stream,
-                       kin, vin, n, shift, hist, nblocks, kout, vout);
-    kin = kout;
-    vin = vout;
-    to_tmp = !to_tmp;
-  }
-}
-
-size_t custom_radix_sort_hist_elems(int64_t n) {
-  return (size_t)RS_RADIX * rs_cdiv(n, RS_TILE);
-}
 
 // Keys-only variant: the backward pipeline packs (id << 32 | position) into
 // one u64, so the sort moves a single array (half the scatter write
