@@ -342,3 +342,49 @@ def test_fused_adagrad_gpu_matches_explicit():
         o2.step()
     assert torch.allclose(e1.weight, e2.weight, atol=1e-3), \
         float((e1.weight - e2.weight).abs().max())
+
+
+@requires_gpu
+def test_distributed_embedding_module_gpu_world1():
+    """Module-level GPU path: mixed hot1/multi-hot/ragged inputs through the
+    fused groups, bf16 output_dtype, fused adagrad training step."""
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd import Ragged
+    torch.manual_seed(17)
+    tables = [de.TableConfig(1000, 64, "sum"), de.TableConfig(50, 64, "sum"),
+              de.TableConfig(300, 32, "mean"), de.TableConfig(77, 64, None)]
+    with torch.device("cuda"):
+        model = de.DistributedEmbedding(tables)
+    g = torch.Generator().manual_seed(3)
+    weights = [torch.randn(c.input_dim, c.output_dim, generator=g)
+               for c in tables]
+    model.set_weights([w.numpy() for w in weights])
+    b = 64
+    x0 = torch.randint(0, 1000, (b, 5), device="cuda")
+    lens = torch.randint(0, 6, (b,), device="cuda")
+    vals = torch.randint(0, 50, (int(lens.sum()),), device="cuda")
+    x1 = Ragged.from_row_lengths(vals, lens)
+    x2 = torch.randint(0, 300, (b, 3), device="cuda")
+    x3 = torch.randint(0, 77, (b,), device="cuda")
+    outs = model([x0, x1, x2, x3], output_dtype=torch.bfloat16)
+    assert all(o.dtype == torch.bfloat16 for o in outs)
+    refs = [weights[0][x0.cpu()].sum(1),
+            None,
+            weights[2][x2.cpu()].mean(1),
+            weights[3][x3.cpu()]]
+    for t in (0, 2, 3):
+        err = (outs[t].float().cpu() - refs[t]).abs().max()
+        assert float(err) < 0.5, f"table {t}: {err}"
+    # ragged row check
+    from distributed_embeddings_amd.ops.embedding_lookup import _csr_lookup_ref
+    ref1 = _csr_lookup_ref(weights[1], vals.cpu(),
+                           Ragged.from_row_lengths(vals.cpu(), lens.cpu()).row_splits,
+                           "sum")
+    assert float((outs[1].float().cpu() - ref1).abs().max()) < 0.5
+    # fused adagrad training step runs end-to-end
+    model.enable_fused_optimizer("adagrad", 0.05)
+    outs = model([x0, x1, x2, x3])
+    loss = sum(o.float().square().sum() for o in outs)
+    loss.backward()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss).item()
