@@ -555,6 +555,58 @@ void launch_scatter_unique(const int64_t* sorted_ids, const int32_t* head,
                      num_unique);
 }
 
+// Packed variant: (masked_id << 32) | position — the whole sort pipeline
+// moves ONE u64 array (digits live in bits [32, 32+log2(vocab+1))).
+__global__ void mask_oob_pack(const int64_t* __restrict__ ids, int64_t n,
+                              int64_t vocab, uint64_t* __restrict__ packed) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const int64_t id = ids[i];
+  const uint64_t m = (id < 0 || id >= vocab) ? (uint64_t)vocab : (uint64_t)id;
+  packed[i] = (m << 32) | (uint32_t)i;
+}
+
+__global__ void unpack_sorted(const uint64_t* __restrict__ packed, int64_t n,
+                              int64_t* __restrict__ sorted_ids,
+                              int32_t* __restrict__ sorted_pos) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const uint64_t p = packed[i];
+  sorted_ids[i] = (int64_t)(p >> 32);
+  sorted_pos[i] = (int32_t)(uint32_t)(p & 0xffffffffu);
+}
+
+void launch_mask_oob_pack(const int64_t* ids, int64_t n, int64_t vocab,
+                          uint64_t* packed, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(mask_oob_pack, dim3((int)cdiv64(n, block)), dim3(block),
+                     0, stream, ids, n, vocab, packed);
+}
+
+void launch_unpack_sorted(const uint64_t* packed, int64_t n,
+                          int64_t* sorted_ids, int32_t* sorted_pos,
+                          hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(unpack_sorted, dim3((int)cdiv64(n, block)), dim3(block),
+                     0, stream, packed, n, sorted_ids, sorted_pos);
+}
+
+hipError_t run_sort_keys_u64(void* temp, size_t temp_bytes,
+                             const uint64_t* keys_in, uint64_t* keys_out,
+                             int64_t n, int begin_bit, int end_bit,
+                             hipStream_t stream) {
+  return rocprim::radix_sort_keys(temp, temp_bytes, keys_in, keys_out,
+                                  (size_t)n, (unsigned)begin_bit,
+                                  (unsigned)end_bit, stream);
+}
+
+size_t rocprim_sort_keys_temp_bytes(int64_t n) {
+  size_t bytes = 0;
+  rocprim::radix_sort_keys(nullptr, bytes, (const uint64_t*)nullptr,
+                           (uint64_t*)nullptr, (size_t)n);
+  return bytes;
+}
+
 // Permute row ids (+ mean weights) by the sorted payload, widening to i64 so
 // the segmented sum can reuse the forward gather-reduce kernels directly.
 __global__ void gather_sorted(const int32_t* __restrict__ perm,
