@@ -180,3 +180,7 @@ void custom_radix_sort_keys(const uint64_t* keys_in, uint64_t* keys_out,
     to_tmp = !to_tmp;
   }
 }
+
+size_t custom_radix_sort_hist_elems(int64_t n) {
+  return (size_t)RS_RADIX * rs_cdiv(n, RS_TILE);
+}
