@@ -21,7 +21,7 @@
 
 #define WAVE 64
 #define RS_RADIX 256
-#define RS_IPT 16                      // items per thread (per lane)
+#define RS_IPT 8                       // items per thread (per lane)
 #define RS_TILE (WAVE * RS_IPT)       // elements per block
 
 static inline int64_t rs_cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
