@@ -21,8 +21,9 @@
 
 #define WAVE 64
 #define RS_RADIX 256
-#define RS_IPT 8                       // items per thread (per lane)
-#define RS_TILE (WAVE * RS_IPT)       // elements per block
+#define RS_IPT 16                      // items per thread (per lane)
+#define RS_WAVES 4
+#define RS_TILE (RS_WAVES * WAVE * RS_IPT)  // elements per block (4096)
 
 static inline int64_t rs_cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
@@ -94,35 +95,59 @@ __global__ void rs_histogram_k(const uint64_t* __restrict__ keys, int64_t n,
                                int shift, int32_t* __restrict__ hist,
                                int64_t nblocks) {
   __shared__ int32_t lhist[RS_RADIX];
-  const int lane = threadIdx.x;
-  for (int i = lane; i < RS_RADIX; i += WAVE) lhist[i] = 0;
+  const int t = threadIdx.x;
+  if (t < RS_RADIX) lhist[t] = 0;
   __syncthreads();
   const int64_t start = (int64_t)blockIdx.x * RS_TILE;
   const int64_t end = min(start + (int64_t)RS_TILE, n);
-  for (int64_t i = start + lane; i < end; i += WAVE) {
+  for (int64_t i = start + t; i < end; i += RS_WAVES * WAVE) {
     const int b = (int)((keys[i] >> shift) & 0xff);
     atomicAdd(&lhist[b], 1);
   }
   __syncthreads();
-  for (int b = lane; b < RS_RADIX; b += WAVE) {
-    hist[(int64_t)b * nblocks + blockIdx.x] = lhist[b];
-  }
+  if (t < RS_RADIX) hist[(int64_t)t * nblocks + blockIdx.x] = lhist[t];
 }
 
+// 4 waves per block; each wave owns a consecutive quarter of the tile with a
+// PRIVATE running-offset row (no barriers in the scatter loop).  Stability:
+// per-wave bases include the counts of earlier quarters.
 __global__ void rs_scatter_k(const uint64_t* __restrict__ keys_in, int64_t n,
                              int shift, const int32_t* __restrict__ offsets,
                              int64_t nblocks, uint64_t* __restrict__ keys_out) {
-  __shared__ int32_t running[RS_RADIX];
-  const int lane = threadIdx.x;
-  for (int i = lane; i < RS_RADIX; i += WAVE) {
-    running[i] = offsets[(int64_t)i * nblocks + blockIdx.x];
+  __shared__ int32_t cnt[RS_WAVES][RS_RADIX];
+  __shared__ int32_t run[RS_WAVES][RS_RADIX];
+  const int t = threadIdx.x;
+  const int wave = t >> 6;
+  const int lane = t & (WAVE - 1);
+  if (t < RS_RADIX) {
+#pragma unroll
+    for (int w = 0; w < RS_WAVES; ++w) cnt[w][t] = 0;
   }
   __syncthreads();
   const int64_t start = (int64_t)blockIdx.x * RS_TILE;
   const int64_t end = min(start + (int64_t)RS_TILE, n);
-  for (int64_t base = start; base < end; base += WAVE) {
+  const int64_t qs = min(start + (int64_t)wave * WAVE * RS_IPT, end);
+  const int64_t qe = min(qs + (int64_t)WAVE * RS_IPT, end);
+  // phase 1: per-quarter bucket counts
+  for (int64_t i = qs + lane; i < qe; i += WAVE) {
+    const int b = (int)((keys_in[i] >> shift) & 0xff);
+    atomicAdd(&cnt[wave][b], 1);
+  }
+  __syncthreads();
+  // phase 2: per-wave bases = global offset + earlier quarters' counts
+  if (t < RS_RADIX) {
+    int base = offsets[(int64_t)t * nblocks + blockIdx.x];
+#pragma unroll
+    for (int w = 0; w < RS_WAVES; ++w) {
+      run[w][t] = base;
+      base += cnt[w][t];
+    }
+  }
+  __syncthreads();
+  // phase 3: each wave scatters its quarter, 64-element groups in order
+  for (int64_t base = qs; base < qe; base += WAVE) {
     const int64_t i = base + lane;
-    const bool active = i < end;
+    const bool active = i < qe;
     const uint64_t key = active ? keys_in[i] : 0;
     const int b = (int)((key >> shift) & 0xff);
     uint64_t same = ~0ull;
@@ -138,15 +163,13 @@ __global__ void rs_scatter_k(const uint64_t* __restrict__ keys_in, int64_t n,
     const int leader = __ffsll((unsigned long long)same) - 1;
     int base_off = 0;
     if (active && lane == leader) {
-      base_off = running[b];
-      running[b] += __popcll(same);
+      base_off = run[wave][b];
+      run[wave][b] += __popcll(same);
     }
-    __syncthreads();
     if (active) {
       const int bo = __shfl(base_off, leader);
       keys_out[bo + rank] = key;
     }
-    __syncthreads();
   }
 }
 
@@ -166,16 +189,18 @@ void custom_radix_sort_keys(const uint64_t* keys_in, uint64_t* keys_out,
   for (int p = 0; p < passes; ++p) {
     kout = to_tmp ? keys_tmp : keys_out;
     const int shift = begin_bit + p * 8;
-    hipLaunchKernelGGL(rs_histogram_k, dim3((int)nblocks), dim3(WAVE), 0,
-                       stream, kin, n, shift, hist, nblocks);
+    hipLaunchKernelGGL(rs_histogram_k, dim3((int)nblocks),
+                       dim3(RS_WAVES * WAVE), 0, stream, kin, n, shift, hist,
+                       nblocks);
     hipLaunchKernelGGL(rs_scan_partials, dim3(nb_scan), dim3(WAVE), 0, stream,
                        hist, m, scan_sums, hist, chunk);
     hipLaunchKernelGGL(rs_scan_top, dim3(1), dim3(WAVE), 0, stream, scan_sums,
                        nb_scan);
     hipLaunchKernelGGL(rs_scan_addback, dim3(nb_scan), dim3(256), 0, stream,
                        hist, m, scan_sums, chunk);
-    hipLaunchKernelGGL(rs_scatter_k, dim3((int)nblocks), dim3(WAVE), 0, stream,
-                       kin, n, shift, hist, nblocks, kout);
+    hipLaunchKernelGGL(rs_scatter_k, dim3((int)nblocks),
+                       dim3(RS_WAVES * WAVE), 0, stream, kin, n, shift, hist,
+                       nblocks, kout);
     kin = kout;
     to_tmp = !to_tmp;
   }
