@@ -35,7 +35,8 @@ __global__ void dot_interact_fwd(const __hip_bfloat16* __restrict__ feats,
   extern __shared__ short lds_all[];
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & (WAVE - 1);
-  short* lds = lds_all + wave * FMAX * D;  // [FMAX][D] bf16 (as short)
+  const int ldst = D + 8;  // +16B row pad: avoids 256B-row bank conflicts
+  short* lds = lds_all + wave * FMAX * ldst;  // [FMAX][D+8] bf16 (as short)
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
   const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
 
@@ -45,12 +46,13 @@ __global__ void dot_interact_fwd(const __hip_bfloat16* __restrict__ feats,
     const int total8 = FMAX * D / 8;
     for (int i = lane; i < total8; i += WAVE) {
       const int elem = i * 8;
+      const int row = elem / D, col = elem % D;
       if (elem < F * D) {
-        *reinterpret_cast<bf16x8*>(&lds[elem]) =
+        *reinterpret_cast<bf16x8*>(&lds[row * ldst + col]) =
             *reinterpret_cast<const bf16x8*>(&src[elem]);
       } else {
         bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
-        *reinterpret_cast<bf16x8*>(&lds[elem]) = z;
+        *reinterpret_cast<bf16x8*>(&lds[row * ldst + col]) = z;
       }
     }
     // per-wave LDS: hipcc inserts the lgkm waits for its own ds ops
@@ -68,9 +70,9 @@ __global__ void dot_interact_fwd(const __hip_bfloat16* __restrict__ feats,
       for (int k0 = 0; k0 < D; k0 += 32) {
         // a_frag: rows of tile mi; b_frag: rows of tile ni (B = A^T)
         bf16x8 a = *reinterpret_cast<const bf16x8*>(
-            &lds[(mi * 16 + r16) * D + k0 + khalf * 8]);
+            &lds[(mi * 16 + r16) * ldst + k0 + khalf * 8]);
         bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
-            &lds[(ni * 16 + r16) * D + k0 + khalf * 8]);
+            &lds[(ni * 16 + r16) * ldst + k0 + khalf * 8]);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
       }
       // scatter lower-triangle entries
@@ -179,7 +181,7 @@ void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,
   const int waves = block / WAVE;
   int64_t blocks = (B + waves - 1) / waves;
   if (blocks > 8192) blocks = 8192;
-  const size_t lds = (size_t)waves * 32 * D * sizeof(short);
+  const size_t lds = (size_t)waves * 32 * (D + 8) * sizeof(short);
   hipLaunchKernelGGL((dot_interact_fwd<32>), dim3((int)blocks), dim3(block),
                      lds, stream, (const __hip_bfloat16*)feats,
                      (__hip_bfloat16*)out, B, F, D, out_w, tri_n);
