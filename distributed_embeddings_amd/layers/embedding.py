@@ -16,7 +16,7 @@ import torch
 from torch import nn
 
 from ..ops.embedding_lookup import (Ragged, csr_lookup_fused_optimizer,
-                                    csr_lookup_fused_sgd, embedding_lookup)
+                                    embedding_lookup)
 
 
 def _default_init(weight: torch.Tensor) -> None:

@@ -18,7 +18,7 @@ Semantics:
 * per-value frequency counts are maintained (``counts`` buffer).
 """
 
-from typing import List, Optional
+from typing import List
 
 import torch
 from torch import nn

@@ -11,7 +11,6 @@ pure-PyTorch reference paths are used and the extension is not required.
 import importlib
 import os
 
-import torch
 
 _ext = None
 _tried = False

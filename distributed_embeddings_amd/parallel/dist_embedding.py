@@ -34,7 +34,7 @@ from torch import nn
 from ..layers.embedding import Embedding
 from ..ops.embedding_lookup import Ragged, embedding_lookup
 from . import comm
-from .strategy import ConcatGroup, DistEmbeddingStrategy, TableConfig
+from .strategy import DistEmbeddingStrategy, TableConfig
 
 
 

@@ -14,7 +14,7 @@ allreduce uses 2 of the 7 point-to-point links, so buckets are sized large
 the env level.
 """
 
-from typing import Iterable, List
+from typing import List
 
 import torch
 import torch.distributed as dist

@@ -13,8 +13,6 @@ the reference examples' per-variable optimizer usage (SGD in
 ``examples/benchmarks/synthetic_models/main.py``).
 """
 
-from typing import Iterable, Optional
-
 import torch
 
 from ..ops import _backend

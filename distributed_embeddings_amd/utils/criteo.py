@@ -10,7 +10,7 @@ features.  ``SyntheticDLRMData`` is the DummyDataset analog
 """
 
 import os
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 import numpy as np
 import torch
