@@ -106,6 +106,10 @@ class Embedding(nn.Module):
             self.register_buffer("_fused_state",
                                  torch.empty(0, dtype=torch.float32,
                                              device=self.weight.device))
+        # optimizer state is shard-local: never broadcast/allreduce it (the
+        # weight itself carries de_local when model-parallel)
+        self._fused_state.de_local = getattr(self.weight, "de_local", False)
+        self._fused_lr.de_local = False  # lr is global; broadcasting is fine
         return self
 
     def set_fused_lr(self, lr: float):

@@ -42,10 +42,12 @@ def broadcast_parameters(module_or_params, root: int = 0):
         params, buffers = list(module_or_params), []
     with torch.no_grad():
         for p in params:
-            if not is_local_param(p):
+            if not is_local_param(p) and p.numel() > 0:
                 dist.broadcast(p.data, src=root)
         for b in buffers:
-            if b.dtype.is_floating_point or b.dtype in (torch.int32, torch.int64):
+            if (b.numel() > 0 and not is_local_param(b)
+                    and (b.dtype.is_floating_point
+                         or b.dtype in (torch.int32, torch.int64))):
                 dist.broadcast(b.data, src=root)
 
 
