@@ -139,7 +139,8 @@ def main():
         return loss
 
     graph = None
-    if args.graph and world == 1:
+    graph_ok = world == 1 or os.environ.get("DE_DIST_GRAPH") == "1"
+    if args.graph and graph_ok:
         # hipGraph capture: static input buffers (one flat cat copy per step),
         # grads pre-materialized, fused SGD has no host syncs.
         s_num, s_flat, s_labels = (t.clone() for t in pool[0])
