@@ -51,6 +51,7 @@ class SparseEmbeddingOptimizer(torch.optim.Optimizer):
             lr = group["lr"]
             eps = group["eps"]
             adagrad = group["method"] == "adagrad"
+            dense = []
             for p in group["params"]:
                 g = p.grad
                 if g is None:
@@ -59,7 +60,9 @@ class SparseEmbeddingOptimizer(torch.optim.Optimizer):
                     self._sparse_update(p, g, lr, eps, adagrad,
                                         group["assume_coalesced"])
                 else:
-                    self._dense_update(p, g, lr, eps, adagrad)
+                    dense.append(p)
+            if dense:
+                self._dense_update_batch(dense, lr, eps, adagrad)
         return loss
 
     def _state_for(self, p):
@@ -87,10 +90,15 @@ class SparseEmbeddingOptimizer(torch.optim.Optimizer):
             else:
                 p.data.index_add_(0, ids, -lr * vals)
 
-    def _dense_update(self, p, g, lr, eps, adagrad):
+    def _dense_update_batch(self, params, lr, eps, adagrad):
+        """foreach-batched dense updates (one fused launch set per step)."""
+        grads = [p.grad for p in params]
         if adagrad:
-            state = self._state_for(p)
-            state.addcmul_(g, g, value=1.0)
-            p.data.addcdiv_(g, state.sqrt().add_(eps), value=-lr)
+            states = [self._state_for(p) for p in params]
+            torch._foreach_addcmul_(states, grads, grads, value=1.0)
+            denoms = torch._foreach_sqrt(states)
+            torch._foreach_add_(denoms, eps)
+            torch._foreach_addcdiv_([p.data for p in params], grads, denoms,
+                                    value=-lr)
         else:
-            p.data.add_(g, alpha=-lr)
+            torch._foreach_add_([p.data for p in params], grads, alpha=-lr)

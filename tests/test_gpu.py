@@ -388,3 +388,17 @@ def test_distributed_embedding_module_gpu_world1():
     loss.backward()
     torch.cuda.synchronize()
     assert torch.isfinite(loss).item()
+
+
+@requires_gpu
+def test_sparse_coo_input_gpu():
+    """torch sparse-COO id input path on GPU (row_to_split + CSR kernel)."""
+    from distributed_embeddings_amd import embedding_lookup
+    w = torch.randn(20, 16, device="cuda")
+    indices = torch.tensor([[0, 0], [0, 1], [2, 0]], device="cuda").t()
+    vals = torch.tensor([4, 7, 9], device="cuda")
+    sp = torch.sparse_coo_tensor(indices, vals, (3, 2))
+    out = embedding_lookup(w, sp, "sum")
+    assert torch.allclose(out[0], w[4] + w[7], atol=1e-5)
+    assert torch.equal(out[1].cpu(), torch.zeros(16))
+    assert torch.allclose(out[2], w[9], atol=1e-5)
