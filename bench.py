@@ -79,9 +79,30 @@ def build_model(args, device):
     return model, table_sizes, hotness, num_numerical, name, keep_hot
 
 
+def maybe_enable_tunableop():
+    """Load pre-tuned hipBLASLt GEMM algorithm selections (read-only).
+
+    profiles/tunableop_gfx950.csv was produced by a PYTORCH_TUNABLEOP_TUNING=1
+    run of this bench on an MI355X; validators (torch/hipblaslt versions) are
+    checked by TunableOp itself — on mismatch it silently falls back to
+    default algorithm selection.
+    """
+    try:
+        import torch.cuda.tunable as tunable
+        csv = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                           "profiles", "tunableop_gfx950.csv")
+        if os.path.exists(csv):
+            tunable.enable(True)
+            tunable.tuning_enable(False)
+            tunable.read_file(csv)
+    except Exception as e:
+        print(f"# tunableop unavailable: {e}")
+
+
 def main():
     args = parse_args()
     rank, world, local_rank = setup_dist(args)
+    maybe_enable_tunableop()
     device = torch.device("cuda", local_rank)
     torch.manual_seed(1234 + rank)
 
