@@ -64,8 +64,22 @@ def auc(scores: torch.Tensor, labels: torch.Tensor) -> float:
     return float((ranks[pos].sum() - n_pos * (n_pos + 1) / 2) / (n_pos * n_neg))
 
 
+def maybe_enable_tunableop():
+    try:
+        import torch.cuda.tunable as tunable
+        csv = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                           "profiles", "tunableop_gfx950.csv")
+        if os.path.exists(csv) and torch.cuda.is_available():
+            tunable.enable(True)
+            tunable.tuning_enable(False)
+            tunable.read_file(csv)
+    except Exception:
+        pass
+
+
 def main():
     args = parse_args()
+    maybe_enable_tunableop()
     if "RANK" in os.environ and int(os.environ.get("WORLD_SIZE", "1")) > 1:
         torch.distributed.init_process_group("nccl" if torch.cuda.is_available() else "gloo")
         local_rank = int(os.environ.get("LOCAL_RANK", 0))

@@ -42,8 +42,22 @@ def parse_args():
     return p.parse_args()
 
 
+def maybe_enable_tunableop():
+    try:
+        import torch.cuda.tunable as tunable
+        csv = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                           "profiles", "tunableop_gfx950.csv")
+        if os.path.exists(csv) and torch.cuda.is_available():
+            tunable.enable(True)
+            tunable.tuning_enable(False)
+            tunable.read_file(csv)
+    except Exception:
+        pass
+
+
 def main():
     args = parse_args()
+    maybe_enable_tunableop()
     if "RANK" in os.environ and int(os.environ.get("WORLD_SIZE", "1")) > 1:
         torch.distributed.init_process_group(
             "nccl" if torch.cuda.is_available() else "gloo")
