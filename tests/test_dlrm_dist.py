@@ -95,3 +95,24 @@ def test_output_dtype_bf16_a2a_world2():
     results = run_distributed(_dtype_worker, world=2)
     for errs in results:
         assert max(errs) < 0.05  # bf16 rounding only
+
+
+def _bcast_callback_worker(rank, world):
+    import distributed_embeddings_amd as de
+    m = torch.nn.Linear(4, 4)
+    with torch.no_grad():
+        m.weight.fill_(float(rank))
+    cb = de.BroadcastParametersOnFirstStep(m)
+    cb()  # first step: broadcast from rank 0
+    first = m.weight.detach().clone()
+    with torch.no_grad():
+        m.weight.fill_(float(rank) + 10)
+    cb()  # later steps: no-op
+    return {"first": first, "after": m.weight.detach().clone()}
+
+
+def test_broadcast_on_first_step_world2():
+    results = run_distributed(_bcast_callback_worker, world=2)
+    for r in range(2):
+        assert torch.equal(results[r]["first"], torch.zeros(4, 4))
+        assert float(results[r]["after"][0, 0]) == r + 10

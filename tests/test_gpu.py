@@ -402,3 +402,32 @@ def test_sparse_coo_input_gpu():
     assert torch.allclose(out[0], w[4] + w[7], atol=1e-5)
     assert torch.equal(out[1].cpu(), torch.zeros(16))
     assert torch.allclose(out[2], w[9], atol=1e-5)
+
+
+@requires_gpu
+def test_hipgraph_capture_fused_train_step():
+    """The fused-optimizer train step must be hipGraph-capturable (no host
+    syncs anywhere in fwd+bwd)."""
+    import distributed_embeddings_amd as de
+    torch.manual_seed(23)
+    with torch.device("cuda"):
+        model = de.DistributedEmbedding(
+            [de.TableConfig(5000, 64), de.TableConfig(300, 64)])
+    model.enable_fused_optimizer("sgd", 0.01)
+    ids0 = torch.randint(0, 5000, (256,), device="cuda")
+    ids1 = torch.randint(0, 300, (256,), device="cuda")
+    def step():
+        outs = model([ids0, ids1], output_dtype=torch.bfloat16)
+        loss = sum(o.float().square().sum() for o in outs)
+        loss.backward()
+    for _ in range(3):
+        step()
+    torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        step()
+    w_before = model.col_layers[0].weight.detach().clone()
+    g.replay()
+    torch.cuda.synchronize()
+    assert not torch.equal(w_before, model.col_layers[0].weight), \
+        "replay did not update weights"
