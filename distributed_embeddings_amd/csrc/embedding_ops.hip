@@ -45,9 +45,9 @@ static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 //   segments <= LONG_T reduce in registers and store directly (no atomics).
 //   Longer rows (power-law mega-segments, tiny-vocab backward) are zero-
 //   filled and pushed to a device-side long-row list.
-//   Kernel B (csr_fwd_long_*) grid-strides waves over (long row, chunk)
-//   pairs, each wave reducing one LONG_T-sized chunk and combining with
-//   atomicAdd.  No host sync, no full-output memset.
+//   Kernel B (csr_fwd_long) consumes an exact device-built (long row,
+//   LONG_T-chunk) work list (expand_long_work), one wave per chunk, and
+//   combines partials with atomicAdd.  No host sync, no full-output memset.
 // This is the wave64 answer to the reference's blockDim.y reduction
 // splitting + round-robin step counter (embedding_lookup_kernels.cu:195-226).
 #define LONG_T 128
