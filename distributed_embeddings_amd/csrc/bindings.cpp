@@ -80,20 +80,25 @@ torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
                                  torch::Tensor row_splits, bool mean) {
   CHECK_CUDA(params); CHECK_CUDA(values); CHECK_CUDA(row_splits);
   CHECK_CONTIG(params); CHECK_CONTIG(values); CHECK_CONTIG(row_splits);
-  TORCH_CHECK(params.dtype() == torch::kFloat32, "params must be fp32");
+  TORCH_CHECK(params.dtype() == torch::kFloat32 ||
+                  params.dtype() == torch::kBFloat16,
+              "params must be fp32 or bf16");
   TORCH_CHECK(values.dtype() == torch::kInt64, "values must be int64");
   TORCH_CHECK(row_splits.dtype() == torch::kInt64, "row_splits must be int64");
   const int64_t num_rows = row_splits.numel() - 1;
   const int64_t vocab = params.size(0);
   const int width = (int)params.size(1);
-  auto out = torch::empty({num_rows, width}, params.options());
+  // output accumulates in fp32 regardless of the table storage dtype
+  auto out = torch::empty({num_rows, width},
+                          params.options().dtype(torch::kFloat32));
   if (num_rows > 0) {
     const int64_t nnz_in = values.numel();
     auto long_rows = torch::empty({num_rows}, values.options());
     auto long_count = torch::empty({1}, values.options().dtype(torch::kInt32));
     auto work_items = torch::empty({nnz_in / 64 + 64}, values.options());
     auto n_work = torch::empty({1}, values.options().dtype(torch::kInt32));
-    launch_csr_lookup_forward(params.data_ptr<float>(),
+    launch_csr_lookup_forward(params.data_ptr(),
+                              params.dtype() == torch::kBFloat16,
                               values.data_ptr<int64_t>(),
                               row_splits.data_ptr<int64_t>(), nullptr,
                               out.data_ptr<float>(), num_rows, nnz_in,
@@ -204,7 +209,7 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
     auto long_count = torch::empty({1}, i32);
     auto work_items = torch::empty({nnz / 64 + 64}, i64);
     auto n_work = torch::empty({1}, i32);
-    launch_csr_lookup_forward(grad_out.data_ptr<float>(),
+    launch_csr_lookup_forward(grad_out.data_ptr(), /*params_bf16=*/false,
                               srow.data_ptr<int64_t>(),
                               seg_tmp.data_ptr<int64_t>(), sw_ptr,
                               unique_grad.data_ptr<float>(), nu, nnz,
@@ -255,7 +260,8 @@ void sparse_row_update(torch::Tensor weight, torch::Tensor state,
                        double eps, bool adagrad) {
   CHECK_CUDA(weight); CHECK_CUDA(ids); CHECK_CUDA(grad);
   CHECK_CONTIG(weight); CHECK_CONTIG(ids); CHECK_CONTIG(grad);
-  TORCH_CHECK(weight.dtype() == torch::kFloat32);
+  TORCH_CHECK(weight.dtype() == torch::kFloat32 ||
+              weight.dtype() == torch::kBFloat16);
   TORCH_CHECK(grad.dtype() == torch::kFloat32);
   TORCH_CHECK(ids.dtype() == torch::kInt64);
   const int64_t n = ids.numel();
@@ -264,9 +270,11 @@ void sparse_row_update(torch::Tensor weight, torch::Tensor state,
   if (adagrad) {
     CHECK_CUDA(state); CHECK_CONTIG(state);
     TORCH_CHECK(state.sizes() == weight.sizes(), "adagrad state shape mismatch");
+    TORCH_CHECK(state.dtype() == torch::kFloat32, "adagrad state must be fp32");
     state_ptr = state.data_ptr<float>();
   }
-  launch_sparse_row_update(weight.data_ptr<float>(), state_ptr,
+  launch_sparse_row_update(weight.data_ptr(),
+                           weight.dtype() == torch::kBFloat16, state_ptr,
                            ids.data_ptr<int64_t>(), grad.data_ptr<float>(), n,
                            (int)weight.size(1), (float)lr, (float)eps, adagrad,
                            current_stream());
@@ -285,9 +293,11 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
   CHECK_CUDA(grad_out); CHECK_CUDA(lr);
   CHECK_CONTIG(weight); CHECK_CONTIG(values); CHECK_CONTIG(row_splits);
   CHECK_CONTIG(grad_out);
-  TORCH_CHECK(weight.dtype() == torch::kFloat32 &&
-              grad_out.dtype() == torch::kFloat32 &&
+  TORCH_CHECK(weight.dtype() == torch::kFloat32 ||
+              weight.dtype() == torch::kBFloat16);
+  TORCH_CHECK(grad_out.dtype() == torch::kFloat32 &&
               lr.dtype() == torch::kFloat32);
+  const bool wbf16 = weight.dtype() == torch::kBFloat16;
   const int64_t num_rows = row_splits.numel() - 1;
   const int64_t nnz = values.numel();
   const int64_t vocab = weight.size(0);
@@ -356,12 +366,15 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
   if (adagrad) {
     CHECK_CUDA(state); CHECK_CONTIG(state);
     TORCH_CHECK(state.sizes() == weight.sizes(), "adagrad state shape mismatch");
+    TORCH_CHECK(state.dtype() == torch::kFloat32, "adagrad state must be fp32");
     state_ptr = state.data_ptr<float>();
+  }
+  if (adagrad || wbf16) {
     scratch_rows = nnz / (128 + 1) + 1;  // max possible long segments
     scratch = torch::empty({scratch_rows, (int64_t)width}, f32);
     scratch_ptr = scratch.data_ptr<float>();
   }
-  launch_sorted_optimizer_update(weight.data_ptr<float>(), state_ptr,
+  launch_sorted_optimizer_update(weight.data_ptr(), wbf16, state_ptr,
                                  (float)eps, sorted_ids.data_ptr<int64_t>(),
                                  seg_tmp.data_ptr<int64_t>(),
                                  srow.data_ptr<int64_t>(), sw_ptr,

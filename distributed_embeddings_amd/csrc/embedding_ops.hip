@@ -20,6 +20,10 @@
 //     reference's cooperative cuCollections kernel (K8-K9 equivalent).
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstring>
+#include <type_traits>
 
 #include <rocprim/device/device_radix_sort.hpp>
 #include <rocprim/device/device_scan.hpp>
@@ -27,6 +31,47 @@
 #include "ops_api.h"
 
 #define WAVE 64
+
+// ---------------------------------------------------------------------------
+// Param dtype abstraction: tables are fp32 or bf16 (storage); all arithmetic
+// accumulates in fp32.  bf16 load = bit-shift; store = RNE via __hip_bfloat16.
+// ---------------------------------------------------------------------------
+struct bf16_t { unsigned short b; };
+
+__device__ __forceinline__ float pt_load(const float* p) { return *p; }
+__device__ __forceinline__ float pt_load(const bf16_t* p) {
+  union { unsigned int u; float f; } c;
+  c.u = ((unsigned int)p->b) << 16;
+  return c.f;
+}
+__device__ __forceinline__ void pt_store(float* p, float v) { *p = v; }
+__device__ __forceinline__ void pt_store(bf16_t* p, float v) {
+  __hip_bfloat16 h(v);
+  p->b = *reinterpret_cast<unsigned short*>(&h);
+}
+// vectorized row loads: 4 contiguous elements starting at p (aligned)
+__device__ __forceinline__ void pt_load4(const float* p, float* o) {
+  const float4 v = *reinterpret_cast<const float4*>(p);
+  o[0] = v.x; o[1] = v.y; o[2] = v.z; o[3] = v.w;
+}
+__device__ __forceinline__ void pt_load4(const bf16_t* p, float* o) {
+  const short4 v = *reinterpret_cast<const short4*>(p);
+  union { unsigned int u; float f; } c;
+  c.u = ((unsigned int)(unsigned short)v.x) << 16; o[0] = c.f;
+  c.u = ((unsigned int)(unsigned short)v.y) << 16; o[1] = c.f;
+  c.u = ((unsigned int)(unsigned short)v.z) << 16; o[2] = c.f;
+  c.u = ((unsigned int)(unsigned short)v.w) << 16; o[3] = c.f;
+}
+__device__ __forceinline__ void pt_load2(const float* p, float* o) {
+  const float2 v = *reinterpret_cast<const float2*>(p);
+  o[0] = v.x; o[1] = v.y;
+}
+__device__ __forceinline__ void pt_load2(const bf16_t* p, float* o) {
+  const short2 v = *reinterpret_cast<const short2*>(p);
+  union { unsigned int u; float f; } c;
+  c.u = ((unsigned int)(unsigned short)v.x) << 16; o[0] = c.f;
+  c.u = ((unsigned int)(unsigned short)v.y) << 16; o[1] = c.f;
+}
 
 static inline int next_pow2(int v) {
   int p = 1;
@@ -53,8 +98,8 @@ static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 #define LONG_T 128
 
 // Narrow kernel A: width <= 64.  TILE = pow2 >= width; 64/TILE rows per wave.
-template <int TILE, bool MEAN, bool HAS_W>
-__global__ void csr_fwd_narrow(const float* __restrict__ params,
+template <int TILE, bool MEAN, bool HAS_W, typename PT>
+__global__ void csr_fwd_narrow(const PT* __restrict__ params,
                                const int64_t* __restrict__ values,
                                const int64_t* __restrict__ splits,
                                const float* __restrict__ per_id_w,
@@ -86,20 +131,20 @@ __global__ void csr_fwd_narrow(const float* __restrict__ params,
       const int64_t i2 = values[k + 2], i3 = values[k + 3];
       float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
       if (i0 >= 0 && i0 < vocab)
-        a0 = (HAS_W ? per_id_w[k] : 1.f) * params[i0 * width + tl];
+        a0 = (HAS_W ? per_id_w[k] : 1.f) * pt_load(&params[i0 * width + tl]);
       if (i1 >= 0 && i1 < vocab)
-        a1 = (HAS_W ? per_id_w[k + 1] : 1.f) * params[i1 * width + tl];
+        a1 = (HAS_W ? per_id_w[k + 1] : 1.f) * pt_load(&params[i1 * width + tl]);
       if (i2 >= 0 && i2 < vocab)
-        a2 = (HAS_W ? per_id_w[k + 2] : 1.f) * params[i2 * width + tl];
+        a2 = (HAS_W ? per_id_w[k + 2] : 1.f) * pt_load(&params[i2 * width + tl]);
       if (i3 >= 0 && i3 < vocab)
-        a3 = (HAS_W ? per_id_w[k + 3] : 1.f) * params[i3 * width + tl];
+        a3 = (HAS_W ? per_id_w[k + 3] : 1.f) * pt_load(&params[i3 * width + tl]);
       acc += (a0 + a1) + (a2 + a3);
     }
     for (; k < e; ++k) {
       const int64_t idx = values[k];
       if (idx < 0 || idx >= vocab) continue;
       const float w = HAS_W ? per_id_w[k] : 1.f;
-      acc += w * params[idx * width + tl];
+      acc += w * pt_load(&params[idx * width + tl]);
     }
     if (MEAN && e > s) acc /= (float)(e - s);
     out[row * width + tl] = acc;
@@ -107,8 +152,8 @@ __global__ void csr_fwd_narrow(const float* __restrict__ params,
 }
 
 // Wide kernel A: width > 64.  One wave per row; VEC elements per lane.
-template <int VEC, bool MEAN, bool HAS_W>
-__global__ void csr_fwd_wide(const float* __restrict__ params,
+template <int VEC, bool MEAN, bool HAS_W, typename PT>
+__global__ void csr_fwd_wide(const PT* __restrict__ params,
                              const int64_t* __restrict__ values,
                              const int64_t* __restrict__ splits,
                              const float* __restrict__ per_id_w,
@@ -138,29 +183,28 @@ __global__ void csr_fwd_wide(const float* __restrict__ params,
         const int64_t idx = values[k];
         if (idx < 0 || idx >= vocab) continue;
         const float w = HAS_W ? per_id_w[k] : 1.f;
-        const float* rowp = params + idx * (int64_t)width + col0;
+        const PT* rowp = params + idx * (int64_t)width + col0;
         if (VEC == 4) {
           if (col0 + 4 <= width) {
-            const float4 r4 = *reinterpret_cast<const float4*>(rowp);
-            acc[0] += w * r4.x;
-            acc[1] += w * r4.y;
-            acc[2] += w * r4.z;
-            acc[3] += w * r4.w;
+            float r[4];
+            pt_load4(rowp, r);
+            acc[0] += w * r[0]; acc[1] += w * r[1];
+            acc[2] += w * r[2]; acc[3] += w * r[3];
           } else {
 #pragma unroll
             for (int v = 0; v < 4; ++v)
-              if (col0 + v < width) acc[v] += w * rowp[v];
+              if (col0 + v < width) acc[v] += w * pt_load(&rowp[v]);
           }
         } else if (VEC == 2) {
           if (col0 + 2 <= width) {
-            const float2 r2 = *reinterpret_cast<const float2*>(rowp);
-            acc[0] += w * r2.x;
-            acc[1] += w * r2.y;
+            float r[2];
+            pt_load2(rowp, r);
+            acc[0] += w * r[0]; acc[1] += w * r[1];
           } else {
-            if (col0 < width) acc[0] += w * rowp[0];
+            if (col0 < width) acc[0] += w * pt_load(&rowp[0]);
           }
         } else {
-          if (col0 < width) acc[0] += w * rowp[0];
+          if (col0 < width) acc[0] += w * pt_load(&rowp[0]);
         }
       }
       float* outp = out + row * (int64_t)width + col0;
@@ -205,8 +249,8 @@ __global__ void expand_long_work(const int64_t* __restrict__ long_rows,
 // Kernel B: long rows.  Work item = (long row, LONG_T-chunk); one wave each.
 // NW lanes-per-row tiling matches kernel A (TILE for narrow, full wave for
 // wide).  Partials combine with global atomicAdd (out pre-zeroed by A).
-template <int TILE, int VEC, bool MEAN, bool HAS_W>
-__global__ void csr_fwd_long(const float* __restrict__ params,
+template <int TILE, int VEC, bool MEAN, bool HAS_W, typename PT>
+__global__ void csr_fwd_long(const PT* __restrict__ params,
                              const int64_t* __restrict__ values,
                              const int64_t* __restrict__ splits,
                              const float* __restrict__ per_id_w,
@@ -247,20 +291,23 @@ __global__ void csr_fwd_long(const float* __restrict__ params,
             const int64_t i2 = values[k + 2 * NSUB], i3 = values[k + 3 * NSUB];
             float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
             if (i0 >= 0 && i0 < vocab)
-              a0 = (HAS_W ? per_id_w[k] : 1.f) * params[i0 * width + tl];
+              a0 = (HAS_W ? per_id_w[k] : 1.f) * pt_load(&params[i0 * width + tl]);
             if (i1 >= 0 && i1 < vocab)
-              a1 = (HAS_W ? per_id_w[k + NSUB] : 1.f) * params[i1 * width + tl];
+              a1 = (HAS_W ? per_id_w[k + NSUB] : 1.f) *
+                   pt_load(&params[i1 * width + tl]);
             if (i2 >= 0 && i2 < vocab)
-              a2 = (HAS_W ? per_id_w[k + 2 * NSUB] : 1.f) * params[i2 * width + tl];
+              a2 = (HAS_W ? per_id_w[k + 2 * NSUB] : 1.f) *
+                   pt_load(&params[i2 * width + tl]);
             if (i3 >= 0 && i3 < vocab)
-              a3 = (HAS_W ? per_id_w[k + 3 * NSUB] : 1.f) * params[i3 * width + tl];
+              a3 = (HAS_W ? per_id_w[k + 3 * NSUB] : 1.f) *
+                   pt_load(&params[i3 * width + tl]);
             acc += (a0 + a1) + (a2 + a3);
           }
           for (; k < ke; k += NSUB) {
             const int64_t idx = values[k];
             if (idx < 0 || idx >= vocab) continue;
             const float w = HAS_W ? per_id_w[k] : 1.f;
-            acc += w * params[idx * width + tl];
+            acc += w * pt_load(&params[idx * width + tl]);
           }
         }
       }
@@ -286,15 +333,16 @@ __global__ void csr_fwd_long(const float* __restrict__ params,
             const int64_t idx = values[k];
             if (idx < 0 || idx >= vocab) continue;
             const float w = HAS_W ? per_id_w[k] : 1.f;
-            const float* rowp = params + idx * (int64_t)width + col0;
+            const PT* rowp = params + idx * (int64_t)width + col0;
             if (V == 4 && col0 + 4 <= width) {
-              const float4 r4 = *reinterpret_cast<const float4*>(rowp);
-              acc[0] += w * r4.x; acc[1] += w * r4.y;
-              acc[2] += w * r4.z; acc[3] += w * r4.w;
+              float r[4];
+              pt_load4(rowp, r);
+              acc[0] += w * r[0]; acc[1] += w * r[1];
+              acc[2] += w * r[2]; acc[3] += w * r[3];
             } else {
 #pragma unroll
               for (int v = 0; v < V; ++v)
-                if (col0 + v < width) acc[v] += w * rowp[v];
+                if (col0 + v < width) acc[v] += w * pt_load(&rowp[v]);
             }
           }
         }
@@ -315,8 +363,8 @@ static int pick_grid(int64_t work_items, int block_waves) {
   return (int)blocks;
 }
 
-template <int TILE, int VEC>
-static void launch_csr_pair(const float* params, const int64_t* values,
+template <int TILE, int VEC, typename PT>
+static void launch_csr_pair(const PT* params, const int64_t* values,
                             const int64_t* splits, const float* per_id_w,
                             float* out, int64_t num_rows, int64_t nnz,
                             int64_t vocab, int width, bool mean,
@@ -341,12 +389,12 @@ static void launch_csr_pair(const float* params, const int64_t* values,
 #define LA(MEAN, HASW)                                                         \
   do {                                                                         \
     if constexpr (TILE > 0)                                                    \
-      hipLaunchKernelGGL((csr_fwd_narrow<(TILE > 0 ? TILE : 1), MEAN, HASW>),  \
-                         grid, dim3(block), 0, stream, params, values, splits, \
-                         per_id_w, out, num_rows, vocab, width, long_thresh,   \
-                         long_rows, long_count);                               \
+      hipLaunchKernelGGL(                                                      \
+          (csr_fwd_narrow<(TILE > 0 ? TILE : 1), MEAN, HASW, PT>), grid,       \
+          dim3(block), 0, stream, params, values, splits, per_id_w, out,       \
+          num_rows, vocab, width, long_thresh, long_rows, long_count);         \
     else                                                                       \
-      hipLaunchKernelGGL((csr_fwd_wide<(VEC > 0 ? VEC : 1), MEAN, HASW>),      \
+      hipLaunchKernelGGL((csr_fwd_wide<(VEC > 0 ? VEC : 1), MEAN, HASW, PT>),  \
                          grid, dim3(block), 0, stream, params, values, splits, \
                          per_id_w, out, num_rows, vocab, width, long_thresh,   \
                          long_rows, long_count);                               \
@@ -355,7 +403,7 @@ static void launch_csr_pair(const float* params, const int64_t* values,
   do {                                                                         \
     hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,    \
                        long_rows, long_count, splits, work_items, n_work);     \
-    hipLaunchKernelGGL((csr_fwd_long<TILE, VEC, MEAN, HASW>), dim3(2048),      \
+    hipLaunchKernelGGL((csr_fwd_long<TILE, VEC, MEAN, HASW, PT>), dim3(2048),  \
                        dim3(block), 0, stream, params, values, splits,         \
                        per_id_w, out, vocab, width, long_rows, work_items,     \
                        n_work);                                                \
@@ -371,33 +419,53 @@ static void launch_csr_pair(const float* params, const int64_t* values,
 #undef LB
 }
 
-void launch_csr_lookup_forward(const float* params, const int64_t* values,
-                               const int64_t* splits, const float* per_id_w,
-                               float* out, int64_t num_rows, int64_t nnz,
-                               int64_t vocab, int width, bool mean,
-                               int64_t* long_rows, int32_t* long_count,
-                               int64_t* work_items, int32_t* n_work,
-                               hipStream_t stream) {
+template <typename PT>
+static void launch_csr_lookup_forward_t(const PT* params, const int64_t* values,
+                                        const int64_t* splits,
+                                        const float* per_id_w, float* out,
+                                        int64_t num_rows, int64_t nnz,
+                                        int64_t vocab, int width, bool mean,
+                                        int64_t* long_rows, int32_t* long_count,
+                                        int64_t* work_items, int32_t* n_work,
+                                        hipStream_t stream) {
 #define ARGS params, values, splits, per_id_w, out, num_rows, nnz, vocab, \
              width, mean, long_rows, long_count, work_items, n_work, stream
   if (width <= 64) {
     switch (next_pow2(width)) {
-      case 1: launch_csr_pair<1, 0>(ARGS); break;
-      case 2: launch_csr_pair<2, 0>(ARGS); break;
-      case 4: launch_csr_pair<4, 0>(ARGS); break;
-      case 8: launch_csr_pair<8, 0>(ARGS); break;
-      case 16: launch_csr_pair<16, 0>(ARGS); break;
-      case 32: launch_csr_pair<32, 0>(ARGS); break;
-      default: launch_csr_pair<64, 0>(ARGS); break;
+      case 1: launch_csr_pair<1, 0, PT>(ARGS); break;
+      case 2: launch_csr_pair<2, 0, PT>(ARGS); break;
+      case 4: launch_csr_pair<4, 0, PT>(ARGS); break;
+      case 8: launch_csr_pair<8, 0, PT>(ARGS); break;
+      case 16: launch_csr_pair<16, 0, PT>(ARGS); break;
+      case 32: launch_csr_pair<32, 0, PT>(ARGS); break;
+      default: launch_csr_pair<64, 0, PT>(ARGS); break;
     }
   } else if (width % 4 == 0) {
-    launch_csr_pair<0, 4>(ARGS);
+    launch_csr_pair<0, 4, PT>(ARGS);
   } else if (width % 2 == 0) {
-    launch_csr_pair<0, 2>(ARGS);
+    launch_csr_pair<0, 2, PT>(ARGS);
   } else {
-    launch_csr_pair<0, 1>(ARGS);
+    launch_csr_pair<0, 1, PT>(ARGS);
   }
 #undef ARGS
+}
+
+void launch_csr_lookup_forward(const void* params, bool params_bf16,
+                               const int64_t* values, const int64_t* splits,
+                               const float* per_id_w, float* out,
+                               int64_t num_rows, int64_t nnz, int64_t vocab,
+                               int width, bool mean, int64_t* long_rows,
+                               int32_t* long_count, int64_t* work_items,
+                               int32_t* n_work, hipStream_t stream) {
+  if (params_bf16) {
+    launch_csr_lookup_forward_t<bf16_t>(
+        (const bf16_t*)params, values, splits, per_id_w, out, num_rows, nnz,
+        vocab, width, mean, long_rows, long_count, work_items, n_work, stream);
+  } else {
+    launch_csr_lookup_forward_t<float>(
+        (const float*)params, values, splits, per_id_w, out, num_rows, nnz,
+        vocab, width, mean, long_rows, long_count, work_items, n_work, stream);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -705,8 +773,8 @@ __device__ __forceinline__ void seg_grad_reduce(
 // Short segments: one wave per segment, direct (non-atomic) update.
 // Grid-strides only over the REAL segment count (*nu_ptr), not the padded
 // nnz-sized buffer.
-template <int TILE, int VEC, bool HAS_W, bool ADAGRAD>
-__global__ void sorted_opt_update(float* __restrict__ weight,
+template <int TILE, int VEC, bool HAS_W, bool ADAGRAD, typename PT>
+__global__ void sorted_opt_update(PT* __restrict__ weight,
                                   float* __restrict__ state, float eps,
                                   const int64_t* __restrict__ sorted_ids,
                                   const int64_t* __restrict__ seg,
@@ -742,9 +810,9 @@ __global__ void sorted_opt_update(float* __restrict__ weight,
           const float g = acc[0];
           const float st = state[o] + g * g;
           state[o] = st;
-          weight[o] -= lr * g / (sqrtf(st) + eps);
+          pt_store(&weight[o], pt_load(&weight[o]) - lr * g / (sqrtf(st) + eps));
         } else {
-          weight[o] -= lr * acc[0];
+          pt_store(&weight[o], pt_load(&weight[o]) - lr * acc[0]);
         }
       }
     } else {
@@ -770,21 +838,23 @@ __global__ void sorted_opt_update(float* __restrict__ weight,
         }
         if (ADAGRAD) {
           float* sp = state + uid * (int64_t)width;
-          float* wp = weight + uid * (int64_t)width;
+          PT* wp = weight + uid * (int64_t)width;
 #pragma unroll
           for (int v = 0; v < V; ++v) {
             if (col0 + v < width) {
               const float g = acc[v];
               const float st = sp[col0 + v] + g * g;
               sp[col0 + v] = st;
-              wp[col0 + v] -= lr * g / (sqrtf(st) + eps);
+              pt_store(&wp[col0 + v],
+                       pt_load(&wp[col0 + v]) - lr * g / (sqrtf(st) + eps));
             }
           }
         } else {
-          float* wp = weight + uid * (int64_t)width;
+          PT* wp = weight + uid * (int64_t)width;
 #pragma unroll
           for (int v = 0; v < V; ++v)
-            if (col0 + v < width) wp[v + col0] -= lr * acc[v];
+            if (col0 + v < width)
+              pt_store(&wp[v + col0], pt_load(&wp[v + col0]) - lr * acc[v]);
         }
       }
     }
@@ -859,8 +929,9 @@ __global__ void sorted_opt_long(float* __restrict__ target,  // weight or scratc
   }
 }
 
-__global__ void sorted_adagrad_long_finalize(
-    float* __restrict__ weight, float* __restrict__ state, float eps,
+template <bool ADAGRAD, typename PT>
+__global__ void sorted_long_finalize(
+    PT* __restrict__ weight, float* __restrict__ state, float eps,
     const int64_t* __restrict__ sorted_ids, const int64_t* __restrict__ seg,
     const float* __restrict__ lr_ptr, int width,
     const int64_t* __restrict__ long_rows,
@@ -876,17 +947,21 @@ __global__ void sorted_adagrad_long_finalize(
     for (int c = lane; c < width; c += WAVE) {
       const float g = scratch[li * (int64_t)width + c];
       const int64_t o = uid * (int64_t)width + c;
-      const float s = state[o] + g * g;
-      state[o] = s;
-      weight[o] -= lr * g / (sqrtf(s) + eps);
+      if (ADAGRAD) {
+        const float s = state[o] + g * g;
+        state[o] = s;
+        pt_store(&weight[o], pt_load(&weight[o]) - lr * g / (sqrtf(s) + eps));
+      } else {
+        pt_store(&weight[o], pt_load(&weight[o]) - lr * g);
+      }
     }
   }
 }
 
 
 
-template <int TILE, int VEC>
-static void launch_sorted_opt_pair(float* weight, float* state, float eps,
+template <int TILE, int VEC, typename PT>
+static void launch_sorted_opt_pair(PT* weight, float* state, float eps,
                                    const int64_t* sorted_ids,
                                    const int64_t* seg, const int64_t* srow,
                                    const float* sw, const float* grad_out,
@@ -900,16 +975,29 @@ static void launch_sorted_opt_pair(float* weight, float* state, float eps,
   const int64_t row_waves =
       TILE > 0 ? cdiv64(max_segs, WAVE / (TILE > 0 ? TILE : 1)) : max_segs;
   const int grid = pick_grid(row_waves, block / WAVE);
+  // bf16 weights have no atomicAdd: their long-SGD path also goes through the
+  // fp32 scratch + finalize pair.
+  constexpr bool PT_F32 = std::is_same<PT, float>::value;
+  const bool use_scratch = adagrad || !PT_F32;
 #define SU(HASW, ADA)                                                          \
-  hipLaunchKernelGGL((sorted_opt_update<TILE, VEC, HASW, ADA>), dim3(grid),    \
-                     dim3(block), 0, stream, weight, state, eps, sorted_ids,   \
-                     seg, srow, sw, grad_out, lr, nu_ptr, (int64_t)LONG_T,     \
-                     width, long_rows, long_count)
-#define SL(HASW, SCR)                                                          \
-  hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, SCR>), dim3(2048),      \
-                     dim3(block), 0, stream, SCR ? long_scratch : weight,      \
-                     sorted_ids, seg, srow, sw, grad_out, lr, width,           \
-                     long_rows, work_items, n_work)
+  hipLaunchKernelGGL((sorted_opt_update<TILE, VEC, HASW, ADA, PT>),            \
+                     dim3(grid), dim3(block), 0, stream, weight, state, eps,   \
+                     sorted_ids, seg, srow, sw, grad_out, lr, nu_ptr,          \
+                     (int64_t)LONG_T, width, long_rows, long_count)
+#define SL_SCRATCH(HASW)                                                       \
+  hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, true>), dim3(2048),     \
+                     dim3(block), 0, stream, long_scratch, sorted_ids, seg,    \
+                     srow, sw, grad_out, lr, width, long_rows, work_items,     \
+                     n_work)
+#define SL_DIRECT(HASW)                                                        \
+  hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, false>), dim3(2048),    \
+                     dim3(block), 0, stream, (float*)weight, sorted_ids, seg,  \
+                     srow, sw, grad_out, lr, width, long_rows, work_items,     \
+                     n_work)
+#define FIN(ADA)                                                               \
+  hipLaunchKernelGGL((sorted_long_finalize<ADA, PT>), dim3(256), dim3(block),  \
+                     0, stream, weight, state, eps, sorted_ids, seg, lr,       \
+                     width, long_rows, long_count, long_scratch)
   if (adagrad) {
     if (sw) SU(true, true); else SU(false, true);
   } else {
@@ -917,19 +1005,51 @@ static void launch_sorted_opt_pair(float* weight, float* state, float eps,
   }
   hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,
                      long_rows, long_count, seg, work_items, n_work);
-  if (adagrad) {
-    if (sw) SL(true, true); else SL(false, true);
-    hipLaunchKernelGGL(sorted_adagrad_long_finalize, dim3(256), dim3(block), 0,
-                       stream, weight, state, eps, sorted_ids, seg, lr, width,
-                       long_rows, long_count, long_scratch);
+  if (use_scratch) {
+    if (sw) SL_SCRATCH(true); else SL_SCRATCH(false);
+    if (adagrad) FIN(true); else FIN(false);
   } else {
-    if (sw) SL(true, false); else SL(false, false);
+    if (sw) SL_DIRECT(true); else SL_DIRECT(false);
   }
 #undef SU
-#undef SL
+#undef SL_SCRATCH
+#undef SL_DIRECT
+#undef FIN
 }
 
-void launch_sorted_optimizer_update(float* weight, float* state, float eps,
+template <typename PT>
+static void launch_sorted_optimizer_update_t(
+    PT* weight, float* state, float eps, const int64_t* sorted_ids,
+    const int64_t* seg, const int64_t* srow, const float* sw,
+    const float* grad_out, const float* lr, const int32_t* nu_ptr,
+    int64_t max_segs, int width, int64_t* long_rows, int32_t* long_count,
+    int64_t* work_items, int32_t* n_work, float* long_scratch, bool adagrad,
+    hipStream_t stream) {
+#define ARGS weight, state, eps, sorted_ids, seg, srow, sw, grad_out, lr,    \
+             nu_ptr, max_segs, width, long_rows, long_count, work_items,     \
+             n_work, long_scratch, adagrad, stream
+  if (width <= 64) {
+    switch (next_pow2(width)) {
+      case 1: launch_sorted_opt_pair<1, 0, PT>(ARGS); break;
+      case 2: launch_sorted_opt_pair<2, 0, PT>(ARGS); break;
+      case 4: launch_sorted_opt_pair<4, 0, PT>(ARGS); break;
+      case 8: launch_sorted_opt_pair<8, 0, PT>(ARGS); break;
+      case 16: launch_sorted_opt_pair<16, 0, PT>(ARGS); break;
+      case 32: launch_sorted_opt_pair<32, 0, PT>(ARGS); break;
+      default: launch_sorted_opt_pair<64, 0, PT>(ARGS); break;
+    }
+  } else if (width % 4 == 0) {
+    launch_sorted_opt_pair<0, 4, PT>(ARGS);
+  } else if (width % 2 == 0) {
+    launch_sorted_opt_pair<0, 2, PT>(ARGS);
+  } else {
+    launch_sorted_opt_pair<0, 1, PT>(ARGS);
+  }
+#undef ARGS
+}
+
+void launch_sorted_optimizer_update(void* weight, bool weight_bf16,
+                                    float* state, float eps,
                                     const int64_t* sorted_ids,
                                     const int64_t* seg, const int64_t* srow,
                                     const float* sw, const float* grad_out,
@@ -941,31 +1061,21 @@ void launch_sorted_optimizer_update(float* weight, float* state, float eps,
                                     bool adagrad, hipStream_t stream) {
   hipMemsetAsync(long_count, 0, sizeof(int32_t), stream);
   hipMemsetAsync(n_work, 0, sizeof(int32_t), stream);
-  if (adagrad) {
+  if (long_scratch) {
     hipMemsetAsync(long_scratch, 0,
                    sizeof(float) * scratch_rows * (int64_t)width, stream);
   }
-#define ARGS weight, state, eps, sorted_ids, seg, srow, sw, grad_out, lr,    \
-             nu_ptr, max_segs, width, long_rows, long_count, work_items,     \
-             n_work, long_scratch, adagrad, stream
-  if (width <= 64) {
-    switch (next_pow2(width)) {
-      case 1: launch_sorted_opt_pair<1, 0>(ARGS); break;
-      case 2: launch_sorted_opt_pair<2, 0>(ARGS); break;
-      case 4: launch_sorted_opt_pair<4, 0>(ARGS); break;
-      case 8: launch_sorted_opt_pair<8, 0>(ARGS); break;
-      case 16: launch_sorted_opt_pair<16, 0>(ARGS); break;
-      case 32: launch_sorted_opt_pair<32, 0>(ARGS); break;
-      default: launch_sorted_opt_pair<64, 0>(ARGS); break;
-    }
-  } else if (width % 4 == 0) {
-    launch_sorted_opt_pair<0, 4>(ARGS);
-  } else if (width % 2 == 0) {
-    launch_sorted_opt_pair<0, 2>(ARGS);
+  if (weight_bf16) {
+    launch_sorted_optimizer_update_t<bf16_t>(
+        (bf16_t*)weight, state, eps, sorted_ids, seg, srow, sw, grad_out, lr,
+        nu_ptr, max_segs, width, long_rows, long_count, work_items, n_work,
+        long_scratch, adagrad, stream);
   } else {
-    launch_sorted_opt_pair<0, 1>(ARGS);
+    launch_sorted_optimizer_update_t<float>(
+        (float*)weight, state, eps, sorted_ids, seg, srow, sw, grad_out, lr,
+        nu_ptr, max_segs, width, long_rows, long_count, work_items, n_work,
+        long_scratch, adagrad, stream);
   }
-#undef ARGS
 }
 
 // ---------------------------------------------------------------------------
@@ -976,8 +1086,8 @@ void launch_sorted_optimizer_update(float* weight, float* state, float eps,
 // ---------------------------------------------------------------------------
 
 // SGD: w[id] -= lr * g.   Adagrad: s[id] += g^2; w[id] -= lr*g/(sqrt(s)+eps).
-template <bool ADAGRAD>
-__global__ void sparse_row_update(float* __restrict__ weight,
+template <bool ADAGRAD, typename PT>
+__global__ void sparse_row_update(PT* __restrict__ weight,
                                   float* __restrict__ state,
                                   const int64_t* __restrict__ ids,
                                   const float* __restrict__ grad,
@@ -988,56 +1098,38 @@ __global__ void sparse_row_update(float* __restrict__ weight,
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
   for (int64_t r = wave_id; r < num_rows; r += n_waves) {
     const int64_t row = ids[r];
-    float* wp = weight + row * (int64_t)width;
+    PT* wp = weight + row * (int64_t)width;
     float* sp = ADAGRAD ? state + row * (int64_t)width : nullptr;
     const float* gp = grad + r * (int64_t)width;
-    for (int c = lane * 4; c < width; c += WAVE * 4) {
-      if (c + 4 <= width) {
-        const float4 g4 = *reinterpret_cast<const float4*>(gp + c);
-        float4 w4 = *reinterpret_cast<float4*>(wp + c);
-        if (ADAGRAD) {
-          float4 s4 = *reinterpret_cast<float4*>(sp + c);
-          s4.x += g4.x * g4.x; s4.y += g4.y * g4.y;
-          s4.z += g4.z * g4.z; s4.w += g4.w * g4.w;
-          *reinterpret_cast<float4*>(sp + c) = s4;
-          w4.x -= lr * g4.x / (sqrtf(s4.x) + eps);
-          w4.y -= lr * g4.y / (sqrtf(s4.y) + eps);
-          w4.z -= lr * g4.z / (sqrtf(s4.z) + eps);
-          w4.w -= lr * g4.w / (sqrtf(s4.w) + eps);
-        } else {
-          w4.x -= lr * g4.x; w4.y -= lr * g4.y;
-          w4.z -= lr * g4.z; w4.w -= lr * g4.w;
-        }
-        *reinterpret_cast<float4*>(wp + c) = w4;
+    for (int c = lane; c < width; c += WAVE) {
+      const float g = gp[c];
+      if (ADAGRAD) {
+        const float s = sp[c] + g * g;
+        sp[c] = s;
+        pt_store(&wp[c], pt_load(&wp[c]) - lr * g / (sqrtf(s) + eps));
       } else {
-        for (int cc = c; cc < width; ++cc) {
-          const float g = gp[cc];
-          if (ADAGRAD) {
-            sp[cc] += g * g;
-            wp[cc] -= lr * g / (sqrtf(sp[cc]) + eps);
-          } else {
-            wp[cc] -= lr * g;
-          }
-        }
+        pt_store(&wp[c], pt_load(&wp[c]) - lr * g);
       }
     }
   }
 }
 
-void launch_sparse_row_update(float* weight, float* state, const int64_t* ids,
-                              const float* grad, int64_t num_rows, int width,
-                              float lr, float eps, bool adagrad,
-                              hipStream_t stream) {
+void launch_sparse_row_update(void* weight, bool weight_bf16, float* state,
+                              const int64_t* ids, const float* grad,
+                              int64_t num_rows, int width, float lr, float eps,
+                              bool adagrad, hipStream_t stream) {
   const int block = 256;
   const int grid = pick_grid(num_rows, block / WAVE);
-  if (adagrad)
-    hipLaunchKernelGGL(sparse_row_update<true>, dim3(grid), dim3(block), 0,
-                       stream, weight, state, ids, grad, num_rows, width, lr,
-                       eps);
-  else
-    hipLaunchKernelGGL(sparse_row_update<false>, dim3(grid), dim3(block), 0,
-                       stream, weight, state, ids, grad, num_rows, width, lr,
-                       eps);
+#define SRU(ADA, PT)                                                         \
+  hipLaunchKernelGGL((sparse_row_update<ADA, PT>), dim3(grid), dim3(block),  \
+                     0, stream, (PT*)weight, state, ids, grad, num_rows,     \
+                     width, lr, eps)
+  if (weight_bf16) {
+    if (adagrad) SRU(true, bf16_t); else SRU(false, bf16_t);
+  } else {
+    if (adagrad) SRU(true, float); else SRU(false, float);
+  }
+#undef SRU
 }
 
 // ---------------------------------------------------------------------------

@@ -7,13 +7,13 @@
 #include <cstddef>
 #include <cstdint>
 
-void launch_csr_lookup_forward(const float* params, const int64_t* values,
-                               const int64_t* splits, const float* per_id_w,
-                               float* out, int64_t num_rows, int64_t nnz,
-                               int64_t vocab, int width, bool mean,
-                               int64_t* long_rows, int32_t* long_count,
-                               int64_t* work_items, int32_t* n_work,
-                               hipStream_t stream);
+void launch_csr_lookup_forward(const void* params, bool params_bf16,
+                               const int64_t* values, const int64_t* splits,
+                               const float* per_id_w, float* out,
+                               int64_t num_rows, int64_t nnz, int64_t vocab,
+                               int width, bool mean, int64_t* long_rows,
+                               int32_t* long_count, int64_t* work_items,
+                               int32_t* n_work, hipStream_t stream);
 
 void launch_row_to_split(const int64_t* rows, int64_t nnz, int64_t num_rows,
                          int64_t* splits, hipStream_t stream);
@@ -59,7 +59,8 @@ size_t integer_lookup_temp_bytes(int64_t max_tokens);
 void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
                             const int64_t* bounds, hipStream_t stream);
 
-void launch_sorted_optimizer_update(float* weight, float* state, float eps,
+void launch_sorted_optimizer_update(void* weight, bool weight_bf16,
+                                    float* state, float eps,
                                     const int64_t* sorted_ids,
                                     const int64_t* seg, const int64_t* srow,
                                     const float* sw, const float* grad_out,
@@ -70,10 +71,10 @@ void launch_sorted_optimizer_update(float* weight, float* state, float eps,
                                     float* long_scratch, int64_t scratch_rows,
                                     bool adagrad, hipStream_t stream);
 
-void launch_sparse_row_update(float* weight, float* state, const int64_t* ids,
-                              const float* grad, int64_t num_rows, int width,
-                              float lr, float eps, bool adagrad,
-                              hipStream_t stream);
+void launch_sparse_row_update(void* weight, bool weight_bf16, float* state,
+                              const int64_t* ids, const float* grad,
+                              int64_t num_rows, int width, float lr, float eps,
+                              bool adagrad, hipStream_t stream);
 
 void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,
                              int D, int out_w, int tri_n, hipStream_t stream);

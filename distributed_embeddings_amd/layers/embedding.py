@@ -101,7 +101,11 @@ class Embedding(nn.Module):
         self._fused_method = method
         self._fused_eps = float(eps)
         if method == "adagrad":
-            self.register_buffer("_fused_state", torch.zeros_like(self.weight))
+            # fp32 state regardless of table storage dtype
+            self.register_buffer("_fused_state",
+                                 torch.zeros(self.weight.shape,
+                                             dtype=torch.float32,
+                                             device=self.weight.device))
         else:
             self.register_buffer("_fused_state",
                                  torch.empty(0, dtype=torch.float32,

@@ -82,7 +82,11 @@ def _csr_lookup_ref(
     row_splits: torch.Tensor,
     combiner: str,
 ) -> torch.Tensor:
-    """Pure-PyTorch CSR segmented gather-reduce (fp32 oracle + CPU path)."""
+    """Pure-PyTorch CSR segmented gather-reduce (fp32 oracle + CPU path).
+
+    Accumulates in fp32 and returns fp32 regardless of the table storage
+    dtype (fp32 or bf16) — matching the HIP kernels.
+    """
     num_rows = row_splits.numel() - 1
     lengths = row_splits[1:] - row_splits[:-1]
     # OOB ids contribute zero rows (needed by the row-slice parallel path,
@@ -90,7 +94,7 @@ def _csr_lookup_ref(
     # dist_model_parallel.py:889-904).
     valid = (values >= 0) & (values < weight.shape[0])
     safe = torch.where(valid, values, torch.zeros_like(values))
-    rows = weight.index_select(0, safe)
+    rows = weight.index_select(0, safe).float()
     rows = rows * valid.unsqueeze(1).to(rows.dtype)
     seg_ids = torch.repeat_interleave(
         torch.arange(num_rows, device=values.device), lengths
@@ -117,6 +121,7 @@ class _CsrLookup(torch.autograd.Function):
         ctx.combiner = combiner
         ctx.vocab = weight.shape[0]
         ctx.width = weight.shape[1]
+        ctx.wdtype = weight.dtype
         if weight.is_cuda:
             return _backend.ops().csr_lookup_forward(weight, values, row_splits, combiner == "mean")
         return _csr_lookup_ref(weight, values, row_splits, combiner)
@@ -135,7 +140,7 @@ class _CsrLookup(torch.autograd.Function):
             )
         grad_weight = torch.sparse_coo_tensor(
             unique_ids.unsqueeze(0),
-            unique_grad,
+            unique_grad.to(ctx.wdtype),  # autograd: grad dtype == param dtype
             size=(ctx.vocab, ctx.width),
             is_coalesced=True,
         )
@@ -201,9 +206,12 @@ class _CsrLookupFusedOptimizer(torch.autograd.Function):
                 if ctx.adagrad:
                     state.index_add_(0, unique_ids, unique_grad * unique_grad)
                     denom = state.index_select(0, unique_ids).sqrt_().add_(ctx.eps)
-                    weight.index_add_(0, unique_ids, -lr_v * unique_grad / denom)
+                    weight.index_add_(
+                        0, unique_ids,
+                        (-lr_v * unique_grad / denom).to(weight.dtype))
                 else:
-                    weight.index_add_(0, unique_ids, unique_grad * (-lr_v))
+                    weight.index_add_(0, unique_ids,
+                                      (unique_grad * (-lr_v)).to(weight.dtype))
         return None, None, None, None, None, None, None, None
 
 

@@ -68,14 +68,16 @@ class SparseEmbeddingOptimizer(torch.optim.Optimizer):
     def _state_for(self, p):
         st = self.state[p]
         if "sum" not in st:
-            st["sum"] = torch.zeros_like(p)
+            # fp32 accumulator state regardless of param storage dtype
+            st["sum"] = torch.zeros(p.shape, dtype=torch.float32,
+                                    device=p.device)
         return st["sum"]
 
     def _sparse_update(self, p, g, lr, eps, adagrad, assume_coalesced):
         if not (assume_coalesced or g.is_coalesced()):
             g = g.coalesce()
         ids = g._indices()[0]
-        vals = g._values()
+        vals = g._values().float()
         if p.is_cuda:
             state = self._state_for(p) if adagrad else torch.empty(0)
             _backend.ops().sparse_row_update(p.data, state, ids.contiguous(),
@@ -86,9 +88,9 @@ class SparseEmbeddingOptimizer(torch.optim.Optimizer):
                 state = self._state_for(p)
                 state.index_add_(0, ids, vals * vals)
                 denom = state.index_select(0, ids).sqrt_().add_(eps)
-                p.data.index_add_(0, ids, -lr * vals / denom)
+                p.data.index_add_(0, ids, (-lr * vals / denom).to(p.dtype))
             else:
-                p.data.index_add_(0, ids, -lr * vals)
+                p.data.index_add_(0, ids, (-lr * vals).to(p.dtype))
 
     def _dense_update_batch(self, params, lr, eps, adagrad):
         """foreach-batched dense updates (one fused launch set per step)."""

@@ -169,3 +169,45 @@ def test_fused_adagrad_cpu_matches_explicit(seed):
     assert e1.weight.grad is None
     assert torch.allclose(e1.weight, e2.weight, atol=1e-5), \
         float((e1.weight - e2.weight).abs().max())
+
+
+def test_bf16_table_forward_matches_fp32(seed):
+    """bf16 table storage: forward accumulates fp32 and matches the fp32
+    lookup of the (bf16-rounded) values exactly."""
+    from distributed_embeddings_amd import Embedding, Ragged, embedding_lookup
+    w32 = torch.randn(50, 16)
+    wbf = w32.bfloat16()
+    r = Ragged.from_lists([[1, 2, 3], [7], [4, 4, 9]])
+    out_bf = embedding_lookup(wbf, r, "sum")
+    out_ref = embedding_lookup(wbf.float(), r, "sum")
+    assert out_bf.dtype == torch.float32
+    assert torch.equal(out_bf, out_ref)
+
+
+def test_bf16_table_training_cpu(seed):
+    from distributed_embeddings_amd import Embedding, Ragged, SparseEmbeddingOptimizer
+    e = Embedding(40, 8, combiner="sum", dtype=torch.bfloat16)
+    opt = SparseEmbeddingOptimizer(e.parameters(), lr=0.1, method="adagrad")
+    r = Ragged.from_lists([[1, 2], [3, 3], [5]])
+    for _ in range(2):
+        opt.zero_grad()
+        out = e(r)
+        assert out.dtype == torch.float32
+        out.square().sum().backward()
+        assert e.weight.grad.dtype == torch.bfloat16
+        opt.step()
+    assert e.weight.dtype == torch.bfloat16
+    assert torch.isfinite(e.weight.float()).all()
+
+
+def test_bf16_fused_optimizer_cpu(seed):
+    from distributed_embeddings_amd import Embedding, Ragged
+    for method in ("sgd", "adagrad"):
+        e = Embedding(40, 8, combiner="sum", dtype=torch.bfloat16)
+        e.enable_fused_optimizer(method, 0.1)
+        assert e._fused_state.dtype == torch.float32
+        w0 = e.weight.detach().clone()
+        out = e(Ragged.from_lists([[1, 2], [3]]))
+        out.sum().backward()
+        assert e.weight.grad is None
+        assert not torch.equal(e.weight, w0), method

@@ -431,3 +431,76 @@ def test_hipgraph_capture_fused_train_step():
     torch.cuda.synchronize()
     assert not torch.equal(w_before, model.col_layers[0].weight), \
         "replay did not update weights"
+
+
+@requires_gpu
+@pytest.mark.parametrize("width", [16, 32, 64, 128, 256])
+def test_bf16_table_forward_gpu(width):
+    """bf16 table storage on GPU == fp32 lookup of the bf16-rounded values."""
+    from distributed_embeddings_amd import Ragged, embedding_lookup
+    torch.manual_seed(width)
+    wbf = torch.randn(500, width).bfloat16().cuda()
+    r = _ragged_random(200, 500, 6, width, "cuda")
+    out = embedding_lookup(wbf, r, "sum")
+    ref = embedding_lookup(wbf.float(), r, "sum")
+    assert out.dtype == torch.float32
+    assert torch.allclose(out, ref, atol=1e-5), \
+        f"w{width}: {(out - ref).abs().max()}"
+
+
+@requires_gpu
+def test_bf16_table_long_segments_gpu():
+    from distributed_embeddings_amd import Ragged, embedding_lookup
+    torch.manual_seed(2)
+    wbf = torch.randn(20, 128).bfloat16().cuda()
+    lens = torch.tensor([1, 3000, 0, 200])
+    vals = torch.randint(0, 20, (int(lens.sum()),))
+    r = Ragged.from_row_lengths(vals.cuda(), lens.cuda())
+    out = embedding_lookup(wbf, r, "mean")
+    ref = embedding_lookup(wbf.float(), r, "mean")
+    assert torch.allclose(out, ref, atol=1e-2), (out - ref).abs().max()
+
+
+@requires_gpu
+@pytest.mark.parametrize("method", ["sgd", "adagrad"])
+def test_bf16_fused_optimizer_gpu_matches_cpu(method):
+    from distributed_embeddings_amd import Embedding, Ragged
+    torch.manual_seed(5)
+    w0 = torch.randn(300, 64).bfloat16()
+    ids = torch.cat([torch.randint(0, 300, (2000,)),
+                     torch.full((1500,), 7)])[torch.randperm(3500)]
+    splits = torch.arange(0, 3501, 5)
+    up = torch.randn(700, 64)
+    e_g = Embedding(300, 64, combiner="sum", dtype=torch.bfloat16).cuda()
+    e_c = Embedding(300, 64, combiner="sum", dtype=torch.bfloat16)
+    with torch.no_grad():
+        e_g.weight.copy_(w0)
+        e_c.weight.copy_(w0)
+    e_g.enable_fused_optimizer(method, 0.05)
+    e_c.enable_fused_optimizer(method, 0.05)
+    e_g(Ragged(ids.cuda(), splits.cuda())).backward(up.cuda())
+    e_c(Ragged(ids, splits)).backward(up)
+    d = (e_g.weight.detach().cpu().float() - e_c.weight.detach().float()).abs()
+    # bf16 storage rounding: one step within a few ulp of the CPU oracle
+    assert float(d.max()) < 0.05, float(d.max())
+
+
+@requires_gpu
+@pytest.mark.parametrize("method", ["sgd", "adagrad"])
+def test_bf16_sparse_optimizer_gpu(method):
+    from distributed_embeddings_amd import Embedding, Ragged, SparseEmbeddingOptimizer
+    torch.manual_seed(6)
+    e_g = Embedding(200, 32, combiner="sum", dtype=torch.bfloat16).cuda()
+    e_c = Embedding(200, 32, combiner="sum", dtype=torch.bfloat16)
+    with torch.no_grad():
+        e_c.weight.copy_(e_g.weight.cpu())
+    og = SparseEmbeddingOptimizer(e_g.parameters(), lr=0.05, method=method)
+    oc = SparseEmbeddingOptimizer(e_c.parameters(), lr=0.05, method=method)
+    r = Ragged.from_lists([[1, 2, 3], [5], [2, 9]])
+    for _ in range(2):
+        og.zero_grad(); oc.zero_grad()
+        e_g(r.to("cuda")).square().sum().backward()
+        e_c(r).square().sum().backward()
+        og.step(); oc.step()
+    d = (e_g.weight.detach().cpu().float() - e_c.weight.detach().float()).abs()
+    assert float(d.max()) < 0.05, float(d.max())
