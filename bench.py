@@ -32,6 +32,9 @@ def parse_args():
     p.add_argument("--pool", type=int, default=8, help="pre-generated batch pool size")
     p.add_argument("--strategy", type=str, default="memory_balanced")
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--table-dtype", type=str, default="fp32",
+                   choices=["fp32", "bf16"],
+                   help="embedding table storage dtype (fp32 accumulate both)")
     p.add_argument("--fused-sgd", dest="fused_sgd", action="store_true", default=True)
     p.add_argument("--no-fused-sgd", dest="fused_sgd", action="store_false")
     p.add_argument("--graph", dest="graph", action="store_true", default=True,
@@ -60,9 +63,10 @@ def build_model(args, device):
     from distributed_embeddings_amd.models.synthetic import SyntheticModel, expand_tables
 
     if args.model == "dlrm-criteo":
-        with device:  # construct tables directly in HBM (96 GB fp32)
+        tdt = torch.bfloat16 if args.table_dtype == "bf16" else torch.float32
+        with device:  # construct tables directly in HBM (96 GB fp32 / 48 GB bf16)
             model = DLRM(CRITEO_1TB_TABLE_SIZES, embedding_dim=128,
-                         strategy=args.strategy)
+                         strategy=args.strategy, table_dtype=tdt)
         table_sizes = CRITEO_1TB_TABLE_SIZES
         hotness = [1] * len(table_sizes)
         num_numerical = 13

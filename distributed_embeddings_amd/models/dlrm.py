@@ -83,6 +83,7 @@ class DLRM(nn.Module):
         dp_input: bool = True,
         column_slice_threshold: Optional[int] = None,
         data_parallel_threshold: Optional[int] = None,
+        table_dtype: torch.dtype = torch.float32,
     ):
         super().__init__()
         self.table_sizes = list(table_sizes)
@@ -112,7 +113,8 @@ class DLRM(nn.Module):
         self.embeddings = DistributedEmbedding(
             tables, strategy=strategy, dp_input=dp_input,
             column_slice_threshold=column_slice_threshold,
-            data_parallel_threshold=data_parallel_threshold)
+            data_parallel_threshold=data_parallel_threshold,
+            table_dtype=table_dtype)
 
     def local_cat_feature_ids(self) -> List[int]:
         if self.distributed and not self.dp_input:

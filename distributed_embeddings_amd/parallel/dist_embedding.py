@@ -99,8 +99,12 @@ class DistributedEmbedding(nn.Module):
         gpu_embedding_size: Optional[int] = None,
         dp_input: bool = True,
         input_table_map: Optional[Sequence[int]] = None,
+        table_dtype: torch.dtype = torch.float32,
     ):
         super().__init__()
+        if table_dtype not in (torch.float32, torch.bfloat16):
+            raise ValueError("table_dtype must be float32 or bfloat16")
+        self.table_dtype = table_dtype
         self.world_size = comm.world_size()
         self.rank = comm.rank()
         self.dp_input = dp_input
@@ -123,7 +127,7 @@ class DistributedEmbedding(nn.Module):
         # ---- data-parallel layers (replicated; grads allreduced) ----
         self.dp_layers = nn.ModuleList([
             Embedding(configs[t].input_dim, configs[t].output_dim, configs[t].combiner,
-                      initializer=configs[t].initializer)
+                      initializer=configs[t].initializer, dtype=table_dtype)
             for t in plan.dp_table_ids
         ])
 
@@ -135,7 +139,8 @@ class DistributedEmbedding(nn.Module):
         # XLA-fusion hint, dist_model_parallel.py:650).
         col_layers = []
         for grp in plan.local_concat_groups(self.rank):
-            lyr = Embedding(grp.input_dim, grp.output_dim, grp.combiner)
+            lyr = Embedding(grp.input_dim, grp.output_dim, grp.combiner,
+                            dtype=table_dtype)
             for p in lyr.parameters():
                 if self.world_size > 1:
                     p.de_local = True  # model-parallel: excluded from grad allreduce
@@ -154,7 +159,13 @@ class DistributedEmbedding(nn.Module):
                                                    m.col_offset + m.width])
                     elif cfg.initializer is not None:
                         if m.width == cfg.output_dim:
-                            cfg.initializer(dst)
+                            if dst.dtype == torch.float32:
+                                cfg.initializer(dst)
+                            else:
+                                tmp = torch.empty(dst.shape, dtype=torch.float32,
+                                                  device=dst.device)
+                                cfg.initializer(tmp)
+                                dst.copy_(tmp)
                         else:
                             full = torch.empty(cfg.input_dim, cfg.output_dim,
                                                dtype=dst.dtype, device=dst.device)
@@ -167,7 +178,8 @@ class DistributedEmbedding(nn.Module):
         row_layers = []
         for shard in plan.local_row_shards(self.rank):
             cfg = configs[shard.table_id]
-            lyr = Embedding(max(shard.rows, 1), cfg.output_dim, cfg.combiner)
+            lyr = Embedding(max(shard.rows, 1), cfg.output_dim, cfg.combiner,
+                            dtype=table_dtype)
             lyr._oob_zero = True
             for p in lyr.parameters():
                 p.de_local = True

@@ -133,3 +133,19 @@ def test_shared_tables_multihot_world2():
     results = run_distributed(_shared_multihot_worker, world=2)
     for errs in results:
         assert max(errs) < 1e-5
+
+
+def test_bf16_table_dtype_module():
+    import distributed_embeddings_amd as de
+    model = de.DistributedEmbedding(
+        [de.TableConfig(40, 8), de.TableConfig(60, 8, "sum")],
+        table_dtype=torch.bfloat16)
+    assert model.col_layers[0].weight.dtype == torch.bfloat16
+    weights = [np.random.RandomState(0).randn(40, 8).astype(np.float32),
+               np.random.RandomState(1).randn(60, 8).astype(np.float32)]
+    model.set_weights(weights)
+    outs = model([torch.randint(0, 40, (4,)),
+                  torch.randint(0, 60, (4, 3))])
+    assert all(torch.isfinite(o.float()).all() for o in outs)
+    got = model.get_weights()
+    assert np.allclose(got[0], weights[0], atol=0.01)  # bf16 rounding
