@@ -39,6 +39,9 @@ def parse_args():
     p.add_argument("--dp-input", action="store_true", default=True)
     p.add_argument("--optimizer", default="adagrad", choices=["adagrad", "sgd"])
     p.add_argument("--pool", type=int, default=4)
+    p.add_argument("--graph", dest="graph", action="store_true", default=True,
+                   help="hipGraph-capture the step (world==1; eager fallback)")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
     return p.parse_args()
 
 
@@ -92,15 +95,42 @@ def main():
     _loss_sum = torch.nn.BCEWithLogitsLoss(reduction="sum")
     loss_fn = lambda lg, lb: _loss_sum(lg, lb) / args.batch_size
 
-    def step(i):
-        num, cats, labels = pool[i % len(pool)]
-        opt.zero_grad(set_to_none=True)
+    def run_step(num, cats, labels, set_to_none=True):
+        opt.zero_grad(set_to_none=set_to_none)
         with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
                             enabled=device.type == "cuda"):
             loss = loss_fn(model(num, cats).float(), labels)
         loss.backward()
         opt.step()
         return loss
+
+    graph = None
+    if args.graph and world == 1 and device.type == "cuda":
+        s_num, s_cats, s_labels = pool[0]
+        s_num, s_labels = s_num.clone(), s_labels.clone()
+        s_cats = [c.clone() for c in s_cats]
+        try:
+            for i in range(max(args.warmup, 2)):
+                run_step(s_num, s_cats, s_labels, set_to_none=False)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                run_step(s_num, s_cats, s_labels, set_to_none=False)
+            graph = g
+        except Exception as e:
+            print(f"# graph capture failed ({type(e).__name__}); eager")
+            graph = None
+
+    def step(i):
+        num, cats, labels = pool[i % len(pool)]
+        if graph is not None:
+            s_num.copy_(num)
+            for d, s in zip(s_cats, cats):
+                d.copy_(s)
+            s_labels.copy_(labels)
+            graph.replay()
+            return torch.zeros(1)
+        return run_step(num, cats, labels)
 
     for i in range(args.warmup):
         step(i)
