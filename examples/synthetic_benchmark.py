@@ -78,15 +78,25 @@ def main():
     table_sizes = [tables[t][0] for t in input_map]
     local_bs = args.batch_size // world
 
+    # cats flat-packed per entry: graph replay needs ONE device copy
     pool = []
+    cat_shapes = None
     for i in range(args.pool):
         g = torch.Generator().manual_seed(23 + 131 * rank + i)
         cats = [c.to(device) for c in
                 make_batch(table_sizes, hotness, local_bs, args.alpha,
                            generator=g, keep_hot_dim=True)]
+        if cat_shapes is None:
+            cat_shapes = [c.shape for c in cats]
+            cat_sizes = [c.numel() for c in cats]
+        flat = torch.cat([c.reshape(-1) for c in cats])
         num = torch.rand(local_bs, cfg.num_numerical_features, device=device)
         labels = torch.randint(0, 2, (local_bs, 1), device=device).float()
-        pool.append((num, cats, labels))
+        pool.append((num, flat, labels))
+
+    def carve(flat):
+        return [p.view(s) for p, s in
+                zip(torch.split(flat, cat_sizes), cat_shapes)]
 
     opt = de.DistributedOptimizer(SparseEmbeddingOptimizer(
         model.parameters(), lr=0.01, method=args.optimizer), average=False)
@@ -106,9 +116,8 @@ def main():
 
     graph = None
     if args.graph and world == 1 and device.type == "cuda":
-        s_num, s_cats, s_labels = pool[0]
-        s_num, s_labels = s_num.clone(), s_labels.clone()
-        s_cats = [c.clone() for c in s_cats]
+        s_num, s_flat, s_labels = (t.clone() for t in pool[0])
+        s_cats = carve(s_flat)
         try:
             for i in range(max(args.warmup, 2)):
                 run_step(s_num, s_cats, s_labels, set_to_none=False)
@@ -122,15 +131,14 @@ def main():
             graph = None
 
     def step(i):
-        num, cats, labels = pool[i % len(pool)]
+        num, flat, labels = pool[i % len(pool)]
         if graph is not None:
             s_num.copy_(num)
-            for d, s in zip(s_cats, cats):
-                d.copy_(s)
+            s_flat.copy_(flat)
             s_labels.copy_(labels)
             graph.replay()
             return torch.zeros(1)
-        return run_step(num, cats, labels)
+        return run_step(num, carve(flat), labels)
 
     for i in range(args.warmup):
         step(i)
