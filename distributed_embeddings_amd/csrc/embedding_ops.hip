@@ -160,25 +160,33 @@ __global__ void csr_fwd_wide(const PT* __restrict__ params,
                              float* __restrict__ out, int64_t num_rows,
                              int64_t vocab, int width, int64_t long_thresh,
                              int64_t* __restrict__ long_rows,
-                             int32_t* __restrict__ long_count) {
+                             int32_t* __restrict__ long_count,
+                             int tile_w /* pow2 lanes per row, <= WAVE */) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
-  constexpr int CHUNK = WAVE * VEC;
-  for (int64_t row = wave_id; row < num_rows; row += n_waves) {
+  // sub-wave tiling: tile_w lanes own one row, so a width-128 VEC-4 table
+  // runs 2 rows per wave with every lane loading (was: half the wave idle).
+  const int rpw = WAVE / tile_w;
+  const int sub = lane / tile_w;
+  const int tl = lane % tile_w;
+  const int chunk = tile_w * VEC;
+  for (int64_t base = wave_id * rpw; base < num_rows; base += n_waves * rpw) {
+    const int64_t row = base + sub;
+    if (row >= num_rows) continue;
     const int64_t s = splits[row], e = splits[row + 1];
     if (e - s > long_thresh) {
-      for (int c = lane; c < width; c += WAVE) out[row * width + c] = 0.f;
-      if (lane == 0) long_rows[atomicAdd(long_count, 1)] = row;
+      for (int c = tl; c < width; c += tile_w) out[row * width + c] = 0.f;
+      if (tl == 0) long_rows[atomicAdd(long_count, 1)] = row;
       continue;
     }
     const float inv = (MEAN && e > s) ? 1.f / (float)(e - s) : 1.f;
-    for (int cbase = 0; cbase < width; cbase += CHUNK) {
+    for (int cbase = 0; cbase < width; cbase += chunk) {
       float acc[VEC];
 #pragma unroll
       for (int v = 0; v < VEC; ++v) acc[v] = 0.f;
-      const int col0 = cbase + lane * VEC;
+      const int col0 = cbase + tl * VEC;
       for (int64_t k = s; k < e; ++k) {
         const int64_t idx = values[k];
         if (idx < 0 || idx >= vocab) continue;
@@ -372,7 +380,17 @@ static void launch_csr_pair(const PT* params, const int64_t* values,
                             int64_t* work_items, int32_t* n_work,
                             hipStream_t stream) {
   const int block = 256, bw = block / WAVE;
-  const int64_t row_waves = TILE > 0 ? cdiv64(num_rows, WAVE / (TILE > 0 ? TILE : 1)) : num_rows;
+  // wide path: tile_w = pow2 lanes covering one row (rows-per-wave = 64/tile_w)
+  int tile_w = WAVE;
+  if (TILE == 0) {
+    const int lanes_needed = (width + (VEC > 0 ? VEC : 1) - 1) /
+                             (VEC > 0 ? VEC : 1);
+    tile_w = next_pow2(lanes_needed);
+    if (tile_w > WAVE) tile_w = WAVE;
+  }
+  const int64_t row_waves = TILE > 0
+                                ? cdiv64(num_rows, WAVE / (TILE > 0 ? TILE : 1))
+                                : cdiv64(num_rows, WAVE / tile_w);
   const dim3 grid(pick_grid(row_waves, bw));
   // Adaptive long threshold: when there are plenty of rows the chip is full
   // without splitting, so only true outliers (>4x the average and >LONG_T)
@@ -397,7 +415,7 @@ static void launch_csr_pair(const PT* params, const int64_t* values,
       hipLaunchKernelGGL((csr_fwd_wide<(VEC > 0 ? VEC : 1), MEAN, HASW, PT>),  \
                          grid, dim3(block), 0, stream, params, values, splits, \
                          per_id_w, out, num_rows, vocab, width, long_thresh,   \
-                         long_rows, long_count);                               \
+                         long_rows, long_count, tile_w);                       \
   } while (0)
 #define LB(MEAN, HASW)                                                         \
   do {                                                                         \
