@@ -265,13 +265,21 @@ __global__ void csr_fwd_long(const PT* __restrict__ params,
                              float* __restrict__ out, int64_t vocab, int width,
                              const int64_t* __restrict__ long_rows,
                              const int64_t* __restrict__ work_items,
-                             const int32_t* __restrict__ n_work_ptr) {
+                             const int32_t* __restrict__ n_work_ptr,
+                             int tile_w) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id =
       ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  // wide path (TILE==0): tile_w lanes per work item, 64/tile_w items per wave
+  const int rpw_w = TILE > 0 ? 1 : (WAVE / tile_w);
+  const int sub_w = TILE > 0 ? 0 : (lane / tile_w);
+  const int tl_w = TILE > 0 ? lane : (lane % tile_w);
   const int64_t n_items = *n_work_ptr;
-  for (int64_t item = wave_id; item < n_items; item += n_waves) {
+  for (int64_t ibase = wave_id * rpw_w; ibase < n_items;
+       ibase += n_waves * rpw_w) {
+    const int64_t item = ibase + sub_w;
+    if (item >= n_items) continue;
     const int64_t w_it = work_items[item];
     const int64_t li = w_it >> 24;
     const int64_t chunk = w_it & 0xffffff;
@@ -328,12 +336,12 @@ __global__ void csr_fwd_long(const PT* __restrict__ params,
       }
     } else {
       constexpr int V = VEC > 0 ? VEC : 1;
-      constexpr int CH = WAVE * V;
+      const int CH = tile_w * V;
       for (int cbase = 0; cbase < width; cbase += CH) {
         float acc[V];
 #pragma unroll
         for (int v = 0; v < V; ++v) acc[v] = 0.f;
-        const int col0 = cbase + lane * V;
+        const int col0 = cbase + tl_w * V;
         {
           const int64_t ks = k0;
           const int64_t ke = min(ks + (int64_t)LONG_T, e);
@@ -424,7 +432,7 @@ static void launch_csr_pair(const PT* params, const int64_t* values,
     hipLaunchKernelGGL((csr_fwd_long<TILE, VEC, MEAN, HASW, PT>), dim3(2048),  \
                        dim3(block), 0, stream, params, values, splits,         \
                        per_id_w, out, vocab, width, long_rows, work_items,     \
-                       n_work);                                                \
+                       n_work, tile_w);                                        \
   } while (0)
   if (mean) {
     if (per_id_w) { LA(true, true); LB(true, true); }
@@ -803,17 +811,24 @@ __global__ void sorted_opt_update(PT* __restrict__ weight,
                                   const int32_t* __restrict__ nu_ptr,
                                   int64_t long_thresh, int width,
                                   int64_t* __restrict__ long_rows,
-                                  int32_t* __restrict__ long_count) {
+                                  int32_t* __restrict__ long_count,
+                                  int tile_w) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
   const float lr = *lr_ptr;
+  const int rpw_w = TILE > 0 ? 1 : (WAVE / tile_w);
+  const int sub_w = TILE > 0 ? 0 : (lane / tile_w);
+  const int tl_w = TILE > 0 ? lane : (lane % tile_w);
   const int64_t n_segs = *nu_ptr;
-  for (int64_t r = wave_id; r < n_segs; r += n_waves) {
+  for (int64_t rbase = wave_id * rpw_w; rbase < n_segs;
+       rbase += n_waves * rpw_w) {
+    const int64_t r = rbase + sub_w;
+    if (r >= n_segs) continue;
     const int64_t s = seg[r], e = seg[r + 1];
     if (s >= e) continue;
     if (e - s > long_thresh) {
-      if (lane == 0) long_rows[atomicAdd(long_count, 1)] = r;
+      if (tl_w == 0) long_rows[atomicAdd(long_count, 1)] = r;
       continue;
     }
     const int64_t uid = sorted_ids[s];
@@ -835,12 +850,12 @@ __global__ void sorted_opt_update(PT* __restrict__ weight,
       }
     } else {
       constexpr int V = VEC > 0 ? VEC : 1;
-      constexpr int CH = WAVE * V;
+      const int CH = tile_w * V;
       for (int cbase = 0; cbase < width; cbase += CH) {
         float acc[V];
 #pragma unroll
         for (int v = 0; v < V; ++v) acc[v] = 0.f;
-        const int col0 = cbase + lane * V;
+        const int col0 = cbase + tl_w * V;
         for (int64_t k = s; k < e; ++k) {
           const float w = HAS_W ? sw[k] : 1.f;
           const float* gp = grad_out + srow[k] * (int64_t)width + col0;
@@ -891,13 +906,20 @@ __global__ void sorted_opt_long(float* __restrict__ target,  // weight or scratc
                                 const float* __restrict__ lr_ptr, int width,
                                 const int64_t* __restrict__ long_rows,
                                 const int64_t* __restrict__ work_items,
-                                const int32_t* __restrict__ n_work_ptr) {
+                                const int32_t* __restrict__ n_work_ptr,
+                                int tile_w) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
   const float lr = TO_SCRATCH ? 1.f : *lr_ptr;
+  const int rpw_w = TILE > 0 ? 1 : (WAVE / tile_w);
+  const int sub_w = TILE > 0 ? 0 : (lane / tile_w);
+  const int tl_w = TILE > 0 ? lane : (lane % tile_w);
   const int64_t n_items = *n_work_ptr;
-  for (int64_t item = wave_id; item < n_items; item += n_waves) {
+  for (int64_t ibase = wave_id * rpw_w; ibase < n_items;
+       ibase += n_waves * rpw_w) {
+    const int64_t item = ibase + sub_w;
+    if (item >= n_items) continue;
     const int64_t w_it = work_items[item];
     const int64_t li = w_it >> 24;
     const int64_t chunk = w_it & 0xffffff;
@@ -918,12 +940,12 @@ __global__ void sorted_opt_long(float* __restrict__ target,  // weight or scratc
       }
     } else {
       constexpr int V = VEC > 0 ? VEC : 1;
-      constexpr int CH = WAVE * V;
+      const int CH = tile_w * V;
       for (int cbase = 0; cbase < width; cbase += CH) {
         float acc[V];
 #pragma unroll
         for (int v = 0; v < V; ++v) acc[v] = 0.f;
-        const int col0 = cbase + lane * V;
+        const int col0 = cbase + tl_w * V;
         for (int64_t k = ks; k < ke; ++k) {
           const float w = HAS_W ? sw[k] : 1.f;
           const float* gp = grad_out + srow[k] * (int64_t)width + col0;
@@ -990,8 +1012,16 @@ static void launch_sorted_opt_pair(PT* weight, float* state, float eps,
                                    float* long_scratch, bool adagrad,
                                    hipStream_t stream) {
   const int block = 256;
+  int tile_w = WAVE;
+  if (TILE == 0) {
+    const int lanes_needed = (width + (VEC > 0 ? VEC : 1) - 1) /
+                             (VEC > 0 ? VEC : 1);
+    tile_w = next_pow2(lanes_needed);
+    if (tile_w > WAVE) tile_w = WAVE;
+  }
   const int64_t row_waves =
-      TILE > 0 ? cdiv64(max_segs, WAVE / (TILE > 0 ? TILE : 1)) : max_segs;
+      TILE > 0 ? cdiv64(max_segs, WAVE / (TILE > 0 ? TILE : 1))
+               : cdiv64(max_segs, WAVE / tile_w);
   const int grid = pick_grid(row_waves, block / WAVE);
   // bf16 weights have no atomicAdd: their long-SGD path also goes through the
   // fp32 scratch + finalize pair.
@@ -1001,17 +1031,17 @@ static void launch_sorted_opt_pair(PT* weight, float* state, float eps,
   hipLaunchKernelGGL((sorted_opt_update<TILE, VEC, HASW, ADA, PT>),            \
                      dim3(grid), dim3(block), 0, stream, weight, state, eps,   \
                      sorted_ids, seg, srow, sw, grad_out, lr, nu_ptr,          \
-                     (int64_t)LONG_T, width, long_rows, long_count)
+                     (int64_t)LONG_T, width, long_rows, long_count, tile_w)
 #define SL_SCRATCH(HASW)                                                       \
   hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, true>), dim3(2048),     \
                      dim3(block), 0, stream, long_scratch, sorted_ids, seg,    \
                      srow, sw, grad_out, lr, width, long_rows, work_items,     \
-                     n_work)
+                     n_work, tile_w)
 #define SL_DIRECT(HASW)                                                        \
   hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, false>), dim3(2048),    \
                      dim3(block), 0, stream, (float*)weight, sorted_ids, seg,  \
                      srow, sw, grad_out, lr, width, long_rows, work_items,     \
-                     n_work)
+                     n_work, tile_w)
 #define FIN(ADA)                                                               \
   hipLaunchKernelGGL((sorted_long_finalize<ADA, PT>), dim3(256), dim3(block),  \
                      0, stream, weight, state, eps, sorted_ids, seg, lr,       \
