@@ -292,3 +292,77 @@ def _large_testcase_worker(rank, world):
 def test_large_testcase_world2():
     results = run_distributed(_large_testcase_worker, world=2)
     assert max(results) < 1e-4
+
+
+def _mp_input_multihot_worker(rank, world):
+    import distributed_embeddings_amd as de
+    tables = [de.TableConfig(50, 8, "sum"), de.TableConfig(60, 8, "mean")]
+    model = de.DistributedEmbedding(tables, strategy="basic", dp_input=False)
+    weights = _ref_weights([50, 60], 8)
+    model.set_weights([w.numpy() for w in weights])
+    my_ids = model.local_input_ids()
+    g = torch.Generator().manual_seed(31)
+    inputs = [torch.randint(0, 50, (world * 4, 3), generator=g),
+              torch.randint(0, 60, (world * 4, 2), generator=g)]
+    outs = model([inputs[i] for i in my_ids])
+    return {"outs": [o.detach() for o in outs], "my_ids": my_ids,
+            "inputs": inputs}
+
+
+def test_mp_input_multihot_world2():
+    results = run_distributed(_mp_input_multihot_worker, world=2)
+    weights = _ref_weights([50, 60], 8)
+    inputs = results[0]["inputs"]
+    refs = [weights[0][inputs[0]].sum(1), weights[1][inputs[1]].mean(1)]
+    for rank in range(2):
+        outs = results[rank]["outs"]
+        for t in range(2):
+            ref = refs[t][rank * 4:(rank + 1) * 4]
+            assert torch.allclose(outs[t], ref, atol=1e-5), \
+                f"rank{rank} t{t}: {(outs[t] - ref).abs().max()}"
+
+
+def _offload_world2_worker(rank, world):
+    import distributed_embeddings_amd as de
+    sizes = [100, 5000, 50]
+    model = de.DistributedEmbedding(
+        [de.TableConfig(s, 8) for s in sizes],
+        strategy="memory_balanced",
+        gpu_embedding_size=600 * 8)  # the 5000-row slice offloads
+    weights = _ref_weights(sizes, 8)
+    model.set_weights([w.numpy() for w in weights])
+    offloaded = any(getattr(l, "_cpu_offload", False) for l in model.col_layers)
+    g = torch.Generator().manual_seed(13)
+    inputs = [torch.randint(0, s, (world * 4,), generator=g) for s in sizes]
+    sl = slice(rank * 4, (rank + 1) * 4)
+    outs = model([x[sl] for x in inputs])
+    errs = [float((o - weights[t][inputs[t][sl]]).abs().max())
+            for t, o in enumerate(outs)]
+    return max(errs), offloaded
+
+
+def test_cpu_offload_world2():
+    results = run_distributed(_offload_world2_worker, world=2)
+    assert any(off for _, off in results)
+    for err, _ in results:
+        assert err < 1e-5
+
+
+def _rowslice_dtype_worker(rank, world):
+    import distributed_embeddings_amd as de
+    model = de.DistributedEmbedding([de.TableConfig(900, 8)],
+                                    row_slice_threshold=1)
+    weights = _ref_weights([900], 8)
+    model.set_weights([w.numpy() for w in weights])
+    g = torch.Generator().manual_seed(3)
+    ids = torch.randint(0, 900, (world * 4,), generator=g)
+    sl = slice(rank * 4, (rank + 1) * 4)
+    outs = model([ids[sl]], output_dtype=torch.bfloat16)
+    assert outs[0].dtype == torch.bfloat16
+    ref = weights[0][ids[sl]].bfloat16()
+    return float((outs[0].float() - ref.float()).abs().max())
+
+
+def test_row_slice_output_dtype_world2():
+    results = run_distributed(_rowslice_dtype_worker, world=2)
+    assert max(results) < 0.05
