@@ -599,13 +599,13 @@ __global__ void find_valid_bounds(const int64_t* __restrict__ sorted_ids,
 }
 
 size_t csr_backward_temp_bytes(int64_t nnz, int64_t vocab) {
-  size_t sort_bytes = 0, scan_bytes = 0;
-  rocprim::radix_sort_pairs(nullptr, sort_bytes, (const int64_t*)nullptr,
-                            (int64_t*)nullptr, (const int32_t*)nullptr,
-                            (int32_t*)nullptr, (size_t)nnz);
+  // only the head-flag inclusive scan uses rocPRIM temp storage now (the
+  // sort is the hand-written radix_sort.hip path)
+  (void)vocab;
+  size_t scan_bytes = 0;
   rocprim::inclusive_scan(nullptr, scan_bytes, (const int32_t*)nullptr,
                           (int32_t*)nullptr, (size_t)nnz);
-  return sort_bytes > scan_bytes ? sort_bytes : scan_bytes;
+  return scan_bytes;
 }
 
 // Orchestration is done on the host side (bindings.cpp) because output
