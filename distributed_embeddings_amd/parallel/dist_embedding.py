@@ -363,19 +363,26 @@ class DistributedEmbedding(nn.Module):
         plan = self.strategy
         W = self.world_size
         b = col_inputs[0].shape[0] if col_inputs else 0
-        # send: for each dest rank, its pairs' ids from my local batch
-        send_parts, in_splits = [], []
-        for k in range(W):
-            n = 0
-            for i in plan.rank_input_ids[k]:
-                x = col_inputs[i]
-                send_parts.append(x.reshape(-1))
-                n += x.numel()
-            in_splits.append(n)
+        # split sizes are static per input-shape signature: cache them
+        sig = ("dp2mp", tuple(tuple(x.shape) for x in col_inputs))
+        cache = getattr(self, "_split_cache", None)
+        if cache is None:
+            cache = self._split_cache = {}
+        cached = cache.get(sig)
+        if cached is None:
+            in_splits = []
+            for k in range(W):
+                in_splits.append(sum(col_inputs[i].numel()
+                                     for i in plan.rank_input_ids[k]))
+            my_sizes = [col_inputs[i].numel()
+                        for i in plan.rank_input_ids[self.rank]]
+            cached = (in_splits, my_sizes, [sum(my_sizes)] * W)
+            cache[sig] = cached
+        in_splits, my_sizes, out_splits = cached
+        send_parts = [col_inputs[i].reshape(-1)
+                      for k in range(W) for i in plan.rank_input_ids[k]]
         send = torch.cat(send_parts) if send_parts else \
             torch.empty(0, dtype=torch.long)
-        my_sizes = [col_inputs[i].numel() for i in plan.rank_input_ids[self.rank]]
-        out_splits = [sum(my_sizes)] * W
         recv = comm.all_to_all_single(send, out_splits, in_splits)
         # recv: [W, sum(my_sizes)] -> per pair [W*b, ...]
         recv = recv.view(W, -1) if recv.numel() else recv.view(W, 0)
