@@ -714,24 +714,33 @@ class DistributedEmbedding(nn.Module):
             shard = shards[tbl_local]
             layer = self.row_layers[tbl_local]
             if isinstance(x, Ragged):
-                raise NotImplementedError("ragged inputs are not supported for "
-                                          "row-sliced tables")
-            gathered = comm.all_gather(x)            # [W*b, ...]
-            ids = gathered - shard.row_offset        # negative => OOB => zero row
-            out = layer(ids)                         # [W*b, (h,) D]
+                # ragged row-slice (beyond the reference, which is dense-only
+                # here): allgather lengths + values, lookup with offset
+                # (OOB contributes zero), reduce-scatter the combined rows.
+                lens_parts = comm.all_gather_uneven(x.row_lengths())
+                vals_parts = comm.all_gather_uneven(x.values)
+                g_ragged = Ragged.from_row_lengths(
+                    torch.cat(vals_parts) - shard.row_offset,
+                    torch.cat(lens_parts))
+                out = layer(g_ragged)                # [W*b, D]
+            else:
+                gathered = comm.all_gather(x)        # [W*b, ...]
+                ids = gathered - shard.row_offset    # negative => OOB => zero row
+                out = layer(ids)                     # [W*b, (h,) D]
             outs.append(out.reshape(W, -1))
         fused = torch.cat(outs, dim=1).reshape(-1)   # [W * sum(b*c)]
         red = comm.reduce_scatter(fused)             # [sum(b*c)]
         # split back per input
         result, pos = [], 0
         for j, x in enumerate(row_inputs):
-            b = x.shape[0]
+            b = x.nrows if isinstance(x, Ragged) else x.shape[0]
             n = outs[j].shape[1]
             flat = red[pos:pos + n]
             pos += n
             tbl_local = plan.input_maps[2][j]
             width = self.row_layers[tbl_local].output_dim
-            if self.row_layers[tbl_local].combiner is None and x.dim() > 1:
+            if (self.row_layers[tbl_local].combiner is None
+                    and not isinstance(x, Ragged) and x.dim() > 1):
                 out_t = flat.view(b, *x.shape[1:], width)
             else:
                 out_t = flat.view(b, width)

@@ -366,3 +366,27 @@ def _rowslice_dtype_worker(rank, world):
 def test_row_slice_output_dtype_world2():
     results = run_distributed(_rowslice_dtype_worker, world=2)
     assert max(results) < 0.05
+
+
+def _ragged_rowslice_worker(rank, world):
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd import Ragged
+    model = de.DistributedEmbedding([de.TableConfig(300, 8, "sum")],
+                                    row_slice_threshold=1)
+    weights = _ref_weights([300], 8)
+    model.set_weights([w.numpy() for w in weights])
+    all_lists = [[1, 2, 250], [7], [299, 0], [42, 42, 42, 99],
+                 [280], [5, 260], [150], [3, 297]]
+    my = all_lists[rank * 4:(rank + 1) * 4]
+    outs = model([Ragged.from_lists(my)])
+    w = weights[0]
+    errs = []
+    for i, row in enumerate(my):
+        ref = w[torch.tensor(row)].sum(0)
+        errs.append(float((outs[0][i] - ref).abs().max()))
+    return max(errs)
+
+
+def test_ragged_row_slice_world2():
+    results = run_distributed(_ragged_rowslice_worker, world=2)
+    assert max(results) < 1e-5
