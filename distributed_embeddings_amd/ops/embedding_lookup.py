@@ -280,6 +280,6 @@ def embedding_lookup(
 
     if ids.dim() != 2:
         raise ValueError(f"Dense ids with combiner must be 2-D, got {ids.dim()}-D")
-    if ids.shape[1] == 1:
-        return weight.index_select(0, ids.reshape(-1))
+    # hotness-1 included: the CSR path keeps the sparse-grad (IndexedSlices)
+    # contract and the OOB->zero semantics on every combiner lookup.
     return _dense_fixed_hotness(weight, ids, combiner)
