@@ -11,9 +11,12 @@
 //     loads; widths >256 loop over 256-element chunks.  Grid-stride over rows.
 //   * row_to_split        — COO rows -> CSR splits by per-row binary search
 //     (K4 equivalent).
-//   * csr_lookup_backward — rowid expansion + radix sort (rocPRIM bring-up,
-//     SURVEY.md §2.2 note) + hand-written head-flag scan unique + segmented
-//     sum reusing the forward kernel with per-id weights (K5-K7 equivalent).
+//   * csr_lookup_backward — rowid expansion + packed (id<<32|pos) keys-only
+//     radix sort (hand-written, radix_sort.hip; rocPRIM via
+//     DE_USE_ROCPRIM_SORT=1) + head-flag scan unique + segmented sum reusing
+//     the forward kernels with per-id weights (K5-K7 equivalent).
+//   * csr_fused_optimizer_apply — the same pipeline with the SGD/Adagrad
+//     update applied in place (no host sync; hipGraph-capturable).
 //   * integer_lookup      — open-addressing (linear probe) int64 hash
 //     resident in torch buffers; two-kernel design (free-slot scan, then
 //     insert_and_find with device-scope 64-bit atomicCAS) replacing the

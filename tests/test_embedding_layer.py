@@ -211,3 +211,32 @@ def test_bf16_fused_optimizer_cpu(seed):
         out.sum().backward()
         assert e.weight.grad is None
         assert not torch.equal(e.weight, w0), method
+
+
+def test_embedding_forward_property(seed):
+    """Property test: random shapes/dtypes/combiners vs a plain oracle."""
+    from hypothesis import given, settings, strategies as st
+    from distributed_embeddings_amd import Embedding
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.integers(1, 200), st.sampled_from([1, 3, 8, 17, 64, 96, 130]),
+           st.sampled_from([None, "sum", "mean"]),
+           st.integers(1, 12), st.integers(1, 6), st.booleans())
+    def check(vocab, width, combiner, batch, hot, use_bf16):
+        dtype = torch.bfloat16 if use_bf16 else torch.float32
+        e = Embedding(vocab, width, combiner, dtype=dtype)
+        if combiner is None:
+            ids = torch.randint(0, vocab, (batch,))
+            out = e(ids)
+            ref = e.weight[ids]
+            assert out.shape == (batch, width)
+        else:
+            ids = torch.randint(0, vocab, (batch, hot))
+            out = e(ids)
+            w = e.weight.float()
+            ref = w[ids].sum(1) if combiner == "sum" else w[ids].mean(1)
+            assert out.shape == (batch, width)
+        assert torch.allclose(out.float(), ref.float(), atol=1e-4), \
+            (vocab, width, combiner, batch, hot, use_bf16)
+
+    check()
