@@ -149,3 +149,25 @@ def test_bf16_table_dtype_module():
     assert all(torch.isfinite(o.float()).all() for o in outs)
     got = model.get_weights()
     assert np.allclose(got[0], weights[0], atol=0.01)  # bf16 rounding
+
+
+def test_checkpoint_roundtrip_property():
+    """Property: set_weights(get_weights()) is identity for random plans."""
+    from hypothesis import given, settings, strategies as st
+    import distributed_embeddings_amd as de
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.lists(st.integers(2, 300), min_size=1, max_size=6),
+           st.sampled_from([8, 16, 32]),
+           st.sampled_from(["basic", "memory_balanced", "memory_optimized"]))
+    def check(sizes, width, strategy):
+        model = de.DistributedEmbedding(
+            [de.TableConfig(s, width) for s in sizes], strategy=strategy)
+        rng = np.random.RandomState(0)
+        weights = [rng.randn(s, width).astype(np.float32) for s in sizes]
+        model.set_weights(weights)
+        got = model.get_weights()
+        for w, g in zip(weights, got):
+            assert np.array_equal(w, g)
+
+    check()
