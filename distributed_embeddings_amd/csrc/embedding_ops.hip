@@ -9,6 +9,9 @@
 //     Wave64 tiling: width<=64 uses sub-wave tiles (64/TILE rows per wave),
 //     width>64 uses one wave per row with per-lane vectorized float4/float2
 //     loads; widths >256 loop over 256-element chunks.  Grid-stride over rows.
+//     Power-law skew: rows longer than an adaptive threshold are pushed to a
+//     device-side list, expanded to exact (row, 128-id chunk) work items, and
+//     reduced by a second kernel with atomic combine (no host sync).
 //   * row_to_split        — COO rows -> CSR splits by per-row binary search
 //     (K4 equivalent).
 //   * csr_lookup_backward — rowid expansion + packed (id<<32|pos) keys-only

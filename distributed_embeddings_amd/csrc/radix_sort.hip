@@ -1,19 +1,22 @@
-// Hand-written LSD radix sort for (int64 key, int32 payload) pairs, tuned for
-// CDNA4 wave64 — replaces rocPRIM in the sparse-backward pipeline
-// ("hand-written LDS-staged segmented sort ... with rocPRIM
-// only as a bring-up crutch").
+// Hand-written LSD radix sort, tuned for CDNA4 wave64 — replaces rocPRIM in
+// the sparse-backward pipeline ("hand-written LDS-staged
+// segmented sort ... with rocPRIM only as a bring-up crutch").
 //
-// 8-bit digits, ceil(end_bit/8) passes.  Each pass:
-//   1. histogram: per-block bucket counts (LDS), written digit-major
+// The pipeline packs (id << 32 | position) into ONE u64 key, so the sort
+// moves a single array; digits start at bit 32 (custom_radix_sort_keys,
+// begin_bit/end_bit args).  8-bit digits, one pass per digit:
+//   1. rs_histogram_k: per-block bucket counts (LDS), written digit-major
 //      [bucket][block] for the stable global scan;
-//   2. exclusive scan over the [256 x nblocks] table (two hand-written scan
-//      kernels: block partials + single-block top scan + add-back);
-//   3. stable scatter: ONE wave per block walks its tile in 64-element
-//      groups; within a group the stable rank comes from 8 rounds of
-//      __ballot (bucket-bit match masks) + popcount over the preceding-lane
-//      mask; per-bucket running offsets live in LDS.
-// Keys must be non-negative (the pipeline masks OOB ids to the `vocab`
-// sentinel before sorting).
+//   2. exclusive scan over the [256 x nblocks] table (hand-written:
+//      per-chunk partials + single-wave top scan + add-back);
+//   3. rs_scatter_k: 4 waves per block, each owning a consecutive quarter of
+//      the 4096-element tile with a PRIVATE running-offset row (barrier-free
+//      scatter loop); within each 64-element group the stable rank comes
+//      from 8 __ballot rounds (bucket-bit match masks) + popcount over the
+//      preceding-lane mask.  Per-wave bases fold in earlier quarters' counts
+//      so the sort stays stable.
+// Keys' digit bits must be non-negative ids (the pipeline masks OOB ids to
+// the `vocab` sentinel before packing).
 
 #include <hip/hip_runtime.h>
 
