@@ -130,6 +130,15 @@ class DistributedEmbedding(nn.Module):
                       initializer=configs[t].initializer, dtype=table_dtype)
             for t in plan.dp_table_ids
         ])
+        # Preserve weights of layer instances handed to us (PyTorch modules
+        # have materialized weights at construction, unlike unbuilt Keras
+        # layers — same contract as the col-group copy below).
+        with torch.no_grad():
+            for local_t, t in enumerate(plan.dp_table_ids):
+                src = embeddings[t] if isinstance(
+                    embeddings[t], (Embedding, nn.Embedding)) else None
+                if src is not None:
+                    self.dp_layers[local_t].weight.copy_(src.weight)
 
         # ---- column/table-parallel fused variables ----
         # The fused group runs at EVERY world size (world==1 included): one
@@ -183,6 +192,12 @@ class DistributedEmbedding(nn.Module):
             lyr._oob_zero = True
             for p in lyr.parameters():
                 p.de_local = True
+            src = embeddings[shard.table_id] if isinstance(
+                embeddings[shard.table_id], (Embedding, nn.Embedding)) else None
+            if src is not None and shard.rows > 0:
+                with torch.no_grad():
+                    lyr.weight[:shard.rows].copy_(
+                        src.weight[shard.row_offset:shard.row_offset + shard.rows])
             row_layers.append(lyr)
         self.row_layers = nn.ModuleList(row_layers)
 

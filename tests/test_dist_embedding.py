@@ -390,3 +390,68 @@ def _ragged_rowslice_worker(rank, world):
 def test_ragged_row_slice_world2():
     results = run_distributed(_ragged_rowslice_worker, world=2)
     assert max(results) < 1e-5
+
+
+def test_layer_instances_preserve_weights_world1():
+    """Passing pre-initialized layer instances preserves their weights in
+    every placement mode (dp / column-group / row-slice).  PyTorch modules
+    have materialized weights at construction (unlike unbuilt Keras layers),
+    so DistributedEmbedding adopts them instead of re-initializing."""
+    import torch.nn as nn
+    import distributed_embeddings_amd as de
+    torch.manual_seed(0)
+    sizes = [8, 400, 900]
+    layers = [de.Embedding(sizes[0], 16),          # -> dp
+              nn.Embedding(sizes[1], 16),          # -> col group
+              de.Embedding(sizes[2], 16)]          # -> row slice
+    model = de.DistributedEmbedding(
+        layers, data_parallel_threshold=8 * 16,
+        row_slice_threshold=900 * 16)
+    plan = model.strategy
+    assert plan.dp_table_ids and plan.col_table_ids and plan.row_table_ids
+    got = model.get_weights()
+    for lyr, w in zip(layers, got):
+        assert torch.allclose(lyr.weight.detach(), torch.as_tensor(w))
+
+
+def _preserve_weights_worker(rank, world):
+    import torch.nn as nn
+    import distributed_embeddings_amd as de
+    torch.manual_seed(42)  # same layers on every rank (user contract)
+    sizes = [8, 400, 900]
+    layers = [de.Embedding(sizes[0], 16), nn.Embedding(sizes[1], 16),
+              de.Embedding(sizes[2], 16)]
+    srcs = [l.weight.detach().clone() for l in layers]
+    model = de.DistributedEmbedding(
+        layers, strategy="memory_balanced",
+        data_parallel_threshold=8 * 16, row_slice_threshold=900 * 16)
+    got = model.get_weights(all_ranks=True)
+    return [float((s - torch.as_tensor(w)).abs().max())
+            for s, w in zip(srcs, got)]
+
+
+def test_layer_instances_preserve_weights_world2():
+    results = run_distributed(_preserve_weights_worker, world=2)
+    for errs in results:
+        assert max(errs) == 0.0
+
+
+def _single_table_worker(rank, world):
+    # one table, world 2: one rank holds the table, the other has NO local
+    # col layers — exercises the empty-rank a2a splits and output reassembly.
+    import distributed_embeddings_amd as de
+    model = de.DistributedEmbedding([de.TableConfig(50, 8, "sum")])
+    weights = _ref_weights([50], 8)
+    model.set_weights([w.numpy() for w in weights])
+    inputs = _make_inputs([50], 4, hotness=3, world=world)
+    local = [x[rank * 4:(rank + 1) * 4] for x in inputs]
+    outs = model(local)
+    loss = sum((o * o).sum() for o in outs)
+    loss.backward()
+    ref = weights[0][inputs[0][rank * 4:(rank + 1) * 4]].sum(1)
+    return float((outs[0] - ref).abs().max())
+
+
+def test_single_table_world2():
+    results = run_distributed(_single_table_worker, world=2)
+    assert max(results) < 1e-5
