@@ -455,3 +455,86 @@ def _single_table_worker(rank, world):
 def test_single_table_world2():
     results = run_distributed(_single_table_worker, world=2)
     assert max(results) < 1e-5
+
+
+# --- direct unit tests of the dp->mp input redistribution helpers
+#     (parity: reference dist_model_parallel_test.py:741-831) ---
+
+def _redistribute_dense_worker(rank, world):
+    import distributed_embeddings_amd as de
+    sizes = [40, 50, 60]
+    model = de.DistributedEmbedding(
+        [de.TableConfig(s, 8, "sum") for s in sizes])
+    full = [torch.randint(0, sizes[0], (world * 4,),
+                          generator=torch.Generator().manual_seed(1)),
+            torch.randint(0, sizes[1], (world * 4, 3),
+                          generator=torch.Generator().manual_seed(2)),
+            torch.randint(0, sizes[2], (world * 4, 2),
+                          generator=torch.Generator().manual_seed(3))]
+    local = [x[rank * 4:(rank + 1) * 4] for x in full]
+    got = model._dp_to_mp_dense(local)
+    errs = []
+    for j, i in enumerate(model.strategy.rank_input_ids[rank]):
+        # pair j must hold input i's GLOBAL batch in rank order
+        errs.append(int((got[j] != full[i]).sum()))
+    return errs
+
+
+def test_dp_to_mp_dense_direct_world2():
+    results = run_distributed(_redistribute_dense_worker, world=2)
+    for errs in results:
+        assert errs and sum(errs) == 0
+
+
+def _redistribute_ragged_worker(rank, world):
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd import Ragged
+    sizes = [40, 50]
+    model = de.DistributedEmbedding(
+        [de.TableConfig(s, 8, "sum") for s in sizes])
+    all_lists = [[1, 2, 3], [7], [39, 0], [4, 4, 4, 9],
+                 [30], [5, 6], [15], [3, 37]]
+    dense_full = torch.randint(0, sizes[1], (world * 4, 2),
+                               generator=torch.Generator().manual_seed(4))
+    local = [Ragged.from_lists(all_lists[rank * 4:(rank + 1) * 4]),
+             dense_full[rank * 4:(rank + 1) * 4]]
+    got = model._dp_to_mp_ragged(local, None)
+    errs = []
+    for j, i in enumerate(model.strategy.rank_input_ids[rank]):
+        if i == 0:
+            ref = Ragged.from_lists(all_lists)
+            errs.append(int((got[j].values != ref.values).sum()) +
+                        int((got[j].row_splits != ref.row_splits).sum()))
+        else:
+            errs.append(int((got[j] != dense_full).sum()))
+    return errs
+
+
+def test_dp_to_mp_ragged_direct_world2():
+    results = run_distributed(_redistribute_ragged_worker, world=2)
+    for errs in results:
+        assert errs and sum(errs) == 0
+
+
+def _redistribute_unbalanced_worker(rank, world):
+    # shared table: BOTH inputs land on the table's single rank; the other
+    # rank serves zero pairs (reference "all features to rank 0" case :824).
+    import distributed_embeddings_amd as de
+    model = de.DistributedEmbedding([de.TableConfig(40, 8, "sum")],
+                                    input_table_map=[0, 0])
+    full = [torch.randint(0, 40, (world * 4, h),
+                          generator=torch.Generator().manual_seed(h))
+            for h in (1, 3)]
+    local = [x[rank * 4:(rank + 1) * 4] for x in full]
+    got = model._dp_to_mp_dense(local)
+    mine = model.strategy.rank_input_ids[rank]
+    assert len(got) == len(mine)
+    return [int((got[j] != full[i]).sum()) for j, i in enumerate(mine)]
+
+
+def test_dp_to_mp_unbalanced_world2():
+    results = run_distributed(_redistribute_unbalanced_worker, world=2)
+    # one rank serves both pairs, the other none — both outcomes valid
+    assert any(len(e) == 2 for e in results)
+    for errs in results:
+        assert sum(errs) == 0
