@@ -84,7 +84,7 @@ def main():
             loss = loss_fn(model(batch), lb)
             loss.backward()
             opt.step()
-        print(f"epoch {epoch}: loss {float(loss):.4f}, "
+        print(f"epoch {epoch}: loss {float(loss.detach()):.4f}, "
               f"vocab[0] size {model.lookups[0].vocabulary_size()}")
 
 

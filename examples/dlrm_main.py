@@ -136,7 +136,7 @@ def main():
         opt.step()
         sched.step()
         if rank == 0 and step % 50 == 0:
-            l = float(loss)
+            l = float(loss.detach())
             print(f"step {step:5d} loss {l:.4f} "
                   f"({args.batch_size * (step + 1) / (time.time() - t0):.0f} samples/s)")
 

@@ -143,7 +143,7 @@ def main():
     for i in range(args.warmup):
         step(i)
     loss = step(0)
-    lt = torch.tensor([float(loss)])
+    lt = torch.tensor([float(loss.detach())])
     de.comm.allreduce_sum_(lt)  # sync + flush (parity: reference main.py:140-158)
 
     if device.type == "cuda":
@@ -151,7 +151,7 @@ def main():
     t0 = time.time()
     for i in range(args.num_steps):
         loss = step(i)
-    lt = torch.tensor([float(loss)])
+    lt = torch.tensor([float(loss.detach())])
     de.comm.allreduce_sum_(lt)
     if device.type == "cuda":
         torch.cuda.synchronize()

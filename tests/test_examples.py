@@ -1,0 +1,50 @@
+"""Example-script smoke tests (CPU, tiny configs).
+
+The reference ships its examples untested; here each example is exercised
+end-to-end with tiny arguments so the scripts cannot bit-rot (reference
+parity targets: examples/dlrm/main.py, examples/benchmarks/synthetic_models/
+main.py, examples/criteo/main.py, examples/benchmarks/benchmark.py).
+"""
+
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(script, args, timeout=240):
+    proc = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "examples", script)] + args,
+        capture_output=True, text=True, timeout=timeout, cwd=ROOT)
+    assert proc.returncode == 0, (
+        f"{script} failed:\n{proc.stdout[-2000:]}\n{proc.stderr[-2000:]}")
+    return proc.stdout
+
+
+def test_dlrm_example_smoke():
+    out = _run("dlrm_main.py", ["--batch-size", "128", "--num-batches", "3",
+                                "--embedding-dim", "16",
+                                "--table-size-cap", "1000"])
+    assert "loss" in out
+
+
+def test_synthetic_benchmark_smoke():
+    out = _run("synthetic_benchmark.py",
+               ["--model", "tiny", "--batch-size", "256",
+                "--num-steps", "3", "--warmup", "1"])
+    assert "ms/iteration" in out
+
+
+def test_integer_lookup_example_smoke():
+    out = _run("criteo_integer_lookup.py", ["--rows", "500", "--epochs", "1"])
+    assert "vocab" in out
+
+
+def test_lookup_benchmark_smoke():
+    out = _run("lookup_benchmark.py",
+               ["--vocab", "2000", "--batch", "64", "--width", "16",
+                "--max-hotness", "5"])
+    assert "fwd" in out
