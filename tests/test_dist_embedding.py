@@ -538,3 +538,29 @@ def test_dp_to_mp_unbalanced_world2():
     assert any(len(e) == 2 for e in results)
     for errs in results:
         assert sum(errs) == 0
+
+
+def _single_table_ragged_worker(rank, world):
+    # ragged inputs with ONE table at world 2: the rank without the table
+    # exercises the zero-pair branch of _dp_to_mp_ragged and the empty
+    # reassembly of the output a2a.
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd import Ragged
+    model = de.DistributedEmbedding([de.TableConfig(50, 8, "sum")])
+    weights = _ref_weights([50], 8)
+    model.set_weights([w.numpy() for w in weights])
+    all_lists = [[1, 2], [3], [4, 5, 6], [7], [20], [21, 22], [23], [24]]
+    my = all_lists[rank * 4:(rank + 1) * 4]
+    outs = model([Ragged.from_lists(my)])
+    loss = (outs[0] * outs[0]).sum()
+    loss.backward()
+    errs = []
+    for i, row in enumerate(my):
+        ref = weights[0][torch.tensor(row)].sum(0)
+        errs.append(float((outs[0][i] - ref).abs().max()))
+    return max(errs)
+
+
+def test_single_table_ragged_world2():
+    results = run_distributed(_single_table_ragged_worker, world=2)
+    assert max(results) < 1e-5
