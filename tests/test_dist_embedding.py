@@ -564,3 +564,27 @@ def _single_table_ragged_worker(rank, world):
 def test_single_table_ragged_world2():
     results = run_distributed(_single_table_ragged_worker, world=2)
     assert max(results) < 1e-5
+
+
+@pytest.mark.parametrize("strategy,kwargs,hotness,combiner", [
+    ("memory_balanced", {}, 1, None),
+    ("basic", {"column_slice_threshold": 64 * 100}, 4, "sum"),
+    ("basic", {"row_slice_threshold": 1}, 4, "sum"),
+])
+def test_forward_equivalence_world4(strategy, kwargs, hotness, combiner):
+    """World-4 equivalence: de-risks the driver's N=4/8 scaling run (all
+    other distributed tests are world=2)."""
+    table_sizes = [67, 130, 259, 40, 91]
+    width = 16
+    world = 4
+    results = run_distributed(
+        _dist_forward_backward, world=world,
+        args=(table_sizes, width, strategy, kwargs, hotness, combiner))
+    ref_outs, _ = _single_forward_backward(table_sizes, width, hotness,
+                                           combiner, world)
+    for rank in range(world):
+        outs = results[rank]["outs"]
+        for t, o in enumerate(outs):
+            ref = ref_outs[t][rank * 4:(rank + 1) * 4]
+            assert torch.allclose(o, ref, atol=1e-5), \
+                f"rank {rank} table {t}: max err {(o - ref).abs().max()}"
