@@ -116,3 +116,21 @@ def test_broadcast_on_first_step_world2():
     for r in range(2):
         assert torch.equal(results[r]["first"], torch.zeros(4, 4))
         assert float(results[r]["after"][0, 0]) == r + 10
+
+
+def test_dlrm_world4_matches_world1():
+    """World-4 hybrid dp+mp DLRM == single process (the CPU proxy for the
+    driver's 8-GPU scaling bench; B=8 -> 2 samples/rank)."""
+    results4 = run_distributed(_dlrm_worker, world=4, args=(True, None))
+    results1 = run_distributed(_dlrm_worker, world=1, args=(True, None))
+    full_out = results1[0]["out"]
+    for rank in range(4):
+        got = results4[rank]["out"]
+        ref = full_out[rank * 2:(rank + 1) * 2]
+        assert torch.allclose(got, ref, atol=1e-4), \
+            f"rank{rank} fwd err {(got - ref).abs().max()}"
+    for t in range(len(SIZES)):
+        a = results4[0]["tables"][t]
+        b = results1[0]["tables"][t]
+        assert torch.allclose(a, b, atol=1e-4), \
+            f"table {t} err {(a - b).abs().max()}"
