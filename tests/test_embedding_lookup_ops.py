@@ -130,3 +130,43 @@ def test_combiner_validation():
         embedding_lookup(weight, torch.zeros(3, dtype=torch.long), "max")
     with pytest.raises(ValueError):
         embedding_lookup(weight, Ragged.from_lists([[1]]), None)
+
+
+def test_representation_consistency_property():
+    """Property: the same logical bags expressed as fixed-hotness dense,
+    ragged, and sparse COO produce identical outputs AND identical weight
+    gradients through the dispatcher."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.integers(0, 10_000), st.integers(1, 5), st.integers(1, 6),
+           st.sampled_from(["sum", "mean"]))
+    def check(seed, batch, hotness, combiner):
+        g = torch.Generator().manual_seed(seed)
+        vocab, width = 30, 8
+        dense = torch.randint(0, vocab, (batch, hotness), generator=g)
+        lists = [row.tolist() for row in dense]
+
+        def run(make_input):
+            w = torch.randn(vocab, width, generator=torch.Generator()
+                            .manual_seed(seed + 1), requires_grad=True)
+            out = embedding_lookup(w, make_input(), combiner=combiner)
+            out.sum().backward()
+            grad = w.grad
+            if grad.is_sparse:
+                grad = grad.to_dense()
+            return out.detach(), grad
+
+        o_dense, g_dense = run(lambda: dense)
+        o_ragged, g_ragged = run(lambda: Ragged.from_lists(lists))
+        idx = torch.tensor([[b, j] for b in range(batch)
+                            for j in range(hotness)]).T
+        o_sparse, g_sparse = run(lambda: torch.sparse_coo_tensor(
+            idx, dense.reshape(-1), (batch, hotness)).coalesce())
+
+        for o in (o_ragged, o_sparse):
+            assert torch.allclose(o_dense, o, atol=1e-5)
+        for gr in (g_ragged, g_sparse):
+            assert torch.allclose(g_dense, gr, atol=1e-5)
+
+    check()
