@@ -134,3 +134,21 @@ def test_dlrm_world4_matches_world1():
         b = results1[0]["tables"][t]
         assert torch.allclose(a, b, atol=1e-4), \
             f"table {t} err {(a - b).abs().max()}"
+
+
+def test_dlrm_world8_matches_world1():
+    """World-8 (the driver's max scaling point): one sample per rank."""
+    results8 = run_distributed(_dlrm_worker, world=8, args=(True, None),
+                               timeout=300)
+    results1 = run_distributed(_dlrm_worker, world=1, args=(True, None))
+    full_out = results1[0]["out"]
+    for rank in range(8):
+        got = results8[rank]["out"]
+        ref = full_out[rank:rank + 1]
+        assert torch.allclose(got, ref, atol=1e-4), \
+            f"rank{rank} fwd err {(got - ref).abs().max()}"
+    for t in range(len(SIZES)):
+        a = results8[0]["tables"][t]
+        b = results1[0]["tables"][t]
+        assert torch.allclose(a, b, atol=1e-4), \
+            f"table {t} err {(a - b).abs().max()}"
