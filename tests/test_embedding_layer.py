@@ -240,3 +240,34 @@ def test_embedding_forward_property(seed):
             (vocab, width, combiner, batch, hot, use_bf16)
 
     check()
+
+
+def test_sparse_optimizer_state_dict_roundtrip(seed):
+    """Standard torch optimizer checkpointing: save/load state_dict mid-run
+    must continue identically to an uninterrupted run (adagrad state)."""
+    from distributed_embeddings_amd import Embedding
+    from distributed_embeddings_amd.parallel.optim import SparseEmbeddingOptimizer
+
+    def train(n_steps, reload_at=None):
+        torch.manual_seed(seed)
+        emb = Embedding(50, 8, combiner="sum")
+        opt = SparseEmbeddingOptimizer(emb.parameters(), lr=0.1,
+                                       method="adagrad")
+        g = torch.Generator().manual_seed(3)
+        for i in range(n_steps):
+            if i == reload_at:
+                sd_o, sd_m = opt.state_dict(), emb.state_dict()
+                emb = Embedding(50, 8, combiner="sum")
+                emb.load_state_dict(sd_m)
+                opt = SparseEmbeddingOptimizer(emb.parameters(), lr=0.1,
+                                               method="adagrad")
+                opt.load_state_dict(sd_o)
+            ids = torch.randint(0, 50, (16, 3), generator=g)
+            opt.zero_grad()
+            (emb(ids) ** 2).sum().backward()
+            opt.step()
+        return emb.weight.detach()
+
+    w_plain = train(6)
+    w_reload = train(6, reload_at=3)
+    assert torch.allclose(w_plain, w_reload, atol=1e-6)
