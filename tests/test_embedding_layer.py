@@ -271,3 +271,28 @@ def test_sparse_optimizer_state_dict_roundtrip(seed):
     w_plain = train(6)
     w_reload = train(6, reload_at=3)
     assert torch.allclose(w_plain, w_reload, atol=1e-6)
+
+
+def test_fused_adagrad_state_in_state_dict(seed):
+    """The fused optimizer's accumulator is a registered buffer, so module
+    state_dict checkpointing resumes fused-Adagrad training exactly."""
+    from distributed_embeddings_amd import Embedding
+
+    def train(n_steps, reload_at=None):
+        torch.manual_seed(seed)
+        emb = Embedding(50, 8, combiner="sum")
+        emb.enable_fused_optimizer("adagrad", lr=0.1)
+        g = torch.Generator().manual_seed(5)
+        for i in range(n_steps):
+            if i == reload_at:
+                sd = emb.state_dict()
+                emb = Embedding(50, 8, combiner="sum")
+                emb.enable_fused_optimizer("adagrad", lr=0.1)
+                emb.load_state_dict(sd)
+            ids = torch.randint(0, 50, (16, 3), generator=g)
+            (emb(ids) ** 2).sum().backward()   # update applied in backward
+        return emb.weight.detach()
+
+    w_plain = train(6)
+    w_reload = train(6, reload_at=3)
+    assert torch.allclose(w_plain, w_reload, atol=1e-6)
