@@ -46,7 +46,7 @@ def barrier(group=None):
         dist.barrier(group)
 
 
-def _as_list(x, n):
+def _as_list(x):
     return None if x is None else [int(v) for v in x]
 
 
@@ -82,8 +82,8 @@ def all_to_all_single(
     """Fused all-to-all along dim 0 with optional uneven per-peer splits."""
     if world_size(group) == 1:
         return inp
-    return _AllToAllSingle.apply(inp, _as_list(output_split_sizes, None),
-                                 _as_list(input_split_sizes, None), group)
+    return _AllToAllSingle.apply(inp, _as_list(output_split_sizes),
+                                 _as_list(input_split_sizes), group)
 
 
 class _AllGather(torch.autograd.Function):
