@@ -34,9 +34,8 @@ Every rank computes the same global plan deterministically (parity:
 import dataclasses
 from typing import Callable, Dict, List, Optional, Sequence, Tuple
 
-# Default thresholds sized for 288 GB HBM3E per MI355X (fp32 elements).
-DEFAULT_DP_THRESHOLD = 0  # no tables data-parallel unless asked
-DEFAULT_ROW_SLICE_THRESHOLD = None
+# Threshold defaults (None everywhere) are sized for 288 GB HBM3E per
+# MI355X: no table is data-parallel, row-sliced, or offloaded unless asked.
 
 
 @dataclasses.dataclass
