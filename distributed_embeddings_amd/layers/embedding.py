@@ -95,6 +95,9 @@ class Embedding(nn.Module):
         update this weight (its ``.grad`` stays ``None``)."""
         if method not in ("sgd", "adagrad"):
             raise ValueError(f"unknown fused optimizer {method!r}")
+        for name in ("_fused_lr", "_fused_state"):  # allow re-configuring
+            if name in self._buffers:
+                del self._buffers[name]
         self.register_buffer("_fused_lr",
                              torch.tensor([float(lr)], dtype=torch.float32,
                                           device=self.weight.device))
@@ -168,9 +171,8 @@ class Embedding(nn.Module):
                 raise ValueError("Sparse input requires a combiner")
             return embedding_lookup(self.weight, ids, self.combiner)
 
-        if torch.is_floating_point(ids):
-            ids = ids.long()  # parity: reference casts non-int inputs (embedding.py:121-123)
-        elif ids.dtype != torch.long:
+        if ids.dtype != torch.long:
+            # parity: reference casts non-int inputs (embedding.py:121-123)
             ids = ids.long()
 
         if self.combiner is None:
@@ -181,7 +183,14 @@ class Embedding(nn.Module):
                              "(parity: reference embedding.py:133-135)")
         lead_shape = ids.shape[:-1]
         flat2d = ids.reshape(-1, ids.shape[-1])
-        if self._oob_zero or flat2d.shape[1] > 1:
+        if getattr(self, "_fused_lr", None) is not None and self.training:
+            # fused in-backward optimizer: dense [b, h] as CSR so the update
+            # applies here too (not only for Ragged inputs)
+            b, h = flat2d.shape
+            splits = torch.arange(b + 1, device=flat2d.device,
+                                  dtype=torch.long) * h
+            out = self.csr_lookup(flat2d.reshape(-1), splits, self.combiner)
+        elif self._oob_zero or flat2d.shape[1] > 1:
             out = embedding_lookup(self.weight, flat2d, self.combiner)
         else:
             out = self._gather(flat2d.reshape(-1))

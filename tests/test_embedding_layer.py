@@ -296,3 +296,40 @@ def test_fused_adagrad_state_in_state_dict(seed):
     w_plain = train(6)
     w_reload = train(6, reload_at=3)
     assert torch.allclose(w_plain, w_reload, atol=1e-6)
+
+
+def test_enable_fused_optimizer_reconfigure(seed):
+    """enable_fused_optimizer may be called again (e.g. sgd -> adagrad)."""
+    from distributed_embeddings_amd import Embedding
+    emb = Embedding(20, 8, combiner="sum")
+    emb.enable_fused_optimizer("sgd", 0.1)
+    emb.enable_fused_optimizer("adagrad", 0.05)
+    assert emb._fused_method == "adagrad"
+    assert emb._fused_state.shape == emb.weight.shape
+    ids = torch.randint(0, 20, (4, 2))
+    (emb(ids) ** 2).sum().backward()
+    assert (emb._fused_state > 0).any()
+
+
+def test_fused_sgd_dense_input_matches_explicit(seed):
+    """Dense [b, hotness] inputs must hit the fused update too (they are
+    converted to CSR when a fused optimizer is enabled)."""
+    from distributed_embeddings_amd import Embedding, SparseEmbeddingOptimizer
+    w0 = torch.randn(40, 8)
+    ids = torch.randint(0, 40, (6, 3), generator=torch.Generator().manual_seed(seed))
+    up = torch.randn(6, 8)
+    for combiner in ("sum", "mean"):
+        e1 = Embedding(40, 8, combiner=combiner)
+        e2 = Embedding(40, 8, combiner=combiner)
+        with torch.no_grad():
+            e1.weight.copy_(w0); e2.weight.copy_(w0)
+        e1.enable_fused_sgd(0.1)
+        out1 = e1(ids)
+        out1.backward(up)
+        assert e1.weight.grad is None
+        o2 = SparseEmbeddingOptimizer(e2.parameters(), lr=0.1)
+        out2 = e2(ids)
+        out2.backward(up)
+        o2.step()
+        assert torch.allclose(out1, out2, atol=1e-6)
+        assert torch.allclose(e1.weight, e2.weight, atol=1e-6), combiner
