@@ -7,10 +7,11 @@ bottom MLP over dense features, per-category embedding tables behind
 interaction (lower triangle) re-concatenated with the bottom MLP output, and
 a top MLP producing one logit.
 
-MI355X notes: the MLPs run as plain ``nn.Linear`` (hipBLASLt GEMMs) under
-bf16 autocast; ``dot_interact`` is the fused-MFMA target (torch fallback here,
-HIP kernel in ``csrc/``; the batched 27x128 @ 128x27 pairwise dots are
-MFMA-shaped work — SURVEY.md §7 step 9).
+MI355X notes: the MLPs run as plain ``nn.Linear`` (hipBLASLt GEMMs, TunableOp
+selections shipped in profiles/) under bf16 autocast; the pairwise-dot
+interaction runs as one fused MFMA kernel each direction
+(``csrc/dot_interact.hip``, dispatched by ``ops/dot_interact.py``; plain
+bmm fallback when the shape or dtype gate fails).
 """
 
 import math

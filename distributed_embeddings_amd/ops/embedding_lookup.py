@@ -6,7 +6,7 @@ the same input-type routing (dense / ragged / sparse / fixed-hotness) and
 combiner semantics, re-designed for PyTorch + hand-written HIP kernels.
 
 The hot path is a CSR (values, row_splits) segmented gather-reduce.  On GPU it
-runs a hand-written CDNA4 kernel (``csrc/embedding_lookup.hip``); on CPU a pure
+runs a hand-written CDNA4 kernel (``csrc/embedding_ops.hip``); on CPU a pure
 PyTorch reference implementation with identical numerics is used (it is also
 the comparison oracle for the GPU numerics tests).
 
