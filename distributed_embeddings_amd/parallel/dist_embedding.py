@@ -601,9 +601,14 @@ class DistributedEmbedding(nn.Module):
                     outs[j] = part
         return outs
 
-    def _offset_vector(self, gi, spec, device):
-        """Per-element fused-table offsets (cached; None if all zero)."""
-        key = ("elem", gi, tuple(spec), str(device))
+    def _offset_vector(self, gi, spec, device, kind="elem"):
+        """Cached fused-table offset vector for ``spec`` = [(offset, count)].
+
+        ``kind='elem'``: one entry per id element; ``kind='row'``: one entry
+        per CSR row (expanded by repeat_interleave at the call site).
+        Returns None when every offset is zero.
+        """
+        key = (kind, gi, tuple(spec), str(device))
         cache = getattr(self, "_off_cache", None)
         if cache is None:
             cache = self._off_cache = {}
@@ -617,19 +622,7 @@ class DistributedEmbedding(nn.Module):
         return cache[key]
 
     def _row_offset_vector(self, gi, spec, device):
-        """Per-row fused-table offsets (cached; None if all zero)."""
-        key = ("row", gi, tuple(spec), str(device))
-        cache = getattr(self, "_off_cache", None)
-        if cache is None:
-            cache = self._off_cache = {}
-        if key not in cache:
-            if all(off == 0 for off, _ in spec):
-                cache[key] = None
-            else:
-                cache[key] = torch.cat([
-                    torch.full((n,), off, dtype=torch.long, device=device)
-                    for off, n in spec])
-        return cache[key]
+        return self._offset_vector(gi, spec, device, kind="row")
 
     def _call_table_parallel(self, col_inputs):
         plan = self.strategy
