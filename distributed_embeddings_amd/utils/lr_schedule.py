@@ -7,9 +7,13 @@ linear warmup to ``base_lr`` over ``warmup_steps``, constant until
 
 
 class WarmupPolyDecay:
+    """``fused_modules``: modules with ``set_fused_lr`` (e.g.
+    ``DistributedEmbedding`` with the in-backward fused optimizer enabled)
+    whose device-resident lr follows the schedule too."""
+
     def __init__(self, optimizer, base_lr: float, warmup_steps: int = 0,
                  decay_start: int = 0, decay_steps: int = 0, power: float = 2.0,
-                 end_lr: float = 0.0):
+                 end_lr: float = 0.0, fused_modules=()):
         self.opt = optimizer
         self.base_lr = base_lr
         self.warmup_steps = warmup_steps
@@ -17,6 +21,7 @@ class WarmupPolyDecay:
         self.decay_steps = decay_steps
         self.power = power
         self.end_lr = end_lr
+        self.fused_modules = list(fused_modules)
         self._step = 0
 
     def lr_at(self, step: int) -> float:
@@ -34,5 +39,7 @@ class WarmupPolyDecay:
             groups = self.opt.optimizer.param_groups
         for g in groups:
             g["lr"] = lr
+        for m in self.fused_modules:
+            m.set_fused_lr(lr)
         self._step += 1
         return lr

@@ -101,3 +101,21 @@ def test_synthetic_dlrm_data():
     assert len(batches) == 3
     num, cats, labels = batches[0]
     assert num.shape == (8, 4) and len(cats) == 2 and labels.shape == (8, 1)
+
+
+def test_warmup_poly_decay_updates_fused_lr():
+    import torch
+    from distributed_embeddings_amd import Embedding
+
+    class Opt:
+        param_groups = [{"lr": 0.0}]
+
+    emb = Embedding(10, 4, combiner="sum")
+    emb.enable_fused_sgd(1.0)
+    sched = WarmupPolyDecay(Opt(), base_lr=1.0, warmup_steps=4,
+                            fused_modules=[emb])
+    sched.step()
+    assert abs(float(emb._fused_lr) - 0.25) < 1e-6
+    for _ in range(3):
+        sched.step()
+    assert abs(float(emb._fused_lr) - 1.0) < 1e-6
