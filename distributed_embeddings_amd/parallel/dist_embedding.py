@@ -333,7 +333,9 @@ class DistributedEmbedding(nn.Module):
         plan = self.strategy
         out: List[Optional[np.ndarray]] = [None] * len(cfgs)
         for local_t, t in enumerate(plan.dp_table_ids):
-            out[t] = self.dp_layers[local_t].weight.detach().cpu().numpy().copy()
+            # .float(): checkpoints are fp32 numpy regardless of table_dtype
+            # (bf16 storage has no numpy dtype)
+            out[t] = self.dp_layers[local_t].weight.detach().float().cpu().numpy()
 
         # column slices: broadcast each slice from its owner, reassemble.
         device = self._comm_device()
@@ -354,7 +356,8 @@ class DistributedEmbedding(nn.Module):
         for t in plan.row_table_ids:
             local_t = plan.row_table_ids.index(t)
             rows = plan.row_shards[t][self.rank].rows
-            shard_w = self.row_layers[local_t].weight.detach()[:rows].to(device)
+            # fp32 for the gather + numpy conversion (bf16-safe on gloo too)
+            shard_w = self.row_layers[local_t].weight.detach()[:rows].float().to(device)
             parts = comm.all_gather_uneven(shard_w)
             out[t] = torch.cat([p.cpu() for p in parts], dim=0).numpy()
         return out

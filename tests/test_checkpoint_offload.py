@@ -171,3 +171,23 @@ def test_checkpoint_roundtrip_property():
             assert np.array_equal(w, g)
 
     check()
+
+
+def test_bf16_get_weights_all_modes():
+    """get_weights must work for bf16 tables in every placement mode
+    (bf16 tensors have no numpy dtype — conversion must go through fp32)."""
+    import distributed_embeddings_amd as de
+    sizes = [8, 400, 900]
+    model = de.DistributedEmbedding(
+        [de.TableConfig(s, 16) for s in sizes],
+        data_parallel_threshold=8 * 16, row_slice_threshold=900 * 16,
+        table_dtype=torch.bfloat16)
+    plan = model.strategy
+    assert plan.dp_table_ids and plan.col_table_ids and plan.row_table_ids
+    weights = [np.random.RandomState(t).randn(s, 16).astype(np.float32)
+               for t, s in enumerate(sizes)]
+    model.set_weights(weights)
+    got = model.get_weights()
+    for w, g in zip(weights, got):
+        assert g.dtype == np.float32
+        assert np.allclose(w, g, rtol=0.01, atol=0.01)  # bf16 rounding
