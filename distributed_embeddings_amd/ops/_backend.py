@@ -1,15 +1,13 @@
 """Loader for the hand-written HIP (gfx950) extension.
 
-The extension is built in-tree (``setup.py build_ext --inplace`` or
-``python -m distributed_embeddings_amd.build``) into
-``distributed_embeddings_amd/_hip_ops*.so``.  On a GPU machine the HIP path is
+The extension is built in-tree (``PYTORCH_ROCM_ARCH=gfx950 python setup.py
+build_ext --inplace``) into ``distributed_embeddings_amd/_hip_ops*.so``.  On a GPU machine the HIP path is
 mandatory: ops fail loudly if the extension is missing so a silent eager
 fallback can never masquerade as the native path.  On CPU-only machines the
 pure-PyTorch reference paths are used and the extension is not required.
 """
 
 import importlib
-import os
 
 
 _ext = None
