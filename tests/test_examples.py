@@ -48,3 +48,13 @@ def test_lookup_benchmark_smoke():
                ["--vocab", "2000", "--batch", "64", "--width", "16",
                 "--max-hotness", "5"])
     assert "fwd" in out
+
+
+def test_bench_contract_help():
+    """bench.py is the driver contract — it must always parse its args."""
+    proc = subprocess.run([sys.executable, os.path.join(ROOT, "bench.py"),
+                           "--help"], capture_output=True, text=True,
+                          timeout=120, cwd=ROOT)
+    assert proc.returncode == 0
+    for flag in ("--gpus", "--steps", "--warmup"):
+        assert flag in proc.stdout
