@@ -152,3 +152,49 @@ def test_dlrm_world8_matches_world1():
         b = results1[0]["tables"][t]
         assert torch.allclose(a, b, atol=1e-4), \
             f"table {t} err {(a - b).abs().max()}"
+
+
+def _synth_worker(rank, world):
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd.models.config import EmbeddingConfig, ModelConfig
+    from distributed_embeddings_amd.models.synthetic import SyntheticModel
+    from distributed_embeddings_amd.utils.input_gen import make_batch
+    from distributed_embeddings_amd.models.synthetic import expand_tables
+
+    cfg = ModelConfig(
+        name="unit",
+        embedding_configs=[
+            EmbeddingConfig(2, [1, 3], 50, 8, True),   # shared multi-hot
+            EmbeddingConfig(3, [1], 80, 8, False),
+        ],
+        mlp_sizes=[16, 8], num_numerical_features=4, interact_stride=2)
+    torch.manual_seed(11)
+    model = SyntheticModel(cfg, strategy="memory_balanced")
+    tables, input_map, hotness = expand_tables(cfg)
+    gw = torch.Generator().manual_seed(9)
+    weights = [torch.randn(r, w, generator=gw).numpy() for r, w in tables]
+    model.embeddings.set_weights(weights)
+    for p in model.mlp.parameters():
+        torch.nn.init.normal_(p, generator=gw) if p.dim() > 1 else p.data.zero_()
+    de.broadcast_parameters(model)
+
+    B = 8
+    gi = torch.Generator().manual_seed(13)
+    sizes = [tables[t][0] for t in input_map]
+    cats = make_batch(sizes, hotness, B, generator=gi, keep_hot_dim=True)
+    num = torch.rand(B, 4, generator=gi)
+    lb = B // world
+    sl = slice(rank * lb, (rank + 1) * lb)
+    out = model(num[sl], [c[sl] for c in cats]).detach()
+    return {"out": out}
+
+
+def test_synthetic_model_world2_matches_world1():
+    r2 = run_distributed(_synth_worker, world=2)
+    r1 = run_distributed(_synth_worker, world=1)
+    full = r1[0]["out"]
+    for rank in range(2):
+        ref = full[rank * 4:(rank + 1) * 4]
+        got = r2[rank]["out"]
+        assert torch.allclose(got, ref, atol=1e-4), \
+            f"rank{rank} err {(got - ref).abs().max()}"
