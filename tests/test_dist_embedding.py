@@ -588,3 +588,42 @@ def test_forward_equivalence_world4(strategy, kwargs, hotness, combiner):
             ref = ref_outs[t][rank * 4:(rank + 1) * 4]
             assert torch.allclose(o, ref, atol=1e-5), \
                 f"rank {rank} table {t}: max err {(o - ref).abs().max()}"
+
+
+def test_error_paths_world1():
+    """Fail-fast error paths (parity: reference error tests :461-490 and
+    fail-fast ValueErrors)."""
+    import distributed_embeddings_amd as de
+    model = de.DistributedEmbedding([de.TableConfig(10, 4, "sum"),
+                                     de.TableConfig(20, 4, "sum")])
+    with pytest.raises(ValueError, match="wrong number of inputs"):
+        model([torch.randint(0, 10, (4, 1))])
+    with pytest.raises(ValueError, match="expected 2 tables"):
+        model.set_weights([torch.randn(10, 4).numpy()])
+    with pytest.raises(ValueError, match="unknown strategy"):
+        de.DistEmbeddingStrategy([de.TableConfig(10, 4)], 1, strategy="nope")
+    with pytest.raises(ValueError, match="table_dtype"):
+        de.DistributedEmbedding([de.TableConfig(10, 4)],
+                                table_dtype=torch.float16)
+    with pytest.raises(ValueError, match="out of range"):
+        de.DistEmbeddingStrategy([de.TableConfig(10, 4)], 1,
+                                 input_table_map=[1])
+
+
+def _dp_input_false_with_dp_tables(rank, world):
+    import distributed_embeddings_amd as de
+    try:
+        model = de.DistributedEmbedding(
+            [de.TableConfig(4, 4, "sum"), de.TableConfig(500, 4, "sum")],
+            data_parallel_threshold=4 * 4, dp_input=False)
+        model([torch.randint(0, 4, (world * 2, 1)),
+               torch.randint(0, 500, (world * 2, 1))])
+        return "no error"
+    except ValueError as e:
+        return str(e)
+
+
+def test_dp_input_false_requires_all_tp_world2():
+    results = run_distributed(_dp_input_false_with_dp_tables, world=2)
+    for msg in results:
+        assert "dp_input=False requires" in msg
