@@ -51,3 +51,22 @@ def test_state_roundtrip_through_buffers():
     lk2.load_state_dict(sd)
     out = lk2(torch.tensor([17, 42]))
     assert out.tolist() == [2, 1]
+
+
+def test_state_dict_round_trip():
+    """The hash lives in registered buffers: checkpoint + fresh module must
+    resolve the same keys to the same values (no re-insertion)."""
+    import torch
+    from distributed_embeddings_amd import IntegerLookup
+    lk = IntegerLookup(max_tokens=100)
+    keys = torch.tensor([11, 22, 33, 44, 11])
+    vals1 = lk(keys)
+    sd = lk.state_dict()
+    lk2 = IntegerLookup(max_tokens=100)
+    lk2.load_state_dict(sd)
+    vals2 = lk2(keys)
+    assert torch.equal(vals1[:4], vals2[:4])
+    assert lk2.vocabulary_size() == lk.vocabulary_size()
+    # unseen key gets a NEW value, not a collision with restored ones
+    v_new = int(lk2(torch.tensor([99]))[0])
+    assert v_new not in vals2[:4].tolist()
