@@ -46,10 +46,15 @@ def run_distributed(fn, world=2, args=(), backend="gloo", timeout=180):
         procs.append(p)
     for p in procs:
         p.join(timeout)
+    hung = [r for r, p in enumerate(procs) if p.is_alive()]
+    if hung:
+        for p in procs:           # terminate ALL stragglers, not just the first
+            if p.is_alive():
+                p.terminate()
+        for p in procs:
+            p.join(10)
+        raise RuntimeError(f"ranks {hung} timed out")
     for r, p in enumerate(procs):
-        if p.is_alive():
-            p.terminate()
-            raise RuntimeError(f"rank {r} timed out")
         assert p.exitcode == 0, f"rank {r} exited with {p.exitcode}"
     return [results.get(r) for r in range(world)]
 
