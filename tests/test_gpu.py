@@ -504,3 +504,29 @@ def test_bf16_sparse_optimizer_gpu(method):
         og.step(); oc.step()
     d = (e_g.weight.detach().cpu().float() - e_c.weight.detach().float()).abs()
     assert float(d.max()) < 0.05, float(d.max())
+
+
+@requires_gpu
+def test_fused_sgd_dense_input_gpu_matches_cpu():
+    """Dense [b, hotness] inputs route through the fused update on GPU too
+    (converted to CSR in Embedding.forward)."""
+    from distributed_embeddings_amd import Embedding
+    torch.manual_seed(31)
+    w0 = torch.randn(300, 64)
+    ids = torch.randint(0, 300, (512, 4))
+    up = torch.randn(512, 64)
+    e_g = Embedding(300, 64, combiner="sum").cuda()
+    e_c = Embedding(300, 64, combiner="sum")
+    with torch.no_grad():
+        e_g.weight.copy_(w0)
+        e_c.weight.copy_(w0)
+    e_g.enable_fused_sgd(0.1)
+    e_c.enable_fused_sgd(0.1)
+    out_g = e_g(ids.cuda())
+    out_g.backward(up.cuda())
+    out_c = e_c(ids)
+    out_c.backward(up)
+    assert e_g.weight.grad is None
+    assert torch.allclose(out_g.cpu(), out_c, atol=1e-4)
+    d = (e_g.weight.detach().cpu() - e_c.weight.detach()).abs()
+    assert float(d.max()) < 1e-3, float(d.max())
