@@ -58,3 +58,22 @@ def test_bench_contract_help():
     assert proc.returncode == 0
     for flag in ("--gpus", "--steps", "--warmup"):
         assert flag in proc.stdout
+
+
+def test_synthetic_benchmark_torchrun_world2():
+    """The torchrun env-rendezvous path (exactly how the driver launches
+    bench.py at N>1): WORLD_SIZE/RANK from env, init_process_group('gloo')."""
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         os.path.join(ROOT, "examples", "synthetic_benchmark.py"),
+         "--model", "tiny", "--batch-size", "128",
+         "--num-steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=280, cwd=ROOT)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    assert "world=2" in proc.stdout and "ms/iteration" in proc.stdout
