@@ -41,18 +41,24 @@ def parse_args():
                    help="capture the train step in a hipGraph (world==1; "
                         "falls back to eager if capture fails)")
     p.add_argument("--no-graph", dest="graph", action="store_false")
+    p.add_argument("--table-size-cap", type=int, default=None,
+                   help="cap vocab sizes (CPU smoke only — a capped run is "
+                        "NOT a valid benchmark measurement)")
     return p.parse_args()
 
 
 def setup_dist(args):
     import torch.distributed as dist
+    have_gpu = torch.cuda.is_available()
     if "RANK" in os.environ and int(os.environ.get("WORLD_SIZE", "1")) > 1:
         rank = int(os.environ["RANK"])
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+        if have_gpu:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl" if have_gpu else "gloo")
         return rank, dist.get_world_size(), local_rank
-    torch.cuda.set_device(0)
+    if have_gpu:
+        torch.cuda.set_device(0)
     return 0, 1, 0
 
 
@@ -64,10 +70,13 @@ def build_model(args, device):
 
     if args.model == "dlrm-criteo":
         tdt = torch.bfloat16 if args.table_dtype == "bf16" else torch.float32
+        sizes = CRITEO_1TB_TABLE_SIZES
+        if args.table_size_cap:  # CPU smoke only; never a valid measurement
+            sizes = [min(s, args.table_size_cap) for s in sizes]
         with device:  # construct tables directly in HBM (96 GB fp32 / 48 GB bf16)
-            model = DLRM(CRITEO_1TB_TABLE_SIZES, embedding_dim=128,
+            model = DLRM(sizes, embedding_dim=128,
                          strategy=args.strategy, table_dtype=tdt)
-        table_sizes = CRITEO_1TB_TABLE_SIZES
+        table_sizes = sizes
         hotness = [1] * len(table_sizes)
         num_numerical = 13
         name = "DLRM-Criteo-1TB"
@@ -95,7 +104,7 @@ def maybe_enable_tunableop():
         import torch.cuda.tunable as tunable
         csv = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                            "profiles", "tunableop_gfx950.csv")
-        if os.path.exists(csv):
+        if os.path.exists(csv) and torch.cuda.is_available():
             tunable.enable(True)
             tunable.tuning_enable(False)
             tunable.read_file(csv)
@@ -107,7 +116,8 @@ def main():
     args = parse_args()
     rank, world, local_rank = setup_dist(args)
     maybe_enable_tunableop()
-    device = torch.device("cuda", local_rank)
+    device = torch.device("cuda", local_rank) if torch.cuda.is_available() \
+        else torch.device("cpu")
     torch.manual_seed(1234 + rank)
 
     import distributed_embeddings_amd as de
@@ -152,7 +162,7 @@ def main():
     _loss_sum = torch.nn.BCEWithLogitsLoss(reduction="sum")
     global_batch_ = b * world
     loss_fn = lambda logits, labels: _loss_sum(logits, labels) / global_batch_
-    use_bf16 = args.dtype == "bf16"
+    use_bf16 = args.dtype == "bf16" and torch.cuda.is_available()
 
     def run_fwd_bwd_opt(num, cats, labels, set_to_none=True):
         opt.zero_grad(set_to_none=set_to_none)
@@ -164,7 +174,8 @@ def main():
         return loss
 
     graph = None
-    graph_ok = world == 1 or os.environ.get("DE_DIST_GRAPH") == "1"
+    graph_ok = torch.cuda.is_available() and (
+        world == 1 or os.environ.get("DE_DIST_GRAPH") == "1")
     if args.graph and graph_ok:
         # hipGraph capture: static input buffers (one flat cat copy per step),
         # grads pre-materialized, fused SGD has no host syncs.
@@ -173,7 +184,7 @@ def main():
         try:
             for i in range(max(args.warmup, 2)):
                 run_fwd_bwd_opt(s_num, s_cats, s_labels, set_to_none=False)
-            torch.cuda.synchronize()
+            torch.cuda.synchronize()  # graph block runs on GPU only
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
                 run_fwd_bwd_opt(s_num, s_cats, s_labels, set_to_none=False)
@@ -196,12 +207,14 @@ def main():
         step(i)
 
     de.comm.barrier()
-    torch.cuda.synchronize()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for i in range(args.steps):
         step(i)
     de.comm.barrier()
-    torch.cuda.synchronize()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
 
     # max over ranks
@@ -226,7 +239,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": samples_per_sec / baseline,
-            "dtype": args.dtype,
+            "dtype": "bf16" if use_bf16 else "fp32",
             "data": "synthetic (power-law ids alpha=1.05, random-init weights)",
             "config": {
                 "model": name,

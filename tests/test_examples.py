@@ -77,3 +77,33 @@ def test_synthetic_benchmark_torchrun_world2():
         capture_output=True, text=True, timeout=280, cwd=ROOT)
     assert proc.returncode == 0, proc.stderr[-2000:]
     assert "world=2" in proc.stdout and "ms/iteration" in proc.stdout
+
+
+def test_bench_torchrun_world2_json_contract():
+    """The EXACT driver launch (torch.distributed.run, env rendezvous) at
+    world 2 on CPU with capped tables: one JSON line from rank 0 with every
+    contract field."""
+    import json
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         os.path.join(ROOT, "bench.py"), "--gpus", "2",
+         "--table-size-cap", "1000", "--batch-per-gpu", "64",
+         "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=280, cwd=ROOT)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    lines = [l for l in proc.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, proc.stdout  # ONE JSON line, rank 0 only
+    rec = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in rec, key
+    assert rec["n_gpus"] == 2 and rec["steps"] == 2
+    assert rec["scaling"] == "weak"
+    assert rec["config"]["global_batch"] == 128
