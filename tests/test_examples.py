@@ -107,3 +107,19 @@ def test_bench_torchrun_world2_json_contract():
     assert rec["n_gpus"] == 2 and rec["steps"] == 2
     assert rec["scaling"] == "weak"
     assert rec["config"]["global_batch"] == 128
+
+
+def test_auc_matches_sklearn():
+    import importlib.util
+    import torch as t
+    spec = importlib.util.spec_from_file_location(
+        "dlrm_main", os.path.join(ROOT, "examples", "dlrm_main.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    from sklearn.metrics import roc_auc_score
+    g = t.Generator().manual_seed(0)
+    scores = t.rand(1000, generator=g)
+    labels = (t.rand(1000, generator=g) < scores).float()  # correlated
+    ours = mod.auc(scores, labels)
+    ref = roc_auc_score(labels.numpy(), scores.numpy())
+    assert abs(ours - ref) < 1e-6
