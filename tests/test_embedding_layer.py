@@ -333,3 +333,37 @@ def test_fused_sgd_dense_input_matches_explicit(seed):
         o2.step()
         assert torch.allclose(out1, out2, atol=1e-6)
         assert torch.allclose(e1.weight, e2.weight, atol=1e-6), combiner
+
+
+def test_fused_optimizer_property(seed):
+    """Property: in-backward fused update == explicit sparse-optimizer step
+    for random shapes, hotness, combiner and method (CPU oracle path)."""
+    from hypothesis import given, settings, strategies as st
+    from distributed_embeddings_amd import (Embedding, Ragged,
+                                            SparseEmbeddingOptimizer)
+
+    @settings(max_examples=30, deadline=None)
+    @given(st.integers(0, 10_000), st.integers(2, 60), st.integers(1, 32),
+           st.integers(1, 12), st.integers(1, 6),
+           st.sampled_from(["sum", "mean"]), st.sampled_from(["sgd", "adagrad"]))
+    def check(s, vocab, width, batch, hot, combiner, method):
+        g = torch.Generator().manual_seed(s)
+        w0 = torch.randn(vocab, width, generator=g)
+        lengths = torch.randint(0, hot + 1, (batch,), generator=g)
+        values = torch.randint(0, vocab, (int(lengths.sum()),), generator=g)
+        ragged = Ragged.from_row_lengths(values, lengths)
+        up = torch.randn(batch, width, generator=g)
+
+        e1 = Embedding(vocab, width, combiner=combiner)
+        e2 = Embedding(vocab, width, combiner=combiner)
+        with torch.no_grad():
+            e1.weight.copy_(w0); e2.weight.copy_(w0)
+        e1.enable_fused_optimizer(method, 0.1)
+        e1(ragged).backward(up)
+        opt = SparseEmbeddingOptimizer(e2.parameters(), lr=0.1, method=method)
+        e2(ragged).backward(up)
+        opt.step()
+        assert torch.allclose(e1.weight, e2.weight, atol=1e-5), \
+            float((e1.weight - e2.weight).abs().max())
+
+    check()
