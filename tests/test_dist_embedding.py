@@ -627,3 +627,42 @@ def test_dp_input_false_requires_all_tp_world2():
     results = run_distributed(_dp_input_false_with_dp_tables, world=2)
     for msg in results:
         assert "dp_input=False requires" in msg
+
+
+def test_world1_fused_groups_property():
+    """Property (world=1): DistributedEmbedding's fused concat-group lookup
+    == per-table Embedding layers, over random table sets, widths, hotness,
+    combiners and strategies (exercises offset vectors, group splits and
+    output reordering)."""
+    from hypothesis import given, settings, strategies as st
+    import distributed_embeddings_amd as de
+
+    @settings(max_examples=30, deadline=None)
+    @given(st.integers(0, 10_000),
+           st.lists(st.tuples(st.integers(2, 200),       # vocab
+                              st.sampled_from([4, 8, 16]),  # width
+                              st.integers(1, 4),          # hotness
+                              st.sampled_from(["sum", "mean", None])),
+                    min_size=1, max_size=6),
+           st.sampled_from(["basic", "memory_balanced", "memory_optimized"]))
+    def check(s, tables, strategy):
+        g = torch.Generator().manual_seed(s)
+        cfgs = [de.TableConfig(v, w, c) for v, w, h, c in tables]
+        model = de.DistributedEmbedding(cfgs, strategy=strategy)
+        weights = [torch.randn(c.input_dim, c.output_dim, generator=g)
+                   for c in cfgs]
+        model.set_weights([w.numpy() for w in weights])
+        inputs, refs = [], []
+        for (v, w, h, c), wt in zip(tables, weights):
+            if c is None:
+                ids = torch.randint(0, v, (3,), generator=g)
+                refs.append(wt[ids])
+            else:
+                ids = torch.randint(0, v, (3, h), generator=g)
+                refs.append(wt[ids].sum(1) if c == "sum" else wt[ids].mean(1))
+            inputs.append(ids)
+        outs = model(inputs)
+        for o, r in zip(outs, refs):
+            assert torch.allclose(o, r, atol=1e-5), float((o - r).abs().max())
+
+    check()
