@@ -666,3 +666,46 @@ def test_world1_fused_groups_property():
             assert torch.allclose(o, r, atol=1e-5), float((o - r).abs().max())
 
     check()
+
+
+def test_world1_ragged_mixed_property():
+    """Property (world=1): mixed Ragged + dense inputs through the fused
+    group path == per-table oracle."""
+    from hypothesis import given, settings, strategies as st
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd import Ragged
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.integers(0, 10_000),
+           st.lists(st.tuples(st.integers(2, 100),        # vocab
+                              st.integers(0, 1),          # 1 = ragged input
+                              st.integers(1, 5)),         # max hotness
+                    min_size=1, max_size=5))
+    def check(s, tables):
+        g = torch.Generator().manual_seed(s)
+        cfgs = [de.TableConfig(v, 8, "sum") for v, _, _ in tables]
+        model = de.DistributedEmbedding(cfgs)
+        weights = [torch.randn(c.input_dim, 8, generator=g) for c in cfgs]
+        model.set_weights([w.numpy() for w in weights])
+        inputs, refs = [], []
+        batch = 3
+        for (v, ragged, hot), wt in zip(tables, weights):
+            if ragged:
+                lens = torch.randint(0, hot + 1, (batch,), generator=g)
+                vals = torch.randint(0, v, (int(lens.sum()),), generator=g)
+                inputs.append(Ragged.from_row_lengths(vals, lens))
+                r, pos = [], 0
+                for n in lens.tolist():
+                    r.append(wt[vals[pos:pos + n]].sum(0) if n else
+                             torch.zeros(8))
+                    pos += n
+                refs.append(torch.stack(r))
+            else:
+                ids = torch.randint(0, v, (batch, hot), generator=g)
+                inputs.append(ids)
+                refs.append(wt[ids].sum(1))
+        outs = model(inputs)
+        for o, r in zip(outs, refs):
+            assert torch.allclose(o, r, atol=1e-5), float((o - r).abs().max())
+
+    check()
