@@ -69,7 +69,11 @@ def _layer_to_config(layer) -> TableConfig:
     if isinstance(layer, TableConfig):
         return layer
     if isinstance(layer, dict):
-        return TableConfig(**layer)
+        # tolerate stock-Keras config keys (parity: reference from_config
+        # drops mask_zero/input_length, embedding.py:163-170)
+        import dataclasses
+        fields = {f.name for f in dataclasses.fields(TableConfig)}
+        return TableConfig(**{k: v for k, v in layer.items() if k in fields})
     raise TypeError(f"unsupported layer/config type {type(layer)}")
 
 

@@ -709,3 +709,16 @@ def test_world1_ragged_mixed_property():
             assert torch.allclose(o, r, atol=1e-5), float((o - r).abs().max())
 
     check()
+
+
+def test_dict_config_with_keras_extras():
+    """Dict configs may carry stock-Keras-only keys (parity: reference
+    from_config drops mask_zero/input_length, embedding.py:163-170)."""
+    import distributed_embeddings_amd as de
+    model = de.DistributedEmbedding([
+        {"input_dim": 30, "output_dim": 8, "combiner": "sum",
+         "mask_zero": False, "input_length": None},
+        {"input_dim": 40, "output_dim": 8},
+    ])
+    outs = model([torch.randint(0, 30, (4, 2)), torch.randint(0, 40, (4,))])
+    assert outs[0].shape == (4, 8) and outs[1].shape == (4, 8)
