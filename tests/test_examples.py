@@ -123,3 +123,17 @@ def test_auc_matches_sklearn():
     ours = mod.auc(scores, labels)
     ref = roc_auc_score(labels.numpy(), scores.numpy())
     assert abs(ours - ref) < 1e-6
+
+
+def test_dlrm_example_eval_and_dump(tmp_path):
+    """--eval (AUC over allgathered predictions) and --dump-embeddings
+    (np.savez via get_weights) end-to-end."""
+    import numpy as np
+    dump = str(tmp_path / "emb.npz")
+    out = _run("dlrm_main.py", ["--batch-size", "64", "--num-batches", "2",
+                                "--embedding-dim", "8",
+                                "--table-size-cap", "200",
+                                "--eval", "--dump-embeddings", dump])
+    assert "AUC:" in out and "dumped" in out
+    tables = np.load(dump)
+    assert len(tables.files) == 26
