@@ -137,3 +137,29 @@ def test_dlrm_example_eval_and_dump(tmp_path):
     assert "AUC:" in out and "dumped" in out
     tables = np.load(dump)
     assert len(tables.files) == 26
+
+
+def test_dlrm_example_with_binary_dataset(tmp_path):
+    """The example's real-dataset path: tiny split-binary Criteo layout on
+    disk, 26 capped tables, 2 training steps."""
+    import numpy as np
+    n, nnum, cap = 64, 13, 200
+    sizes = [min(s, cap) for s in
+             [39884407, 39043, 17289, 7420, 20263, 3, 7120, 1543, 63,
+              38532952, 2953546, 403346, 10, 2208, 11938, 155, 4, 976, 14,
+              39979772, 25641295, 39664985, 585935, 12972, 108, 36]]
+    d = tmp_path / "train"
+    d.mkdir()
+    rng = np.random.RandomState(0)
+    (d / "label.bin").write_bytes(rng.randint(0, 2, n).astype(np.int8).tobytes())
+    (d / "numerical.bin").write_bytes(
+        rng.rand(n, nnum).astype(np.float16).tobytes())
+    for i, s in enumerate(sizes):
+        dt = np.int8 if s <= 256 else np.int16 if s <= 65536 else np.int32
+        (d / f"cat_{i}.bin").write_bytes(
+            rng.randint(0, s, n).astype(dt).tobytes())
+    out = _run("dlrm_main.py", ["--batch-size", "32", "--num-batches", "2",
+                                "--embedding-dim", "8",
+                                "--table-size-cap", str(cap),
+                                "--dataset-path", str(tmp_path)])
+    assert "loss" in out
