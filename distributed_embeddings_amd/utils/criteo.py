@@ -17,10 +17,11 @@ import torch
 
 
 def _cat_dtype(vocab: int):
-    # parity: reference utils.py:116-123 (int8/16/32 by category count)
-    if vocab <= (1 << 8):
+    # parity: reference utils.py:116-123 — smallest SIGNED dtype whose max
+    # exceeds the vocab (int8 only below 127: ids are stored signed)
+    if vocab < np.iinfo(np.int8).max:
         return np.int8, 1
-    if vocab <= (1 << 16):
+    if vocab < np.iinfo(np.int16).max:
         return np.int16, 2
     return np.int32, 4
 

@@ -155,7 +155,8 @@ def test_dlrm_example_with_binary_dataset(tmp_path):
     (d / "numerical.bin").write_bytes(
         rng.rand(n, nnum).astype(np.float16).tobytes())
     for i, s in enumerate(sizes):
-        dt = np.int8 if s <= 256 else np.int16 if s <= 65536 else np.int32
+        dt = (np.int8 if s < np.iinfo(np.int8).max else
+              np.int16 if s < np.iinfo(np.int16).max else np.int32)
         (d / f"cat_{i}.bin").write_bytes(
             rng.randint(0, s, n).astype(dt).tobytes())
     out = _run("dlrm_main.py", ["--batch-size", "32", "--num-batches", "2",

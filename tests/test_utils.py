@@ -119,3 +119,17 @@ def test_warmup_poly_decay_updates_fused_lr():
     for _ in range(3):
         sched.step()
     assert abs(float(emb._fused_lr) - 1.0) < 1e-6
+
+
+def test_cat_dtype_boundaries():
+    """Signed-dtype boundaries (parity: reference get_categorical_feature_type
+    uses iinfo(dtype).max): vocab 200 must NOT be int8 (ids >=128 would wrap
+    negative)."""
+    from distributed_embeddings_amd.utils.criteo import _cat_dtype
+    assert _cat_dtype(100) == (np.int8, 1)
+    assert _cat_dtype(126) == (np.int8, 1)
+    assert _cat_dtype(127) == (np.int16, 2)
+    assert _cat_dtype(200) == (np.int16, 2)
+    assert _cat_dtype(32766) == (np.int16, 2)
+    assert _cat_dtype(32767) == (np.int32, 4)
+    assert _cat_dtype(40_000_000) == (np.int32, 4)
