@@ -191,3 +191,23 @@ def test_bf16_get_weights_all_modes():
     for w, g in zip(weights, got):
         assert g.dtype == np.float32
         assert np.allclose(w, g, rtol=0.01, atol=0.01)  # bf16 rounding
+
+
+def _get_weights_rank0_worker(rank, world):
+    import distributed_embeddings_amd as de
+    sizes = [30, 200]
+    model = de.DistributedEmbedding([de.TableConfig(s, 8) for s in sizes])
+    g = torch.Generator().manual_seed(2)
+    weights = [torch.randn(s, 8, generator=g).numpy() for s in sizes]
+    model.set_weights(weights)
+    got = model.get_weights(all_ranks=False)  # collective: all ranks call
+    if rank != 0:
+        return 0.0
+    return max(float(np.abs(got[t] - weights[t]).max()) for t in range(2))
+
+
+def test_get_weights_rank0_only_world2():
+    """all_ranks=False is still a collective — every rank participates, no
+    deadlock, rank 0 gets the full tables."""
+    results = run_distributed(_get_weights_rank0_worker, world=2)
+    assert results[0] < 1e-6
