@@ -84,6 +84,14 @@ class Embedding(nn.Module):
     def extra_repr(self) -> str:
         return f"input_dim={self.input_dim}, output_dim={self.output_dim}, combiner={self.combiner}"
 
+    def _apply(self, fn, recurse=True):
+        # CPU-offloaded tables are pinned: .to('cuda') / .cuda() on the parent
+        # module must not move the weight (parity: reference builds offloaded
+        # variables under tf.device('CPU:0'), dist_model_parallel.py:1186-1189).
+        if getattr(self, "_cpu_offload", False):
+            return self
+        return super()._apply(fn, recurse)
+
     def enable_fused_sgd(self, lr: float):
         """In-backward SGD (see enable_fused_optimizer)."""
         return self.enable_fused_optimizer("sgd", lr)

@@ -50,6 +50,51 @@ def _as_list(x):
     return None if x is None else [int(v) for v in x]
 
 
+def backend_device(group=None) -> torch.device:
+    """Device collectives must run on: CUDA for the nccl(RCCL) backend, CPU
+    for gloo / uninitialized."""
+    if not is_initialized():
+        return torch.device("cpu")
+    if dist.get_backend(group) == "nccl":
+        return torch.device("cuda", torch.cuda.current_device())
+    return torch.device("cpu")
+
+
+def all_gather_ints(vals: Sequence[int], group=None) -> List[List[int]]:
+    """All-gather one equal-length int vector per rank via tensor collectives.
+
+    RCCL-safe replacement for ``all_gather_object`` (which pickles through
+    CPU staging buffers and is neither graph-capturable nor portable across
+    backends).  Every rank must pass the SAME number of ints.
+    """
+    w = world_size(group)
+    if w == 1:
+        return [[int(v) for v in vals]]
+    dev = backend_device(group)
+    t = torch.tensor([int(v) for v in vals], dtype=torch.long, device=dev)
+    out = torch.empty(w * t.numel(), dtype=torch.long, device=dev)
+    dist.all_gather_into_tensor(out, t, group=group)
+    return out.view(w, -1).cpu().tolist()
+
+
+def all_gather_int_vectors(vals: Sequence[int], group=None) -> List[List[int]]:
+    """All-gather per-rank int vectors of DIFFERING lengths (two tensor
+    collectives: lengths, then max-padded payload)."""
+    w = world_size(group)
+    if w == 1:
+        return [[int(v) for v in vals]]
+    lens = [row[0] for row in all_gather_ints([len(vals)], group)]
+    mx = max(lens + [1])
+    dev = backend_device(group)
+    t = torch.zeros(mx, dtype=torch.long, device=dev)
+    if len(vals):
+        t[:len(vals)] = torch.tensor([int(v) for v in vals], dtype=torch.long)
+    out = torch.empty(w * mx, dtype=torch.long, device=dev)
+    dist.all_gather_into_tensor(out, t, group=group)
+    rows = out.view(w, mx).cpu().tolist()
+    return [row[:k] for row, k in zip(rows, lens)]
+
+
 class _AllToAllSingle(torch.autograd.Function):
     @staticmethod
     def forward(ctx, inp, out_splits, in_splits, group):
@@ -147,15 +192,18 @@ def all_gather_uneven(inp: torch.Tensor, group=None) -> List[torch.Tensor]:
     w = world_size(group)
     if w == 1:
         return [inp]
-    sizes = [None] * w
-    dist.all_gather_object(sizes, int(inp.shape[0]), group=group)
+    sizes = [row[0] for row in all_gather_ints([int(inp.shape[0])], group)]
     mx = max(sizes)
-    padded = inp
+    # stage through the backend's device (nccl refuses CPU tensors), return
+    # on the caller's device
+    dev = backend_device(group)
+    padded = inp.to(dev)
     if inp.shape[0] < mx:
-        padded = torch.cat([inp, inp.new_zeros((mx - inp.shape[0],) + tuple(inp.shape[1:]))])
-    out = [inp.new_empty((mx,) + tuple(inp.shape[1:])) for _ in range(w)]
+        padded = torch.cat(
+            [padded, padded.new_zeros((mx - inp.shape[0],) + tuple(inp.shape[1:]))])
+    out = [padded.new_empty((mx,) + tuple(inp.shape[1:])) for _ in range(w)]
     dist.all_gather(out, padded.contiguous(), group=group)
-    return [o[:s] for o, s in zip(out, sizes)]
+    return [o[:s].to(inp.device) for o, s in zip(out, sizes)]
 
 
 def broadcast(t: torch.Tensor, src: int = 0, group=None) -> torch.Tensor:

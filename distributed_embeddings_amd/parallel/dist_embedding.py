@@ -152,8 +152,12 @@ class DistributedEmbedding(nn.Module):
         # XLA-fusion hint, dist_model_parallel.py:650).
         col_layers = []
         for grp in plan.local_concat_groups(self.rank):
+            # offloaded tables are built ON CPU and pinned there (the explicit
+            # device= overrides any enclosing torch.device('cuda') context;
+            # Embedding._apply then refuses later moves).
             lyr = Embedding(grp.input_dim, grp.output_dim, grp.combiner,
-                            dtype=table_dtype)
+                            dtype=table_dtype,
+                            device="cpu" if grp.cpu_offload else None)
             for p in lyr.parameters():
                 if self.world_size > 1:
                     p.de_local = True  # model-parallel: excluded from grad allreduce
@@ -220,8 +224,12 @@ class DistributedEmbedding(nn.Module):
     def enable_fused_optimizer(self, method: str, lr: float, eps: float = 1e-10):
         """Enables in-backward fused SGD/Adagrad on all model-parallel tables
         (col + row groups).  Data-parallel tables keep sparse grads (they
-        need the allreduce).  See Embedding.enable_fused_optimizer."""
+        need the allreduce); CPU-offloaded tables keep sparse grads too (their
+        lookup runs through the CPU fallback, not the HIP fused kernel).
+        See Embedding.enable_fused_optimizer."""
         for lyr in list(self.col_layers) + list(self.row_layers):
+            if getattr(lyr, "_cpu_offload", False):
+                continue
             lyr.enable_fused_optimizer(method, lr, eps)
         return self
 
@@ -243,9 +251,7 @@ class DistributedEmbedding(nn.Module):
     def _validate_batch(self, b: int):
         if self._built or self.world_size == 1:
             return
-        import torch.distributed as dist
-        sizes = [None] * self.world_size
-        dist.all_gather_object(sizes, int(b))
+        sizes = [row[0] for row in comm.all_gather_ints([int(b)])]
         if any(s != sizes[0] for s in sizes):
             raise ValueError(f"per-rank batch sizes differ: {sizes} "
                              "(parity: reference build() check :1171-1173)")
@@ -325,12 +331,20 @@ class DistributedEmbedding(nn.Module):
             copy_into(self.row_layers[local_t].weight,
                       w[shard.row_offset:shard.row_offset + shard.rows])
 
-    def get_weights(self, all_ranks: bool = False) -> List[np.ndarray]:
+    def get_weights(self, all_ranks: bool = False,
+                    chunk_elements: int = 128 * 1024 * 1024) -> List[np.ndarray]:
         """Reassembles full global per-table weights (on CPU, numpy).
 
         Parity: reference ``get_weights`` (:1139-1162) — returns the tables in
         original order; with ``all_ranks=False`` only rank 0's return value is
         meaningful on GPU backends (collectives still run on every rank).
+
+        Every collective moves at most ``chunk_elements`` elements and the
+        transient device buffers are O(chunk), not O(table) — 2e9-row tables
+        reassemble with bounded peak memory (parity: reference ≤2e9-element
+        chunked allgather + ``_split_1d``, dist_model_parallel.py:1024-1046,
+        1069-1098; per-chunk counts stay far below int32 limits by
+        construction).
         """
         import torch.distributed as dist
         cfgs = self.strategy.configs
@@ -341,29 +355,49 @@ class DistributedEmbedding(nn.Module):
             # (bf16 storage has no numpy dtype)
             out[t] = self.dp_layers[local_t].weight.detach().float().cpu().numpy()
 
-        # column slices: broadcast each slice from its owner, reassemble.
+        # column slices: broadcast each slice from its owner in row chunks.
         device = self._comm_device()
         for t in plan.col_table_ids:
             cfg = cfgs[t]
             cols = []
             for s in plan.table_slices[t]:
-                buf = torch.empty(cfg.input_dim, s.width, device=device)
-                if s.rank == self.rank:
-                    src = self.col_layers[s.concat_group].weight.detach()
-                    buf.copy_(src[s.concat_row_offset:s.concat_row_offset + cfg.input_dim])
-                if self.world_size > 1:
-                    dist.broadcast(buf, src=s.rank)
-                cols.append(buf.cpu())
+                step = max(1, chunk_elements // s.width)
+                pieces = []
+                for r0 in range(0, cfg.input_dim, step):
+                    r1 = min(cfg.input_dim, r0 + step)
+                    buf = torch.empty(r1 - r0, s.width, device=device)
+                    if s.rank == self.rank:
+                        src = self.col_layers[s.concat_group].weight.detach()
+                        buf.copy_(src[s.concat_row_offset + r0:
+                                      s.concat_row_offset + r1])
+                    if self.world_size > 1:
+                        dist.broadcast(buf, src=s.rank)
+                    pieces.append(buf.cpu())
+                cols.append(torch.cat(pieces, dim=0))
             out[t] = torch.cat(cols, dim=1).numpy()
 
-        # row shards: allgather uneven rows.
+        # row shards: chunked allgather of uneven rows (every rank iterates
+        # the same global chunk count so the collectives stay matched).
         for t in plan.row_table_ids:
             local_t = plan.row_table_ids.index(t)
-            rows = plan.row_shards[t][self.rank].rows
-            # fp32 for the gather + numpy conversion (bf16-safe on gloo too)
-            shard_w = self.row_layers[local_t].weight.detach()[:rows].float().to(device)
-            parts = comm.all_gather_uneven(shard_w)
-            out[t] = torch.cat([p.cpu() for p in parts], dim=0).numpy()
+            my_rows = plan.row_shards[t][self.rank].rows
+            width = cfgs[t].output_dim
+            max_rows = max(sh.rows for sh in plan.row_shards[t])
+            step = max(1, chunk_elements // width)
+            nchunks = max(-(-max_rows // step), 1)
+            per_rank = [[] for _ in range(self.world_size)]
+            w_local = self.row_layers[local_t].weight.detach()
+            for ci in range(nchunks):
+                r0 = ci * step
+                r1 = min(my_rows, r0 + step)
+                piece = w_local[r0:max(r0, r1)].float().to(device)
+                parts = comm.all_gather_uneven(piece)
+                for k, p in enumerate(parts):
+                    if p.shape[0]:
+                        per_rank[k].append(p.cpu())
+            out[t] = torch.cat(
+                [torch.cat(p, dim=0) if p else
+                 torch.empty(0, width) for p in per_rank], dim=0).numpy()
         return out
 
     def _comm_device(self):
@@ -404,7 +438,7 @@ class DistributedEmbedding(nn.Module):
         send_parts = [col_inputs[i].reshape(-1)
                       for k in range(W) for i in plan.rank_input_ids[k]]
         send = torch.cat(send_parts) if send_parts else \
-            torch.empty(0, dtype=torch.long)
+            torch.empty(0, dtype=torch.long, device=self._comm_device())
         recv = comm.all_to_all_single(send, out_splits, in_splits)
         # recv: [W, sum(my_sizes)] -> per pair [W*b, ...]
         recv = recv.view(W, -1) if recv.numel() else recv.view(W, 0)
@@ -637,12 +671,14 @@ class DistributedEmbedding(nn.Module):
         if not plan.col_table_ids:
             return []
 
+        local_b = None
         if self.dp_input:
             any_ragged = any(isinstance(x, Ragged) for x in col_inputs)
             if col_inputs:
                 b = (col_inputs[0].nrows if isinstance(col_inputs[0], Ragged)
                      else col_inputs[0].shape[0])
                 self._validate_batch(b)
+                local_b = b  # dp-side batch known on every rank
             if any_ragged:
                 pair_ids = self._dp_to_mp_ragged(col_inputs, None)
             else:
@@ -656,15 +692,32 @@ class DistributedEmbedding(nn.Module):
                     raise ValueError(
                         f"model-parallel input batch {n0} not divisible by world {W} "
                         "(parity: reference :1175-1177)")
-            b = None
+                local_b = n0 // W
+        # A rank left without column slices by the planner still participates
+        # in the output all-to-all: exchange the batch size when the plan
+        # leaves ANY rank empty (global plan => every rank enters the
+        # collective consistently).
+        if any(len(plan.rank_input_ids[k]) == 0 for k in range(W)):
+            vals = comm.all_gather_ints([local_b if local_b is not None else -1])
+            known = [v[0] for v in vals if v[0] >= 0]
+            local_b = known[0] if known else 0
+        if local_b is None:
+            local_b = 0
 
         outs = self._fused_group_lookup(pair_ids)  # per pair [W*b, c]
 
         # mp->dp output all-to-all (parity: reference :868-878).
-        rows = outs[0].shape[0] if outs else 0
-        local_b = rows // W
-        send = torch.cat([o.reshape(W, -1) for o in outs], dim=1).reshape(-1) if outs else \
-            torch.empty(0)
+        if outs:
+            send = torch.cat([o.reshape(W, -1) for o in outs], dim=1).reshape(-1)
+        else:
+            # empty-contribution rank: buffer must match peers' device/dtype
+            # for the RCCL collective, and must REQUIRE GRAD so this rank
+            # enters the reverse all-to-all in backward alongside its peers
+            send = torch.empty(
+                0,
+                dtype=getattr(self, "_output_dtype", None) or self.table_dtype,
+                device=self._comm_device(),
+                requires_grad=torch.is_grad_enabled())
         my_cols = sum(o.shape[1] for o in outs) if outs else 0
         in_splits = [local_b * my_cols] * W
         # Per-pair output column counts of every rank (static after first call;
@@ -683,34 +736,62 @@ class DistributedEmbedding(nn.Module):
             parts = torch.split(chunks[k], sizes)
             worker_outs.extend(p.view(local_b, c) for p, c in zip(parts, all_cols[k]))
         ordered = [worker_outs[i] for i in plan.rev_tp_order]
+        # per ordered entry, the slice width (ordered slices of one input are
+        # in col_offset order — worker order == rank walk order in the plan)
+        ordered_widths = [plan.widths_list_flat[i] for i in plan.rev_tp_order]
 
-        # concat column slices back together (parity: reference :884-886).
+        # concat column slices back together (parity: reference :884-886) and
+        # restore the hotness dimension for no-combiner multi-hot inputs
+        # (matching the Embedding layer / row-slice output contract: the
+        # column count of such a pair is hotness*width, so slice concat must
+        # interleave on the LAST dim, not blockwise on dim 1).
         merged = []
-        pos = 0
         ranges = {start: stop for start, stop in plan.sliced_out_ranges}
         idx = 0
+        inp_local = 0
         while idx < len(ordered):
-            if idx in ranges:
-                stop = ranges[idx]
-                merged.append(torch.cat(ordered[idx:stop], dim=1))
-                idx = stop
+            stop = ranges.get(idx, idx + 1)
+            parts = ordered[idx:stop]
+            widths = ordered_widths[idx:stop]
+            t = plan.col_table_ids[plan.input_maps[1][inp_local]]
+            feat_shape = None
+            if plan.configs[t].combiner is None:
+                if self.dp_input and inp_local < len(col_inputs):
+                    x = col_inputs[inp_local]
+                    if not isinstance(x, Ragged) and x.dim() > 1:
+                        feat_shape = tuple(x.shape[1:])
+                else:
+                    # mp-input mode: derive hotness from the column count
+                    hot = parts[0].shape[1] // widths[0]
+                    if hot > 1:
+                        feat_shape = (hot,)
+            if feat_shape is not None:
+                parts = [p.view(local_b, *feat_shape, w)
+                         for p, w in zip(parts, widths)]
+                merged.append(parts[0] if len(parts) == 1
+                              else torch.cat(parts, dim=-1))
             else:
-                merged.append(ordered[idx])
-                idx += 1
+                merged.append(parts[0] if len(parts) == 1
+                              else torch.cat(parts, dim=1))
+            idx = stop
+            inp_local += 1
         return merged
 
     def _exchange_pair_cols(self, my_pair_cols):
-        """Share per-pair output column counts across ranks (cached)."""
-        if getattr(self, "_all_pair_cols", None) is not None:
-            return self._all_pair_cols
-        if self.world_size == 1:
-            self._all_pair_cols = [list(my_pair_cols)]
-            return self._all_pair_cols
-        import torch.distributed as dist
-        gathered = [None] * self.world_size
-        dist.all_gather_object(gathered, list(my_pair_cols))
-        self._all_pair_cols = gathered
-        return self._all_pair_cols
+        """Share per-pair output column counts across ranks.
+
+        Cached PER column-count signature: for no-combiner inputs the count is
+        ``hotness*width``, so a hotness change between calls must re-exchange
+        (shape changes are globally consistent — the same contract as the
+        dp-side ``_split_cache``).  Tensor collectives only (RCCL-safe).
+        """
+        key = tuple(my_pair_cols)
+        cache = getattr(self, "_pair_cols_cache", None)
+        if cache is None:
+            cache = self._pair_cols_cache = {}
+        if key not in cache:
+            cache[key] = comm.all_gather_int_vectors(list(my_pair_cols))
+        return cache[key]
 
     # --------------------------------------------------------- row slice path
 

@@ -111,10 +111,14 @@ def main():
                                  feature_ids=feature_ids,
                                  dp_input=args.dp_input or world == 1)
 
+    # loss below is sum(BCE)/global_batch and dp grads are SUMMED
+    # (average=False), so each grad is already the global-batch mean —
+    # lr is used as-is (parity: reference lr=24 with a mean loss,
+    # examples/dlrm/main.py).
     opt = de.DistributedOptimizer(SparseEmbeddingOptimizer(
-        model.parameters(), lr=args.learning_rate / args.batch_size, method="sgd"),
-        average=False)  # loss normalized by global batch below
-    sched = WarmupPolyDecay(opt, base_lr=args.learning_rate / args.batch_size,
+        model.parameters(), lr=args.learning_rate, method="sgd"),
+        average=False)
+    sched = WarmupPolyDecay(opt, base_lr=args.learning_rate,
                             warmup_steps=args.warmup_steps,
                             decay_start=args.decay_start,
                             decay_steps=args.decay_steps)

@@ -1,0 +1,177 @@
+"""Round-2 regression tests: zero-slice ranks, chunked checkpoint collectives,
+hotness-change re-exchange, CPU-offload pinning, tensor-based int exchange.
+
+Covers the advisor findings and VERDICT.md next-round items
+#3 (chunked get_weights) and the ``all_gather_object`` replacement (#1).
+"""
+
+import numpy as np
+import pytest
+import torch
+
+from conftest import run_distributed
+
+
+# ---------------------------------------------------------------- int exchange
+
+def _gather_ints_worker(rank, world):
+    from distributed_embeddings_amd.parallel import comm
+    same = comm.all_gather_ints([rank * 10, rank * 10 + 1])
+    vecs = comm.all_gather_int_vectors(list(range(rank + 1)))
+    return {"same": same, "vecs": vecs}
+
+
+def test_all_gather_ints_world3():
+    outs = run_distributed(_gather_ints_worker, world=3)
+    for o in outs:
+        assert o["same"] == [[0, 1], [10, 11], [20, 21]]
+        assert o["vecs"] == [[0], [0, 1], [0, 1, 2]]
+
+
+def test_all_gather_ints_world1():
+    from distributed_embeddings_amd.parallel import comm
+    assert comm.all_gather_ints([5, 6]) == [[5, 6]]
+    assert comm.all_gather_int_vectors([]) == [[]]
+
+
+# ------------------------------------------------------------- zero-slice rank
+
+def _zero_slice_worker(rank, world):
+    """One width-1 table at world 2: rank 1 owns no column slices but must
+    still participate in the output all-to-all."""
+    import distributed_embeddings_amd as de
+    tables = [de.TableConfig(20, 1, "sum")]
+    model = de.DistributedEmbedding(tables, strategy="basic")
+    # exactly one rank should own the single slice
+    n_slices = [len(model.strategy.rank_slices[r]) for r in range(world)]
+    assert sorted(n_slices) == [0, 1], n_slices
+
+    w = torch.arange(20, dtype=torch.float32).reshape(20, 1)
+    model.set_weights([w.numpy()])
+    ids = torch.tensor([[0, 1], [2, 3], [4, 5], [6, 7]])[rank * 2:(rank + 1) * 2]
+    (out,) = model(tuple([ids]))
+    expect = w[ids].sum(dim=1)
+    assert torch.allclose(out, expect), (out, expect)
+    out.sum().backward()
+    return True
+
+
+def test_zero_slice_rank_world2():
+    assert all(run_distributed(_zero_slice_worker, world=2))
+
+
+# -------------------------------------------------------- chunked get_weights
+
+def _chunked_weights_worker(rank, world, chunk):
+    import distributed_embeddings_amd as de
+    sizes = [50, 700, 33]
+    tables = [de.TableConfig(s, 8, "sum") for s in sizes]
+    model = de.DistributedEmbedding(tables, strategy="basic",
+                                    row_slice_threshold=5000,
+                                    data_parallel_threshold=300)
+    g = torch.Generator().manual_seed(11)
+    weights = [torch.randn(s, 8, generator=g) for s in sizes]
+    model.set_weights([w.numpy() for w in weights])
+    out = model.get_weights(all_ranks=True, chunk_elements=chunk)
+    return [torch.as_tensor(w) for w in out]
+
+
+@pytest.mark.parametrize("chunk", [17, 64, 128 * 1024 * 1024])
+def test_get_weights_chunked_world2(chunk):
+    """Tiny chunk_elements forces many small collectives; result must equal
+    the unchunked reassembly exactly."""
+    outs = run_distributed(_chunked_weights_worker, world=2, args=(chunk,))
+    g = torch.Generator().manual_seed(11)
+    expect = [torch.randn(s, 8, generator=g) for s in [50, 700, 33]]
+    for o in outs:
+        for got, want in zip(o, expect):
+            assert torch.equal(got, want)
+
+
+def test_get_weights_chunked_world1():
+    out1 = _chunked_weights_worker(0, 1, 17)
+    out2 = _chunked_weights_worker(0, 1, 1 << 30)
+    for a, b in zip(out1, out2):
+        assert torch.equal(a, b)
+
+
+# -------------------------------------------- hotness change between calls
+
+def _hotness_change_worker(rank, world):
+    """No-combiner multi-hot inputs: output col counts are hotness*width, so
+    the mp->dp split exchange must re-key when hotness changes between calls
+    (ADVICE.md low #1)."""
+    import distributed_embeddings_amd as de
+    sizes = [40, 60]
+    tables = [de.TableConfig(s, 4, None) for s in sizes]
+    model = de.DistributedEmbedding(tables, strategy="basic")
+    g = torch.Generator().manual_seed(5)
+    weights = [torch.randn(s, 4, generator=g) for s in sizes]
+    model.set_weights([w.numpy() for w in weights])
+
+    results = []
+    for hot in (3, 5, 3):
+        g2 = torch.Generator().manual_seed(50 + hot)
+        full = [torch.randint(0, s, (world * 2, hot), generator=g2) for s in sizes]
+        local = [x[rank * 2:(rank + 1) * 2] for x in full]
+        outs = model(local)
+        for o, w, ids in zip(outs, weights, local):
+            assert o.shape == (2, hot, 4)
+            assert torch.allclose(o, w[ids]), (hot,)
+        results.append(True)
+    return all(results)
+
+
+def test_hotness_change_reexchange_world2():
+    assert all(run_distributed(_hotness_change_worker, world=2))
+
+
+# ------------------------------------------------------------ offload pinning
+
+def test_offload_pinned_against_apply():
+    """CPU-offloaded tables must survive module-wide _apply moves/casts
+    (ADVICE.md medium): the reference pins with tf.device('CPU:0')."""
+    import distributed_embeddings_amd as de
+    tables = [de.TableConfig(100, 8, "sum"), de.TableConfig(10000, 8, "sum")]
+    model = de.DistributedEmbedding(tables, strategy="basic",
+                                    gpu_embedding_size=5000)
+    offloaded = [l for l in model.col_layers if getattr(l, "_cpu_offload", False)]
+    resident = [l for l in model.col_layers if not getattr(l, "_cpu_offload", False)]
+    assert offloaded, "expected the 10000x8 table to be offloaded"
+    model.double()  # stand-in for .cuda()/.to(device) — same _apply path
+    for l in offloaded:
+        assert l.weight.dtype == torch.float32  # untouched
+        assert l.weight.device.type == "cpu"
+    for l in resident:
+        assert l.weight.dtype == torch.float64  # moved as usual
+
+    # forward still works after the attempted move
+    ids = torch.randint(0, 100, (4, 2))
+    ids2 = torch.randint(0, 10000, (4, 2))
+    outs = model([ids, ids2])
+    assert outs[0].shape == (4, 8) and outs[1].shape == (4, 8)
+
+
+def test_offload_constructed_on_cpu_under_device_context():
+    """Constructing under a device context must still place offloaded tables
+    on CPU (meta stands in for cuda here: no GPU in CI)."""
+    import distributed_embeddings_amd as de
+    tables = [de.TableConfig(10000, 8, "sum")]
+    with torch.device("meta"):
+        model = de.DistributedEmbedding(tables, strategy="basic",
+                                        gpu_embedding_size=5000)
+    (lyr,) = model.col_layers
+    assert getattr(lyr, "_cpu_offload", False)
+    assert lyr.weight.device.type == "cpu"
+
+
+# ----------------------------------------------------- dlrm example lr contract
+
+def test_dlrm_lr_not_double_scaled():
+    """The example passes lr through unscaled: the loss is already normalized
+    by the global batch and dp grads are summed."""
+    import ast
+    import pathlib
+    src = (pathlib.Path(__file__).parent.parent / "examples" / "dlrm_main.py").read_text()
+    assert "args.learning_rate / args.batch_size" not in src
+    assert "lr=args.learning_rate" in src
