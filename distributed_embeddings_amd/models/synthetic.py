@@ -20,17 +20,28 @@ from .config import ModelConfig
 
 
 def expand_tables(model_config: ModelConfig):
-    """EmbeddingConfig list -> (table configs, input_table_map, hotness list)."""
+    """EmbeddingConfig list -> (table configs, input_table_map, hotness list).
+
+    ``shared=True``: one table per ``num_tables``, one input per nnz entry
+    (all mapped to the same table).  ``shared=False``: each (table, nnz)
+    combination is its OWN table — ``num_tables * len(nnz)`` tables in total
+    (parity: reference ``config_v3.py:21-24`` /
+    ``synthetic_models.py:116-176``).
+    """
     tables, input_table_map, hotness = [], [], []
     for cfg in model_config.embedding_configs:
-        if len(cfg.nnz) > 1 and not cfg.shared:
-            raise NotImplementedError("non-shared multi-hot tables")
         for _ in range(cfg.num_tables):
-            t = len(tables)
-            tables.append((cfg.num_rows, cfg.width))
-            for nnz in cfg.nnz:
-                input_table_map.append(t)
-                hotness.append(nnz)
+            if cfg.shared:
+                t = len(tables)
+                tables.append((cfg.num_rows, cfg.width))
+                for nnz in cfg.nnz:
+                    input_table_map.append(t)
+                    hotness.append(nnz)
+            else:
+                for nnz in cfg.nnz:
+                    input_table_map.append(len(tables))
+                    tables.append((cfg.num_rows, cfg.width))
+                    hotness.append(nnz)
     return tables, input_table_map, hotness
 
 
