@@ -255,6 +255,21 @@ torch::Tensor integer_lookup(torch::Tensor keys, torch::Tensor table_keys,
   return out;
 }
 
+void hash_rehash(torch::Tensor old_keys, torch::Tensor old_values,
+                 torch::Tensor new_keys, torch::Tensor new_values) {
+  CHECK_CUDA(old_keys); CHECK_CUDA(old_values);
+  CHECK_CUDA(new_keys); CHECK_CUDA(new_values);
+  CHECK_CONTIG(old_keys); CHECK_CONTIG(old_values);
+  CHECK_CONTIG(new_keys); CHECK_CONTIG(new_values);
+  TORCH_CHECK(new_keys.numel() >= old_keys.numel(),
+              "rehash target must not be smaller");
+  launch_hash_reinsert(old_keys.data_ptr<int64_t>(),
+                       old_values.data_ptr<int64_t>(), old_keys.numel(),
+                       new_keys.data_ptr<int64_t>(),
+                       new_values.data_ptr<int64_t>(), new_keys.numel(),
+                       current_stream());
+}
+
 void sparse_row_update(torch::Tensor weight, torch::Tensor state,
                        torch::Tensor ids, torch::Tensor grad, double lr,
                        double eps, bool adagrad) {
@@ -426,6 +441,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("csr_lookup_backward", &csr_lookup_backward,
         "sparse backward: sort + unique + segmented sum (gfx950)");
   m.def("row_to_split", &row_to_split, "COO rows -> CSR splits (gfx950)");
+  m.def("hash_rehash", &hash_rehash,
+        "re-insert occupied (key,value) pairs into a larger hash (gfx950)");
   m.def("integer_lookup", &integer_lookup,
         "open-addressing hash vocab build + lookup (gfx950)");
   m.def("sparse_row_update", &sparse_row_update,

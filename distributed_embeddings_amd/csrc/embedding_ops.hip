@@ -1272,6 +1272,41 @@ __global__ void hash_find(const int64_t* __restrict__ keys, int64_t n,
   out[i] = value;
 }
 
+// Rehash: re-insert every occupied (key, value) pair of the old table into a
+// larger (sentinel-initialized) one, PRESERVING values.  Keys are unique in
+// the old table, so CAS races are only slot contention between different
+// keys — the loser probes on; no spin-waits (same wave64 deadlock rationale
+// as hash_insert above).
+__global__ void hash_reinsert(const int64_t* __restrict__ old_keys,
+                              const int64_t* __restrict__ old_vals,
+                              int64_t old_cap, int64_t* __restrict__ tkeys,
+                              int64_t* __restrict__ tvals, int64_t capacity) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= old_cap) return;
+  const int64_t key = old_keys[i];
+  if (key == IL_EMPTY) return;
+  uint64_t slot = mix64((uint64_t)key) % (uint64_t)capacity;
+  for (int64_t probe = 0; probe < capacity; ++probe) {
+    int64_t prev = (int64_t)atomicCAS((unsigned long long*)&tkeys[slot],
+                                      (unsigned long long)IL_EMPTY,
+                                      (unsigned long long)key);
+    if (prev == IL_EMPTY) {
+      tvals[slot] = old_vals[i];
+      return;
+    }
+    slot = (slot + 1) % (uint64_t)capacity;
+  }
+}
+
+void launch_hash_reinsert(const int64_t* old_keys, const int64_t* old_vals,
+                          int64_t old_cap, int64_t* tkeys, int64_t* tvals,
+                          int64_t capacity, hipStream_t stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(hash_reinsert, dim3((int)cdiv64(old_cap, block)),
+                     dim3(block), 0, stream, old_keys, old_vals, old_cap,
+                     tkeys, tvals, capacity);
+}
+
 void launch_integer_lookup(const int64_t* keys, int64_t n, int64_t* tkeys,
                            int64_t* tvals, int64_t capacity, int32_t* counts,
                            int64_t max_tokens, void* temp, size_t temp_bytes,

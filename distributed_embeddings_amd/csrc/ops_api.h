@@ -56,6 +56,10 @@ void launch_integer_lookup(const int64_t* keys, int64_t n, int64_t* tkeys,
 
 size_t integer_lookup_temp_bytes(int64_t max_tokens);
 
+void launch_hash_reinsert(const int64_t* old_keys, const int64_t* old_vals,
+                          int64_t old_cap, int64_t* tkeys, int64_t* tvals,
+                          int64_t capacity, hipStream_t stream);
+
 void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
                             const int64_t* bounds, hipStream_t stream);
 

@@ -6,39 +6,64 @@ On GPU the vocabulary lives in a hand-written open-addressing (linear-probe)
 int64 hash resident in module buffers — replacing the reference's
 cuCollections ``static_map`` (``embedding_lookup_kernels.cu:383-516``) with a
 two-kernel CDNA4 design (free-slot scan, then insert-and-find with
-device-scope 64-bit atomicCAS).  On CPU a dict-based path with identical
-semantics is used (parity: the reference's ``DenseHashTable`` CPU path,
-``embedding.py:228-253``).
+device-scope 64-bit atomicCAS).  The CPU path is fully vectorized (numpy
+probing over the same buffer layout — parity with the reference's
+``DenseHashTable`` CPU path, ``embedding.py:228-253``, which is tensorized,
+not a Python loop).
 
 Semantics:
 * value 0 is reserved for OOV / overflow (slot pre-claimed at init,
   parity ``embedding.py:217-220``);
 * new keys are assigned the next free value (1, 2, ...) on first sight;
-* when the table is full, unseen keys map to 0;
+* when the table is full, unseen keys map to 0 (with ``auto_grow=True`` the
+  table instead rehashes into doubled capacity — an MI355X extension beyond
+  the reference's fixed ``max_tokens`` guess);
 * per-value frequency counts are maintained (``counts`` buffer).
 """
 
 from typing import List
 
+import numpy as np
 import torch
 from torch import nn
 
 from ..ops import _backend
 
 _LOAD_FACTOR = 1.5  # capacity multiplier, parity: reference embedding.py:226
+_GROW_AT = 0.8      # auto_grow trigger: assigned values / max_tokens
+
+
+def _mix64_np(k: np.ndarray) -> np.ndarray:
+    """splitmix64 finalizer on uint64 arrays — must match ``mix64`` in
+    embedding_ops.hip so CPU-inserted state probes identically on GPU."""
+    k = k.astype(np.uint64, copy=True)
+    k += np.uint64(0x9E3779B97F4A7C15)
+    k = (k ^ (k >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+    k = (k ^ (k >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+    return k ^ (k >> np.uint64(31))
 
 
 class IntegerLookup(nn.Module):
-    """Maps arbitrary int64 keys to a dense [0, max_tokens] vocabulary."""
+    """Maps arbitrary int64 keys to a dense [0, max_tokens] vocabulary.
 
-    def __init__(self, max_tokens: int = 100000, device=None):
+    Args:
+      max_tokens: vocabulary budget (value 0 is reserved for OOV).
+      auto_grow: if True, reaching 80% load (or overflowing) doubles
+        ``max_tokens`` and rehashes instead of mapping new keys to OOV.
+        Costs one device sync per call on GPU; keep False (reference
+        parity) on hot serving paths.
+    """
+
+    def __init__(self, max_tokens: int = 100000, auto_grow: bool = False,
+                 device=None):
         super().__init__()
         if max_tokens <= 0:
             raise ValueError("max_tokens must be positive")
         self.max_tokens = int(max_tokens)
+        self.auto_grow = bool(auto_grow)
         self.capacity = int(_LOAD_FACTOR * (self.max_tokens + 1))
         # GPU-resident hash state (also the single source of truth for
-        # checkpointing; CPU path mirrors into it lazily).
+        # checkpointing; the CPU path probes the same layout via numpy views).
         self.register_buffer("table_keys", torch.full((self.capacity,), -1, dtype=torch.int64,
                                                       device=device))
         self.register_buffer("table_values", torch.zeros(self.capacity, dtype=torch.int64,
@@ -47,70 +72,164 @@ class IntegerLookup(nn.Module):
         counts = torch.zeros(self.max_tokens + 1, dtype=torch.int32, device=device)
         counts[0] = 1
         self.register_buffer("counts", counts)
-        self._cpu_map = None  # lazy dict for the CPU path
 
     # ------------------------------------------------------------------ CPU
 
-    def _cpu_state(self):
-        if self._cpu_map is None:
-            self._cpu_map = {}
-            keys = self.table_keys.cpu()
-            vals = self.table_values.cpu()
-            occupied = keys >= 0
-            for k, v in zip(keys[occupied].tolist(), vals[occupied].tolist()):
-                self._cpu_map[k] = v
-        return self._cpu_map
+    def _probe_np(self, uniq: np.ndarray):
+        """Vectorized linear probe: per unique key returns (value, found)."""
+        tk = self.table_keys.numpy()
+        tv = self.table_values.numpy()
+        cap = self.capacity
+        slot = (_mix64_np(uniq) % np.uint64(cap)).astype(np.int64)
+        vals = np.zeros(len(uniq), dtype=np.int64)
+        found = np.zeros(len(uniq), dtype=bool)
+        active = np.arange(len(uniq))
+        for _ in range(cap):
+            if not active.size:
+                break
+            s = slot[active]
+            k = tk[s]
+            hit = k == uniq[active]
+            empty = k == -1
+            if hit.any():
+                vals[active[hit]] = tv[s[hit]]
+                found[active[hit]] = True
+            cont = ~(hit | empty)
+            active = active[cont]
+            slot[active] = (slot[active] + 1) % cap
+        return vals, found
+
+    def _insert_np(self, keys_np: np.ndarray, vals_np: np.ndarray):
+        """Vectorized insert of UNIQUE keys with pre-assigned values.
+
+        Collision rounds: contenders probe, first contender per free slot
+        wins, losers (and occupied-slot probes) advance — each round is pure
+        numpy over the still-pending set.
+        """
+        tk = self.table_keys.numpy()
+        tv = self.table_values.numpy()
+        cap = self.capacity
+        slot = (_mix64_np(keys_np) % np.uint64(cap)).astype(np.int64)
+        pending = np.arange(len(keys_np))
+        while pending.size:
+            s = slot[pending]
+            occupied = tk[s] != -1
+            cand = pending[~occupied]
+            if cand.size:
+                cs = s[~occupied]
+                order = np.argsort(cs, kind="stable")
+                cs_o, cand_o = cs[order], cand[order]
+                first = np.ones(cs_o.size, dtype=bool)
+                first[1:] = cs_o[1:] != cs_o[:-1]
+                win, ws = cand_o[first], cs_o[first]
+                tk[ws] = keys_np[win]
+                tv[ws] = vals_np[win]
+                lose = cand_o[~first]
+            else:
+                lose = pending[:0]
+            adv = np.concatenate([pending[occupied], lose])
+            slot[adv] = (slot[adv] + 1) % cap
+            pending = adv
 
     def _forward_cpu(self, keys: torch.Tensor) -> torch.Tensor:
-        m = self._cpu_state()
-        counts = self.counts
-        next_val = int((counts > 0).sum().item())
-        out = torch.empty_like(keys)
-        flat_in = keys.reshape(-1)
-        flat_out = out.reshape(-1)
-        for i, k in enumerate(flat_in.tolist()):
-            v = m.get(k)
-            if v is None:
-                if next_val <= self.max_tokens:
-                    v = next_val
-                    m[k] = v
-                    next_val += 1
-                    # persist into the hash buffers (linear probe)
-                    self._cpu_insert(k, v)
-                else:
-                    v = 0
-            counts[v] += 1
-            flat_out[i] = v
-        return out
+        flat = keys.reshape(-1).numpy()
+        uniq, first_idx, inverse = np.unique(flat, return_index=True,
+                                             return_inverse=True)
+        vals, found = self._probe_np(uniq)
+        miss = np.flatnonzero(~found)
+        if miss.size:
+            # assign values in first-occurrence order (reference CPU parity)
+            miss = miss[np.argsort(first_idx[miss], kind="stable")]
+            next_val = self.vocabulary_size()
+            if self.auto_grow:
+                while next_val + miss.size > self.max_tokens + 1 or \
+                        next_val + miss.size > _GROW_AT * self.max_tokens:
+                    self._grow()
+            n_assign = max(0, min(miss.size, self.max_tokens + 1 - next_val))
+            take = miss[:n_assign]
+            new_vals = next_val + np.arange(n_assign, dtype=np.int64)
+            vals[take] = new_vals
+            self._insert_np(uniq[take], new_vals)
+            # overflow (auto_grow off): remaining keys stay at value 0
+        out = vals[inverse]
+        self.counts += torch.from_numpy(
+            np.bincount(out, minlength=self.counts.numel()).astype(np.int32))
+        return torch.from_numpy(out).view(keys.shape)
 
-    @staticmethod
-    def _mix64(k: int) -> int:
-        """splitmix64 finalizer — must match ``mix64`` in embedding_ops.hip so
-        CPU-inserted state probes identically on GPU."""
-        m = (1 << 64) - 1
-        k = (k + 0x9E3779B97F4A7C15) & m
-        k = ((k ^ (k >> 30)) * 0xBF58476D1CE4E5B9) & m
-        k = ((k ^ (k >> 27)) * 0x94D049BB133111EB) & m
-        return (k ^ (k >> 31)) & m
+    # ------------------------------------------------------------------ grow
 
-    def _cpu_insert(self, key: int, value: int):
-        cap = self.capacity
-        slot = self._mix64(key & ((1 << 64) - 1)) % cap
-        while int(self.table_keys[slot]) != -1:
-            slot = (slot + 1) % cap
-        self.table_keys[slot] = key
-        self.table_values[slot] = value
+    def _grow(self):
+        """Double max_tokens, rehash all pairs into the larger table."""
+        self.max_tokens *= 2
+        new_cap = int(_LOAD_FACTOR * (self.max_tokens + 1))
+        dev = self.table_keys.device
+        new_keys = torch.full((new_cap,), -1, dtype=torch.int64, device=dev)
+        new_vals = torch.zeros(new_cap, dtype=torch.int64, device=dev)
+        if self.table_keys.is_cuda:
+            _backend.ops().hash_rehash(self.table_keys, self.table_values,
+                                       new_keys, new_vals)
+        else:
+            tk = self.table_keys.numpy()
+            occ = tk != -1
+            old_k = tk[occ].copy()
+            old_v = self.table_values.numpy()[occ].copy()
+            self.table_keys = new_keys
+            self.table_values = new_vals
+            self.capacity = new_cap
+            self._insert_np(old_k, old_v)
+            # counts extension below; keys/values already swapped
+            new_counts = torch.zeros(self.max_tokens + 1, dtype=torch.int32)
+            new_counts[:self.counts.numel()] = self.counts
+            self.counts = new_counts
+            return
+        self.table_keys = new_keys
+        self.table_values = new_vals
+        self.capacity = new_cap
+        new_counts = torch.zeros(self.max_tokens + 1, dtype=torch.int32,
+                                 device=dev)
+        new_counts[:self.counts.numel()] = self.counts
+        self.counts = new_counts
 
     # ------------------------------------------------------------------ GPU
 
     def forward(self, keys: torch.Tensor) -> torch.Tensor:
         if keys.dtype != torch.int64:
             keys = keys.long()
-        if keys.is_cuda:
-            return _backend.ops().integer_lookup(
-                keys.contiguous().reshape(-1), self.table_keys, self.table_values,
-                self.counts, self.max_tokens).view(keys.shape)
-        return self._forward_cpu(keys)
+        if not keys.is_cuda:
+            return self._forward_cpu(keys)
+        if self.auto_grow:
+            # proactive growth at 80% load (one D2H sync — documented cost)
+            while self.vocabulary_size() > _GROW_AT * self.max_tokens:
+                self._grow()
+        out = _backend.ops().integer_lookup(
+            keys.contiguous().reshape(-1), self.table_keys, self.table_values,
+            self.counts, self.max_tokens)
+        if self.auto_grow and self.vocabulary_size() >= self.max_tokens + 1:
+            # table filled mid-batch: some new keys overflowed to 0 without
+            # being inserted.  Retract this batch's counts, grow, re-run.
+            self.counts.index_put_((out,),
+                                   torch.full_like(out, -1, dtype=torch.int32).to(torch.int32),
+                                   accumulate=True)
+            self._grow()
+            return self.forward(keys)
+        return out.view(keys.shape)
+
+    # --------------------------------------------------------- (de)serialize
+
+    def _load_from_state_dict(self, state_dict, prefix, *args, **kwargs):
+        # a grown checkpoint may be larger than this module's fresh buffers:
+        # resize to the incoming shapes before the standard copy
+        tk = state_dict.get(prefix + "table_keys")
+        if tk is not None and tk.numel() != self.table_keys.numel():
+            dev = self.table_keys.device
+            self.capacity = tk.numel()
+            self.table_keys = torch.empty_like(tk, device=dev)
+            self.table_values = torch.empty_like(tk, device=dev)
+        cn = state_dict.get(prefix + "counts")
+        if cn is not None and cn.numel() != self.counts.numel():
+            self.max_tokens = cn.numel() - 1
+            self.counts = torch.empty_like(cn, device=self.counts.device)
+        super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
 
     # ------------------------------------------------------------- inspection
 

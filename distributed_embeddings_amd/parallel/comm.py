@@ -19,10 +19,23 @@ All functions are world_size==1 passthrough-safe and work on both the RCCL
 ("nccl") and gloo backends.
 """
 
+import os
 from typing import List, Optional, Sequence
 
 import torch
 import torch.distributed as dist
+
+
+def _strict_check(t: torch.Tensor):
+    """Audit mode (DE_COMM_DEVICE=cuda): fail loudly when a CPU tensor would
+    enter a collective.  On an 8-GPU RCCL run such a tensor aborts the job
+    deep inside NCCL; this surfaces it in the world-2 gloo+CUDA test lane
+    (RCCL refuses 2 ranks on one device — see tools/probe_rccl2.py — so this
+    audit is how single-GPU leases validate device placement)."""
+    if os.environ.get("DE_COMM_DEVICE") == "cuda" and not t.is_cuda:
+        raise RuntimeError(
+            f"CPU tensor (shape {tuple(t.shape)}, dtype {t.dtype}) entering a "
+            "collective under DE_COMM_DEVICE=cuda — this would fail on RCCL")
 
 
 def is_initialized() -> bool:
@@ -101,21 +114,29 @@ class _AllToAllSingle(torch.autograd.Function):
         ctx.group = group
         ctx.out_splits = out_splits
         ctx.in_splits = in_splits
+        _strict_check(inp)
+        # execute on the backend's device (no-op copy on the matched path:
+        # nccl+cuda or gloo+cpu); restore to the caller's device
+        dev = backend_device(group)
+        src = inp.contiguous().to(dev)
         n_out = sum(out_splits) if out_splits is not None else inp.shape[0]
-        out = inp.new_empty((n_out,) + tuple(inp.shape[1:]))
-        dist.all_to_all_single(out, inp.contiguous(),
+        out = src.new_empty((n_out,) + tuple(inp.shape[1:]))
+        dist.all_to_all_single(out, src,
                                output_split_sizes=out_splits,
                                input_split_sizes=in_splits, group=group)
-        return out
+        return out.to(inp.device)
 
     @staticmethod
     def backward(ctx, grad_out):
+        _strict_check(grad_out)
+        dev = backend_device(ctx.group)
+        src = grad_out.contiguous().to(dev)
         n_in = sum(ctx.in_splits) if ctx.in_splits is not None else grad_out.shape[0]
-        grad_in = grad_out.new_empty((n_in,) + tuple(grad_out.shape[1:]))
-        dist.all_to_all_single(grad_in, grad_out.contiguous(),
+        grad_in = src.new_empty((n_in,) + tuple(grad_out.shape[1:]))
+        dist.all_to_all_single(grad_in, src,
                                output_split_sizes=ctx.in_splits,
                                input_split_sizes=ctx.out_splits, group=ctx.group)
-        return grad_in, None, None, None
+        return grad_in.to(grad_out.device), None, None, None
 
 
 def all_to_all_single(
@@ -136,16 +157,22 @@ class _AllGather(torch.autograd.Function):
     def forward(ctx, inp, group):
         ctx.group = group
         ctx.in_rows = inp.shape[0]
+        _strict_check(inp)
+        dev = backend_device(group)
+        src = inp.contiguous().to(dev)
         w = world_size(group)
-        out = inp.new_empty((inp.shape[0] * w,) + tuple(inp.shape[1:]))
-        dist.all_gather_into_tensor(out, inp.contiguous(), group=group)
-        return out
+        out = src.new_empty((inp.shape[0] * w,) + tuple(inp.shape[1:]))
+        dist.all_gather_into_tensor(out, src, group=group)
+        return out.to(inp.device)
 
     @staticmethod
     def backward(ctx, grad_out):
-        grad_in = grad_out.new_empty((ctx.in_rows,) + tuple(grad_out.shape[1:]))
-        dist.reduce_scatter_tensor(grad_in, grad_out.contiguous(), group=ctx.group)
-        return grad_in, None
+        _strict_check(grad_out)
+        dev = backend_device(ctx.group)
+        src = grad_out.contiguous().to(dev)
+        grad_in = src.new_empty((ctx.in_rows,) + tuple(grad_out.shape[1:]))
+        dist.reduce_scatter_tensor(grad_in, src, group=ctx.group)
+        return grad_in.to(grad_out.device), None
 
 
 def all_gather(inp: torch.Tensor, group=None) -> torch.Tensor:
@@ -159,17 +186,23 @@ class _ReduceScatter(torch.autograd.Function):
     @staticmethod
     def forward(ctx, inp, group):
         ctx.group = group
+        _strict_check(inp)
+        dev = backend_device(group)
+        src = inp.contiguous().to(dev)
         w = world_size(group)
-        out = inp.new_empty((inp.shape[0] // w,) + tuple(inp.shape[1:]))
-        dist.reduce_scatter_tensor(out, inp.contiguous(), group=group)
-        return out
+        out = src.new_empty((inp.shape[0] // w,) + tuple(inp.shape[1:]))
+        dist.reduce_scatter_tensor(out, src, group=group)
+        return out.to(inp.device)
 
     @staticmethod
     def backward(ctx, grad_out):
+        _strict_check(grad_out)
+        dev = backend_device(ctx.group)
+        src = grad_out.contiguous().to(dev)
         w = world_size(ctx.group)
-        grad_in = grad_out.new_empty((grad_out.shape[0] * w,) + tuple(grad_out.shape[1:]))
-        dist.all_gather_into_tensor(grad_in, grad_out.contiguous(), group=ctx.group)
-        return grad_in, None
+        grad_in = src.new_empty((grad_out.shape[0] * w,) + tuple(grad_out.shape[1:]))
+        dist.all_gather_into_tensor(grad_in, src, group=ctx.group)
+        return grad_in.to(grad_out.device), None
 
 
 def reduce_scatter(inp: torch.Tensor, group=None) -> torch.Tensor:
@@ -208,11 +241,25 @@ def all_gather_uneven(inp: torch.Tensor, group=None) -> List[torch.Tensor]:
 
 def broadcast(t: torch.Tensor, src: int = 0, group=None) -> torch.Tensor:
     if world_size(group) > 1:
-        dist.broadcast(t, src=src, group=group)
+        dev = backend_device(group)
+        if t.device != dev:
+            # cold-path utility: stage through the backend's device (e.g. a
+            # CPU-offloaded table broadcast on an nccl job)
+            staged = t.to(dev)
+            dist.broadcast(staged, src=src, group=group)
+            t.copy_(staged)
+        else:
+            dist.broadcast(t, src=src, group=group)
     return t
 
 
 def allreduce_sum_(t: torch.Tensor, group=None) -> torch.Tensor:
     if world_size(group) > 1:
-        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=group)
+        dev = backend_device(group)
+        if t.device != dev:
+            staged = t.to(dev)
+            dist.all_reduce(staged, op=dist.ReduceOp.SUM, group=group)
+            t.copy_(staged)
+        else:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM, group=group)
     return t

@@ -401,7 +401,13 @@ class DistributedEmbedding(nn.Module):
         return out
 
     def _comm_device(self):
+        import os
         import torch.distributed as dist
+        ov = os.environ.get("DE_COMM_DEVICE")
+        if ov == "cuda":
+            # test-lane override: build comm buffers as an nccl job would
+            # (see comm._strict_check)
+            return torch.device("cuda", torch.cuda.current_device())
         if self.world_size > 1 and dist.get_backend() == "nccl":
             return torch.device("cuda", torch.cuda.current_device())
         if self.world_size == 1:

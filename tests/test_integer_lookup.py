@@ -70,3 +70,44 @@ def test_state_dict_round_trip():
     # unseen key gets a NEW value, not a collision with restored ones
     v_new = int(lk2(torch.tensor([99]))[0])
     assert v_new not in vals2[:4].tolist()
+
+
+def test_cpu_path_vectorized_1m_keys():
+    """the CPU path must handle 1M keys in seconds (it is
+    numpy-vectorized probing, not a per-element Python loop)."""
+    import time
+    lk = IntegerLookup(max_tokens=1_200_000)
+    keys = torch.randint(0, 1 << 62, (1_000_000,))
+    t0 = time.time()
+    out = lk(keys)
+    assert time.time() - t0 < 30.0
+    assert torch.equal(out, lk(keys))
+    assert lk.vocabulary_size() == len(set(keys.tolist())) + 1
+
+
+def test_auto_grow_cpu():
+    """auto_grow=True rehashes past the max_tokens guess instead of OOV."""
+    lk = IntegerLookup(max_tokens=4, auto_grow=True)
+    out = lk(torch.arange(100, 120))
+    assert lk.max_tokens >= 20
+    assert (out > 0).all()
+    assert torch.equal(out, lk(torch.arange(100, 120)))
+
+
+def test_auto_grow_preserves_assignments():
+    lk = IntegerLookup(max_tokens=8, auto_grow=True)
+    a = lk(torch.tensor([5, 6, 7]))
+    lk(torch.arange(1000, 1030))  # force growth
+    assert torch.equal(a, lk(torch.tensor([5, 6, 7])))
+    # counts survive growth: keys 5/6/7 seen twice each
+    for v in a.tolist():
+        assert int(lk.counts[v]) == 2
+
+
+def test_grown_state_dict_loads_into_fresh_module():
+    lk = IntegerLookup(max_tokens=4, auto_grow=True)
+    out = lk(torch.arange(50, 70))
+    lk2 = IntegerLookup(max_tokens=4, auto_grow=True)
+    lk2.load_state_dict(lk.state_dict())
+    assert lk2.max_tokens == lk.max_tokens
+    assert torch.equal(lk2(torch.arange(50, 70)), out)
