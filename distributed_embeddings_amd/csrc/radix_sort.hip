@@ -9,24 +9,29 @@
 //      [bucket][block] for the stable global scan;
 //   2. exclusive scan over the [256 x nblocks] table (hand-written:
 //      per-chunk partials + single-wave top scan + add-back);
-//   3. rs_scatter_k: 4 waves per block, each owning a consecutive quarter of
-//      the 4096-element tile with a PRIVATE running-offset row (barrier-free
-//      scatter loop); within each 64-element group the stable rank comes
-//      from 8 __ballot rounds (bucket-bit match masks) + popcount over the
-//      preceding-lane mask.  Per-wave bases fold in earlier quarters' counts
-//      so the sort stays stable.
+//   3. rs_scatter_k: WAVES waves per block, each owning a consecutive
+//      1/WAVES slice of the tile with a PRIVATE running-offset row
+//      (barrier-free scatter loop); within each 64-element group the stable
+//      rank comes from 8 __ballot rounds (bucket-bit match masks) + popcount
+//      over the preceding-lane mask.  Per-wave bases fold in earlier slices'
+//      counts so the sort stays stable.
+//
+// Tile geometry is a (WAVES, IPT) template: at DLRM-backward sizes
+// (~200k keys) the original 4x16 = 4096-element tile yields only ~52
+// blocks — far below what 256 CUs need — so smaller tiles win despite the
+// larger histogram/scan table.  The variant is chosen at dispatch
+// (DE_SORT_VARIANT env overrides for measurement; see tools/bench_sort3.py).
 // Keys' digit bits must be non-negative ids (the pipeline masks OOB ids to
 // the `vocab` sentinel before packing).
 
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
 #include "ops_api.h"
 
 #define WAVE 64
 #define RS_RADIX 256
-#define RS_IPT 16                      // items per thread (per lane)
-#define RS_WAVES 4
-#define RS_TILE (RS_WAVES * WAVE * RS_IPT)  // elements per block (4096)
 
 static inline int64_t rs_cdiv(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
@@ -94,60 +99,65 @@ __global__ void rs_scan_addback(int32_t* __restrict__ out, int64_t m,
 // Keys-only variant: the backward pipeline packs (id << 32 | position) into
 // one u64, so the sort moves a single array (half the scatter write
 // traffic of pairs) and digits start at bit 32.
+template <int WAVES, int IPT>
 __global__ void rs_histogram_k(const uint64_t* __restrict__ keys, int64_t n,
                                int shift, int32_t* __restrict__ hist,
                                int64_t nblocks) {
+  constexpr int TILE = WAVES * WAVE * IPT;
   __shared__ int32_t lhist[RS_RADIX];
   const int t = threadIdx.x;
-  if (t < RS_RADIX) lhist[t] = 0;
+  for (int b = t; b < RS_RADIX; b += WAVES * WAVE) lhist[b] = 0;
   __syncthreads();
-  const int64_t start = (int64_t)blockIdx.x * RS_TILE;
-  const int64_t end = min(start + (int64_t)RS_TILE, n);
-  for (int64_t i = start + t; i < end; i += RS_WAVES * WAVE) {
+  const int64_t start = (int64_t)blockIdx.x * TILE;
+  const int64_t end = min(start + (int64_t)TILE, n);
+  for (int64_t i = start + t; i < end; i += WAVES * WAVE) {
     const int b = (int)((keys[i] >> shift) & 0xff);
     atomicAdd(&lhist[b], 1);
   }
   __syncthreads();
-  if (t < RS_RADIX) hist[(int64_t)t * nblocks + blockIdx.x] = lhist[t];
+  for (int b = t; b < RS_RADIX; b += WAVES * WAVE)
+    hist[(int64_t)b * nblocks + blockIdx.x] = lhist[b];
 }
 
-// 4 waves per block; each wave owns a consecutive quarter of the tile with a
-// PRIVATE running-offset row (no barriers in the scatter loop).  Stability:
-// per-wave bases include the counts of earlier quarters.
+// WAVES waves per block; each wave owns a consecutive slice of the tile with
+// a PRIVATE running-offset row (no barriers in the scatter loop).
+// Stability: per-wave bases include the counts of earlier slices.
+template <int WAVES, int IPT>
 __global__ void rs_scatter_k(const uint64_t* __restrict__ keys_in, int64_t n,
                              int shift, const int32_t* __restrict__ offsets,
                              int64_t nblocks, uint64_t* __restrict__ keys_out) {
-  __shared__ int32_t cnt[RS_WAVES][RS_RADIX];
-  __shared__ int32_t run[RS_WAVES][RS_RADIX];
+  constexpr int TILE = WAVES * WAVE * IPT;
+  __shared__ int32_t cnt[WAVES][RS_RADIX];
+  __shared__ int32_t run[WAVES][RS_RADIX];
   const int t = threadIdx.x;
   const int wave = t >> 6;
   const int lane = t & (WAVE - 1);
-  if (t < RS_RADIX) {
+  for (int b = t; b < RS_RADIX; b += WAVES * WAVE) {
 #pragma unroll
-    for (int w = 0; w < RS_WAVES; ++w) cnt[w][t] = 0;
+    for (int w = 0; w < WAVES; ++w) cnt[w][b] = 0;
   }
   __syncthreads();
-  const int64_t start = (int64_t)blockIdx.x * RS_TILE;
-  const int64_t end = min(start + (int64_t)RS_TILE, n);
-  const int64_t qs = min(start + (int64_t)wave * WAVE * RS_IPT, end);
-  const int64_t qe = min(qs + (int64_t)WAVE * RS_IPT, end);
-  // phase 1: per-quarter bucket counts
+  const int64_t start = (int64_t)blockIdx.x * TILE;
+  const int64_t end = min(start + (int64_t)TILE, n);
+  const int64_t qs = min(start + (int64_t)wave * WAVE * IPT, end);
+  const int64_t qe = min(qs + (int64_t)WAVE * IPT, end);
+  // phase 1: per-slice bucket counts
   for (int64_t i = qs + lane; i < qe; i += WAVE) {
     const int b = (int)((keys_in[i] >> shift) & 0xff);
     atomicAdd(&cnt[wave][b], 1);
   }
   __syncthreads();
-  // phase 2: per-wave bases = global offset + earlier quarters' counts
-  if (t < RS_RADIX) {
-    int base = offsets[(int64_t)t * nblocks + blockIdx.x];
+  // phase 2: per-wave bases = global offset + earlier slices' counts
+  for (int b = t; b < RS_RADIX; b += WAVES * WAVE) {
+    int base = offsets[(int64_t)b * nblocks + blockIdx.x];
 #pragma unroll
-    for (int w = 0; w < RS_WAVES; ++w) {
-      run[w][t] = base;
-      base += cnt[w][t];
+    for (int w = 0; w < WAVES; ++w) {
+      run[w][b] = base;
+      base += cnt[w][b];
     }
   }
   __syncthreads();
-  // phase 3: each wave scatters its quarter, 64-element groups in order
+  // phase 3: each wave scatters its slice, 64-element groups in order
   for (int64_t base = qs; base < qe; base += WAVE) {
     const int64_t i = base + lane;
     const bool active = i < qe;
@@ -176,11 +186,51 @@ __global__ void rs_scatter_k(const uint64_t* __restrict__ keys_in, int64_t n,
   }
 }
 
+namespace {
+
+struct SortVariant {
+  int waves, ipt;
+  void (*hist)(const uint64_t*, int64_t, int, int32_t*, int64_t);
+  void (*scat)(const uint64_t*, int64_t, int, const int32_t*, int64_t,
+               uint64_t*);
+};
+
+template <int WAVES, int IPT>
+constexpr SortVariant make_variant() {
+  return {WAVES, IPT, rs_histogram_k<WAVES, IPT>, rs_scatter_k<WAVES, IPT>};
+}
+
+// index 0 is the measured default at DLRM-backward sizes; others selectable
+// via DE_SORT_VARIANT for measurement (tools/bench_sort3.py sweeps these).
+const SortVariant kVariants[] = {
+    make_variant<4, 4>(),   // tile 1024
+    make_variant<4, 16>(),  // tile 4096 (round-1 geometry)
+    make_variant<8, 8>(),   // tile 4096, 512 threads
+    make_variant<4, 8>(),   // tile 2048
+    make_variant<8, 4>(),   // tile 2048, 512 threads
+    make_variant<2, 4>(),   // tile 512
+};
+
+int pick_variant() {
+  static int v = [] {
+    const char* e = std::getenv("DE_SORT_VARIANT");
+    if (!e) return 0;
+    int i = std::atoi(e);
+    const int nv = (int)(sizeof(kVariants) / sizeof(kVariants[0]));
+    return (i >= 0 && i < nv) ? i : 0;
+  }();
+  return v;
+}
+
+}  // namespace
+
 void custom_radix_sort_keys(const uint64_t* keys_in, uint64_t* keys_out,
                             uint64_t* keys_tmp, int32_t* hist,
                             int32_t* scan_sums, int64_t n, int begin_bit,
                             int end_bit, hipStream_t stream) {
-  const int64_t nblocks = rs_cdiv(n, RS_TILE);
+  const SortVariant& var = kVariants[pick_variant()];
+  const int tile = var.waves * WAVE * var.ipt;
+  const int64_t nblocks = rs_cdiv(n, tile);
   const int passes = (end_bit - begin_bit + 7) / 8;
   const int64_t m = (int64_t)RS_RADIX * nblocks;
   const int chunk = (int)rs_cdiv(m, 2048) < 64 ? 64 : (int)rs_cdiv(m, 2048);
@@ -192,8 +242,8 @@ void custom_radix_sort_keys(const uint64_t* keys_in, uint64_t* keys_out,
   for (int p = 0; p < passes; ++p) {
     kout = to_tmp ? keys_tmp : keys_out;
     const int shift = begin_bit + p * 8;
-    hipLaunchKernelGGL(rs_histogram_k, dim3((int)nblocks),
-                       dim3(RS_WAVES * WAVE), 0, stream, kin, n, shift, hist,
+    hipLaunchKernelGGL(var.hist, dim3((int)nblocks),
+                       dim3(var.waves * WAVE), 0, stream, kin, n, shift, hist,
                        nblocks);
     hipLaunchKernelGGL(rs_scan_partials, dim3(nb_scan), dim3(WAVE), 0, stream,
                        hist, m, scan_sums, hist, chunk);
@@ -201,8 +251,8 @@ void custom_radix_sort_keys(const uint64_t* keys_in, uint64_t* keys_out,
                        nb_scan);
     hipLaunchKernelGGL(rs_scan_addback, dim3(nb_scan), dim3(256), 0, stream,
                        hist, m, scan_sums, chunk);
-    hipLaunchKernelGGL(rs_scatter_k, dim3((int)nblocks),
-                       dim3(RS_WAVES * WAVE), 0, stream, kin, n, shift, hist,
+    hipLaunchKernelGGL(var.scat, dim3((int)nblocks),
+                       dim3(var.waves * WAVE), 0, stream, kin, n, shift, hist,
                        nblocks, kout);
     kin = kout;
     to_tmp = !to_tmp;
@@ -210,5 +260,11 @@ void custom_radix_sort_keys(const uint64_t* keys_in, uint64_t* keys_out,
 }
 
 size_t custom_radix_sort_hist_elems(int64_t n) {
-  return (size_t)RS_RADIX * rs_cdiv(n, RS_TILE);
+  // sized for the smallest tile any variant uses (most blocks)
+  int min_tile = kVariants[0].waves * WAVE * kVariants[0].ipt;
+  for (const auto& v : kVariants) {
+    const int t = v.waves * WAVE * v.ipt;
+    if (t < min_tile) min_tile = t;
+  }
+  return (size_t)RS_RADIX * rs_cdiv(n, min_tile);
 }

@@ -205,13 +205,19 @@ class IntegerLookup(nn.Module):
             keys.contiguous().reshape(-1), self.table_keys, self.table_values,
             self.counts, self.max_tokens)
         if self.auto_grow and self.vocabulary_size() >= self.max_tokens + 1:
-            # table filled mid-batch: some new keys overflowed to 0 without
-            # being inserted.  Retract this batch's counts, grow, re-run.
-            self.counts.index_put_((out,),
-                                   torch.full_like(out, -1, dtype=torch.int32).to(torch.int32),
-                                   accumulate=True)
-            self._grow()
-            return self.forward(keys)
+            # Table filled mid-batch: out==0 elements are unresolved keys
+            # (in auto_grow mode nothing is true-OOV).  Retract only their
+            # counts[0] contribution, grow, resolve just that subset —
+            # resolved keys keep their counts, so the counts>0 free-value
+            # invariant holds throughout.
+            unresolved = out == 0
+            n0 = int(unresolved.sum().item())
+            if n0:
+                self.counts[0] -= n0
+                self._grow()
+                out = out.clone()
+                out[unresolved] = self.forward(
+                    keys.contiguous().reshape(-1)[unresolved]).reshape(-1)
         return out.view(keys.shape)
 
     # --------------------------------------------------------- (de)serialize

@@ -530,3 +530,22 @@ def test_fused_sgd_dense_input_gpu_matches_cpu():
     assert torch.allclose(out_g.cpu(), out_c, atol=1e-4)
     d = (e_g.weight.detach().cpu() - e_c.weight.detach()).abs()
     assert float(d.max()) < 1e-3, float(d.max())
+
+
+@requires_gpu
+def test_integer_lookup_gpu_auto_grow():
+    """auto_grow=True: the hash_rehash kernel doubles capacity past the
+    max_tokens guess; existing assignments and counts survive."""
+    from distributed_embeddings_amd import IntegerLookup
+    lk = IntegerLookup(max_tokens=4, auto_grow=True).cuda()
+    a = lk(torch.tensor([5, 6, 7], device="cuda"))
+    out = lk(torch.arange(1000, 1030, device="cuda"))
+    assert lk.max_tokens >= 30
+    assert (out.cpu() > 0).all()
+    b = lk(torch.tensor([5, 6, 7], device="cuda"))
+    assert torch.equal(a.cpu(), b.cpu())
+    out2 = lk(torch.arange(1000, 1030, device="cuda"))
+    assert torch.equal(out.cpu(), out2.cpu())
+    # CPU/GPU interop after growth: the grown table probes identically on CPU
+    c = lk.cpu()(torch.tensor([5, 6, 7]))
+    assert torch.equal(a.cpu(), c)
