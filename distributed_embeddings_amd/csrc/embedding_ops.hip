@@ -1284,7 +1284,10 @@ __global__ void hash_reinsert(const int64_t* __restrict__ old_keys,
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= old_cap) return;
   const int64_t key = old_keys[i];
-  if (key == IL_EMPTY) return;
+  // drop value-0 entries: those are overflow keys hash_insert claimed while
+  // the table was full — after growth they must re-insert with a REAL value
+  // (keeping them would pin the key to OOV forever)
+  if (key == IL_EMPTY || old_vals[i] == 0) return;
   uint64_t slot = mix64((uint64_t)key) % (uint64_t)capacity;
   for (int64_t probe = 0; probe < capacity; ++probe) {
     int64_t prev = (int64_t)atomicCAS((unsigned long long*)&tkeys[slot],

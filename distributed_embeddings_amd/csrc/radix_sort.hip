@@ -200,26 +200,37 @@ constexpr SortVariant make_variant() {
   return {WAVES, IPT, rs_histogram_k<WAVES, IPT>, rs_scatter_k<WAVES, IPT>};
 }
 
-// index 0 is the measured default at DLRM-backward sizes; others selectable
-// via DE_SORT_VARIANT for measurement (tools/bench_sort3.py sweeps these).
+// Variants selectable via DE_SORT_VARIANT for measurement
+// (tools/bench_sort3.py sweeps these vs rocPRIM).
 const SortVariant kVariants[] = {
-    make_variant<4, 4>(),   // tile 1024
-    make_variant<4, 16>(),  // tile 4096 (round-1 geometry)
-    make_variant<8, 8>(),   // tile 4096, 512 threads
-    make_variant<4, 8>(),   // tile 2048
-    make_variant<8, 4>(),   // tile 2048, 512 threads
-    make_variant<2, 4>(),   // tile 512
+    make_variant<4, 4>(),   // 0: tile 1024
+    make_variant<4, 16>(),  // 1: tile 4096 (round-1 geometry)
+    make_variant<8, 8>(),   // 2: tile 4096, 512 threads
+    make_variant<4, 8>(),   // 3: tile 2048
+    make_variant<8, 4>(),   // 4: tile 2048, 512 threads
+    make_variant<2, 4>(),   // 5: tile 512
+    make_variant<8, 16>(),  // 6: tile 8192, 512 threads
+    make_variant<16, 8>(),  // 7: tile 8192, 1024 threads
 };
 
-int pick_variant() {
+int env_variant() {
   static int v = [] {
     const char* e = std::getenv("DE_SORT_VARIANT");
-    if (!e) return 0;
+    if (!e) return -1;
     int i = std::atoi(e);
     const int nv = (int)(sizeof(kVariants) / sizeof(kVariants[0]));
-    return (i >= 0 && i < nv) ? i : 0;
+    return (i >= 0 && i < nv) ? i : -1;
   }();
   return v;
+}
+
+const SortVariant& pick_variant(int64_t n) {
+  const int e = env_variant();
+  if (e >= 0) return kVariants[e];
+  // measured on MI355X (profiles/sort_sweep.md): small-tile 8-wave wins at
+  // DLRM-26-table sizes (needs blocks >> 256 CUs); bigger tiles amortize
+  // better once n is large
+  return n <= 512 * 1024 ? kVariants[4] : kVariants[2];
 }
 
 }  // namespace
@@ -228,7 +239,7 @@ void custom_radix_sort_keys(const uint64_t* keys_in, uint64_t* keys_out,
                             uint64_t* keys_tmp, int32_t* hist,
                             int32_t* scan_sums, int64_t n, int begin_bit,
                             int end_bit, hipStream_t stream) {
-  const SortVariant& var = kVariants[pick_variant()];
+  const SortVariant& var = pick_variant(n);
   const int tile = var.waves * WAVE * var.ipt;
   const int64_t nblocks = rs_cdiv(n, tile);
   const int passes = (end_bit - begin_bit + 7) / 8;

@@ -76,13 +76,15 @@ class IntegerLookup(nn.Module):
     # ------------------------------------------------------------------ CPU
 
     def _probe_np(self, uniq: np.ndarray):
-        """Vectorized linear probe: per unique key returns (value, found)."""
+        """Vectorized linear probe: per unique key returns
+        (value, found, slot-of-hit)."""
         tk = self.table_keys.numpy()
         tv = self.table_values.numpy()
         cap = self.capacity
         slot = (_mix64_np(uniq) % np.uint64(cap)).astype(np.int64)
         vals = np.zeros(len(uniq), dtype=np.int64)
         found = np.zeros(len(uniq), dtype=bool)
+        hit_slot = np.full(len(uniq), -1, dtype=np.int64)
         active = np.arange(len(uniq))
         for _ in range(cap):
             if not active.size:
@@ -94,10 +96,11 @@ class IntegerLookup(nn.Module):
             if hit.any():
                 vals[active[hit]] = tv[s[hit]]
                 found[active[hit]] = True
+                hit_slot[active[hit]] = s[hit]
             cont = ~(hit | empty)
             active = active[cont]
             slot[active] = (slot[active] + 1) % cap
-        return vals, found
+        return vals, found, hit_slot
 
     def _insert_np(self, keys_np: np.ndarray, vals_np: np.ndarray):
         """Vectorized insert of UNIQUE keys with pre-assigned values.
@@ -135,21 +138,31 @@ class IntegerLookup(nn.Module):
         flat = keys.reshape(-1).numpy()
         uniq, first_idx, inverse = np.unique(flat, return_index=True,
                                              return_inverse=True)
-        vals, found = self._probe_np(uniq)
+        vals, found, hit_slot = self._probe_np(uniq)
         miss = np.flatnonzero(~found)
-        if miss.size:
+        # under auto_grow, keys a full GPU table once pinned to value 0 are
+        # assignable again (their slot value upgrades in place)
+        upgrade = np.flatnonzero(found & (vals == 0)) if self.auto_grow \
+            else np.empty(0, dtype=np.int64)
+        need = np.concatenate([miss, upgrade])
+        if need.size:
             # assign values in first-occurrence order (reference CPU parity)
-            miss = miss[np.argsort(first_idx[miss], kind="stable")]
+            need = need[np.argsort(first_idx[need], kind="stable")]
             next_val = self.vocabulary_size()
             if self.auto_grow:
-                while next_val + miss.size > self.max_tokens + 1 or \
-                        next_val + miss.size > _GROW_AT * self.max_tokens:
+                while next_val + need.size > self.max_tokens + 1 or \
+                        next_val + need.size > _GROW_AT * self.max_tokens:
                     self._grow()
-            n_assign = max(0, min(miss.size, self.max_tokens + 1 - next_val))
-            take = miss[:n_assign]
+            n_assign = max(0, min(need.size, self.max_tokens + 1 - next_val))
+            take = need[:n_assign]
             new_vals = next_val + np.arange(n_assign, dtype=np.int64)
             vals[take] = new_vals
-            self._insert_np(uniq[take], new_vals)
+            ins = take[hit_slot[take] < 0]
+            upg = take[hit_slot[take] >= 0]
+            if ins.size:
+                self._insert_np(uniq[ins], vals[ins])
+            if upg.size:
+                self.table_values.numpy()[hit_slot[upg]] = vals[upg]
             # overflow (auto_grow off): remaining keys stay at value 0
         out = vals[inverse]
         self.counts += torch.from_numpy(
@@ -170,7 +183,9 @@ class IntegerLookup(nn.Module):
                                        new_keys, new_vals)
         else:
             tk = self.table_keys.numpy()
-            occ = tk != -1
+            # drop value-0 (overflow-pinned) entries — parity with the GPU
+            # hash_reinsert kernel
+            occ = (tk != -1) & (self.table_values.numpy() != 0)
             old_k = tk[occ].copy()
             old_v = self.table_values.numpy()[occ].copy()
             self.table_keys = new_keys

@@ -15,6 +15,7 @@ bmm fallback when the shape or dtype gate fails).
 """
 
 import math
+import os
 from typing import List, Optional, Sequence
 
 import torch
@@ -124,7 +125,15 @@ class DLRM(nn.Module):
 
     def forward(self, numerical: torch.Tensor,
                 cat_features: Sequence[torch.Tensor]) -> torch.Tensor:
+        cats = list(cat_features)
+        handle = None
+        if self.distributed and self.dp_input and \
+                os.environ.get("DE_OVERLAP_A2A", "1") != "0":
+            # post the dp->mp ID all-to-all before the bottom MLP: the xGMI
+            # id exchange overlaps the MLP GEMMs (ids carry no grad)
+            handle = self.embeddings.redistribute_async(cats)
         bottom = self.bottom_mlp(numerical)
-        emb = self.embeddings(list(cat_features), output_dtype=bottom.dtype)
+        emb = self.embeddings(cats, output_dtype=bottom.dtype,
+                              async_handle=handle)
         x = fused_dot_interact(emb, bottom, pad_to=self.interact_pad)
         return self.top_mlp(x)

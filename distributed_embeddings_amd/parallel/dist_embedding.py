@@ -260,7 +260,8 @@ class DistributedEmbedding(nn.Module):
     # ---------------------------------------------------------------- forward
 
     def forward(self, inputs: Sequence[Union[torch.Tensor, Ragged]],
-                output_dtype: Optional[torch.dtype] = None) -> List[torch.Tensor]:
+                output_dtype: Optional[torch.dtype] = None,
+                async_handle: Optional[dict] = None) -> List[torch.Tensor]:
         plan = self.strategy
         self._output_dtype = output_dtype
         dp_in, col_in, row_in = plan.input_groups
@@ -281,8 +282,8 @@ class DistributedEmbedding(nn.Module):
                   for j, x in enumerate(dp_inputs)]
         if output_dtype is not None:
             dp_out = [o.to(output_dtype) for o in dp_out]
-        col_out = self._call_table_parallel(col_inputs) if (col_inputs or plan.col_table_ids) \
-            else []
+        col_out = self._call_table_parallel(col_inputs, async_handle) \
+            if (col_inputs or plan.col_table_ids) else []
         row_out = self._call_row_slice(row_inputs) if row_inputs else []
 
         outs = dp_out + col_out + row_out
@@ -417,15 +418,10 @@ class DistributedEmbedding(nn.Module):
 
     # ----------------------------------------------------- table parallel path
 
-    def _dp_to_mp_dense(self, col_inputs):
-        """Dense id redistribution dp->mp (parity: reference ``:169-221``).
-
-        Returns per local pair: ids tensor of shape [W*b, ...feature dims].
-        """
+    def _dense_splits(self, col_inputs):
+        """Cached (in_splits, my_sizes, out_splits) per input-shape signature."""
         plan = self.strategy
         W = self.world_size
-        b = col_inputs[0].shape[0] if col_inputs else 0
-        # split sizes are static per input-shape signature: cache them
         sig = ("dp2mp", tuple(tuple(x.shape) for x in col_inputs))
         cache = getattr(self, "_split_cache", None)
         if cache is None:
@@ -440,13 +436,20 @@ class DistributedEmbedding(nn.Module):
                         for i in plan.rank_input_ids[self.rank]]
             cached = (in_splits, my_sizes, [sum(my_sizes)] * W)
             cache[sig] = cached
-        in_splits, my_sizes, out_splits = cached
+        return cached
+
+    def _dense_send_buffer(self, col_inputs):
+        plan = self.strategy
+        W = self.world_size
         send_parts = [col_inputs[i].reshape(-1)
                       for k in range(W) for i in plan.rank_input_ids[k]]
-        send = torch.cat(send_parts) if send_parts else \
+        return torch.cat(send_parts) if send_parts else \
             torch.empty(0, dtype=torch.long, device=self._comm_device())
-        recv = comm.all_to_all_single(send, out_splits, in_splits)
-        # recv: [W, sum(my_sizes)] -> per pair [W*b, ...]
+
+    def _carve_dense_recv(self, recv, my_sizes, col_inputs):
+        """recv: [W * sum(my_sizes)] flat ids -> per local pair [W*b, ...]."""
+        plan = self.strategy
+        W = self.world_size
         recv = recv.view(W, -1) if recv.numel() else recv.view(W, 0)
         parts = torch.split(recv, my_sizes, dim=1) if my_sizes else []
         out = []
@@ -454,6 +457,49 @@ class DistributedEmbedding(nn.Module):
             shape = col_inputs[i].shape
             out.append(parts[j].reshape(W * shape[0], *shape[1:]))
         return out
+
+    def _dp_to_mp_dense(self, col_inputs):
+        """Dense id redistribution dp->mp (parity: reference ``:169-221``).
+
+        Returns per local pair: ids tensor of shape [W*b, ...feature dims].
+        """
+        in_splits, my_sizes, out_splits = self._dense_splits(col_inputs)
+        send = self._dense_send_buffer(col_inputs)
+        recv = comm.all_to_all_single(send, out_splits, in_splits)
+        return self._carve_dense_recv(recv, my_sizes, col_inputs)
+
+    def redistribute_async(self, inputs) -> Optional[dict]:
+        """Posts the dp->mp ID all-to-all WITHOUT waiting, so the caller can
+        overlap independent compute (e.g. DLRM's bottom MLP) with the id
+        exchange on the xGMI links.  Ids carry no gradient, so no autograd
+        plumbing is needed.  Pass the returned handle to
+        ``forward(..., async_handle=h)``.
+
+        Returns None when there is nothing to overlap (world==1, mp-input
+        mode, no table-parallel inputs, or ragged inputs present — those take
+        the two-phase sync path).
+        """
+        import torch.distributed as dist
+        plan = self.strategy
+        if self.world_size == 1 or not self.dp_input or not plan.col_table_ids:
+            return None
+        col_in = plan.input_groups[1]
+        col_inputs = [inputs[i] for i in col_in]
+        if not col_inputs or any(isinstance(x, Ragged) for x in col_inputs):
+            return None
+        self._validate_batch(col_inputs[0].shape[0])
+        in_splits, my_sizes, out_splits = self._dense_splits(col_inputs)
+        send = self._dense_send_buffer(col_inputs)
+        dev = comm.backend_device()
+        src = send.contiguous().to(dev)
+        recv = src.new_empty(sum(out_splits))
+        work = dist.all_to_all_single(recv, src,
+                                      output_split_sizes=[int(v) for v in out_splits],
+                                      input_split_sizes=[int(v) for v in in_splits],
+                                      async_op=True)
+        return {"work": work, "recv": recv, "my_sizes": my_sizes,
+                "device": send.device,
+                "sig": tuple(tuple(x.shape) for x in col_inputs)}
 
     def _dp_to_mp_ragged(self, col_inputs, ragged_mask):
         """Ragged id redistribution dp->mp (two-phase; parity ``:115-166``).
@@ -671,7 +717,7 @@ class DistributedEmbedding(nn.Module):
     def _row_offset_vector(self, gi, spec, device):
         return self._offset_vector(gi, spec, device, kind="row")
 
-    def _call_table_parallel(self, col_inputs):
+    def _call_table_parallel(self, col_inputs, async_handle=None):
         plan = self.strategy
         W = self.world_size
         if not plan.col_table_ids:
@@ -685,7 +731,14 @@ class DistributedEmbedding(nn.Module):
                      else col_inputs[0].shape[0])
                 self._validate_batch(b)
                 local_b = b  # dp-side batch known on every rank
-            if any_ragged:
+            if async_handle is not None:
+                if async_handle["sig"] != tuple(tuple(x.shape) for x in col_inputs):
+                    raise ValueError("async_handle does not match these inputs")
+                async_handle["work"].wait()
+                recv = async_handle["recv"].to(async_handle["device"])
+                pair_ids = self._carve_dense_recv(
+                    recv, async_handle["my_sizes"], col_inputs)
+            elif any_ragged:
                 pair_ids = self._dp_to_mp_ragged(col_inputs, None)
             else:
                 pair_ids = self._dp_to_mp_dense(col_inputs)

@@ -47,7 +47,7 @@ def main():
         worker()
         return
     env = dict(os.environ, _SORT_WORKER="1")
-    for v in range(6):
+    for v in range(8):
         e = dict(env, DE_SORT_VARIANT=str(v))
         print(f"variant {v}: ", end="", flush=True)
         subprocess.run([sys.executable, __file__], env=e, check=True)
