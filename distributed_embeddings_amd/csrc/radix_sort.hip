@@ -228,9 +228,10 @@ const SortVariant& pick_variant(int64_t n) {
   const int e = env_variant();
   if (e >= 0) return kVariants[e];
   // measured on MI355X (profiles/sort_sweep.md): small-tile 8-wave wins at
-  // DLRM-26-table sizes (needs blocks >> 256 CUs); bigger tiles amortize
-  // better once n is large
-  return n <= 512 * 1024 ? kVariants[4] : kVariants[2];
+  // DLRM-26-table sizes (needs blocks >> 256 CUs; 178us vs rocPRIM 195us);
+  // the 16-wave/8192 tile amortizes best once n is large (506us vs 502us
+  // at 1.7M keys — parity)
+  return n <= 512 * 1024 ? kVariants[4] : kVariants[7];
 }
 
 }  // namespace

@@ -41,14 +41,17 @@ def broadcast_parameters(module_or_params, root: int = 0):
     else:
         params, buffers = list(module_or_params), []
     with torch.no_grad():
+        # comm.broadcast stages through the backend's device, so CPU-resident
+        # tensors (offloaded-table state, IntegerLookup buffers) survive an
+        # nccl job
         for p in params:
             if not is_local_param(p) and p.numel() > 0:
-                dist.broadcast(p.data, src=root)
+                comm.broadcast(p.data, src=root)
         for b in buffers:
             if (b.numel() > 0 and not is_local_param(b)
                     and (b.dtype.is_floating_point
                          or b.dtype in (torch.int32, torch.int64))):
-                dist.broadcast(b.data, src=root)
+                comm.broadcast(b.data, src=root)
 
 
 # Backwards-friendly alias matching the reference public name.
