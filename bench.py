@@ -103,12 +103,17 @@ def maybe_enable_tunableop():
     """
     try:
         import torch.cuda.tunable as tunable
-        csv = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                           "profiles", "tunableop_gfx950.csv")
-        if os.path.exists(csv) and torch.cuda.is_available():
+        if os.environ.get("PYTORCH_TUNABLEOP_TUNING") == "1":
+            return  # tuning run: TunableOp env config drives everything
+        base = os.path.join(os.path.dirname(os.path.abspath(__file__)), "profiles")
+        csvs = [os.path.join(base, f) for f in
+                ("tunableop_gfx950.csv", "tunableop_gfx950_bs64k.csv")]
+        csvs = [c for c in csvs if os.path.exists(c)]
+        if csvs and torch.cuda.is_available():
             tunable.enable(True)
             tunable.tuning_enable(False)
-            tunable.read_file(csv)
+            for c in csvs:
+                tunable.read_file(c)
     except Exception as e:
         print(f"# tunableop unavailable: {e}")
 
