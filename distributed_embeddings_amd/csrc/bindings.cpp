@@ -418,6 +418,51 @@ torch::Tensor dot_interact_fwd(torch::Tensor feats, int64_t out_w) {
   return out;
 }
 
+torch::Tensor dot_interact_fwd_packed(torch::Tensor bottom,
+                                      torch::Tensor packed, torch::Tensor perm,
+                                      int64_t out_w) {
+  CHECK_CUDA(bottom); CHECK_CUDA(packed); CHECK_CUDA(perm);
+  CHECK_CONTIG(bottom); CHECK_CONTIG(packed); CHECK_CONTIG(perm);
+  TORCH_CHECK(bottom.dtype() == torch::kBFloat16 &&
+              packed.dtype() == torch::kBFloat16, "bf16 required");
+  TORCH_CHECK(perm.dtype() == torch::kInt32, "perm must be int32");
+  TORCH_CHECK(bottom.dim() == 2 && packed.dim() == 3, "bottom [B,D], packed [P,B,D]");
+  const int64_t B = bottom.size(0);
+  const int P = (int)packed.size(0), D = (int)bottom.size(1);
+  TORCH_CHECK(packed.size(1) == B && packed.size(2) == D, "shape mismatch");
+  TORCH_CHECK(perm.numel() == P, "perm must have P entries");
+  const int F = P + 1;
+  TORCH_CHECK(F <= 32 && D % 32 == 0, "F<=32 and D%32==0 required");
+  const int tri_n = F * (F - 1) / 2;
+  TORCH_CHECK(out_w >= tri_n + D, "out width too small");
+  auto out = torch::empty({B, out_w}, bottom.options());
+  launch_dot_interact_fwd_packed(bottom.data_ptr(), packed.data_ptr(),
+                                 perm.data_ptr<int>(), out.data_ptr(), B, F, D,
+                                 (int)out_w, tri_n, current_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> dot_interact_bwd_packed(torch::Tensor gout,
+                                                   torch::Tensor bottom,
+                                                   torch::Tensor packed,
+                                                   torch::Tensor perm) {
+  CHECK_CUDA(gout); CHECK_CUDA(bottom); CHECK_CUDA(packed); CHECK_CUDA(perm);
+  CHECK_CONTIG(gout); CHECK_CONTIG(bottom); CHECK_CONTIG(packed);
+  TORCH_CHECK(gout.dtype() == torch::kBFloat16);
+  const int64_t B = bottom.size(0);
+  const int P = (int)packed.size(0), D = (int)bottom.size(1);
+  const int F = P + 1;
+  const int tri_n = F * (F - 1) / 2;
+  auto gbottom = torch::empty_like(bottom);
+  auto gpacked = torch::empty_like(packed);
+  launch_dot_interact_bwd_packed(gout.data_ptr(), bottom.data_ptr(),
+                                 packed.data_ptr(), perm.data_ptr<int>(),
+                                 gbottom.data_ptr(), gpacked.data_ptr(), B, F,
+                                 D, (int)gout.size(1), tri_n,
+                                 current_stream());
+  return {gbottom, gpacked};
+}
+
 torch::Tensor dot_interact_bwd(torch::Tensor gout, torch::Tensor feats) {
   CHECK_CUDA(gout); CHECK_CUDA(feats);
   CHECK_CONTIG(gout); CHECK_CONTIG(feats);
@@ -449,6 +494,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused sparse SGD/Adagrad row update (gfx950)");
   m.def("csr_fused_optimizer_apply", &csr_fused_optimizer_apply,
         "in-backward fused SGD/Adagrad update (gfx950)");
+  m.def("dot_interact_fwd_packed", &dot_interact_fwd_packed,
+        "pairwise-dot interaction on (bottom, packed, perm) (MFMA, gfx950)");
+  m.def("dot_interact_bwd_packed", &dot_interact_bwd_packed,
+        "packed interaction backward -> (gbottom, gpacked) (MFMA, gfx950)");
   m.def("dot_interact_fwd", &dot_interact_fwd,
         "fused DLRM pairwise-dot interaction forward (MFMA bf16, gfx950)");
   m.def("dot_interact_bwd", &dot_interact_bwd,

@@ -175,6 +175,200 @@ __global__ void dot_interact_bwd(const __hip_bfloat16* __restrict__ gout,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Packed variants: feats live as TWO tensors — bottom [B, D] (feature 0) and
+// packed [P, B, D] (P = F-1 embedding outputs, feature-major).  The packed
+// block is exactly the fused-group lookup output (world==1) or the mp->dp
+// all-to-all recv buffer (world>1) VIEWED in place, so the torch.stack /
+// per-pair split+merge copies disappear from the step.  `perm[f-1]` maps
+// feature f (model input order) to its packed row (worker order) — a small
+// device-resident buffer built once from the plan, keeping the interaction
+// column order identical at every world size.
+// ---------------------------------------------------------------------------
+
+template <int FMAX>
+__global__ void dot_interact_fwd_packed(
+    const __hip_bfloat16* __restrict__ bottom,
+    const __hip_bfloat16* __restrict__ packed,
+    const int* __restrict__ perm, __hip_bfloat16* __restrict__ out, int64_t B,
+    int F, int D, int out_w, int tri_n) {
+  extern __shared__ short lds_all[];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int ldst = D + 8;
+  short* lds = lds_all + wave * FMAX * ldst;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+
+  for (int64_t b = wave_id; b < B; b += n_waves) {
+    const int d8 = D / 8;
+    for (int i = lane; i < FMAX * d8; i += WAVE) {
+      const int row = i / d8;
+      const int col = (i % d8) * 8;
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (row == 0) {
+        v = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const short*>(bottom) + b * (int64_t)D + col);
+      } else if (row < F) {
+        const int64_t src_row = perm[row - 1];
+        v = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const short*>(packed) +
+            (src_row * B + b) * (int64_t)D + col);
+      }
+      *reinterpret_cast<bf16x8*>(&lds[row * ldst + col]) = v;
+    }
+
+    const int r16 = lane & 15;
+    const int khalf = lane >> 4;
+    const int tiles_mi[3] = {0, 1, 1};
+    const int tiles_ni[3] = {0, 0, 1};
+#pragma unroll
+    for (int t = 0; t < 3; ++t) {
+      const int mi = tiles_mi[t], ni = tiles_ni[t];
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      for (int k0 = 0; k0 < D; k0 += 32) {
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &lds[(mi * 16 + r16) * ldst + k0 + khalf * 8]);
+        bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+            &lds[(ni * 16 + r16) * ldst + k0 + khalf * 8]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
+      }
+      __hip_bfloat16* orow = out + b * (int64_t)out_w;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int i = mi * 16 + (lane >> 4) * 4 + reg;
+        const int j = ni * 16 + (lane & 15);
+        if (i > j && i < F && j < F) {
+          orow[i * (i - 1) / 2 + j] = __hip_bfloat16(acc[reg]);
+        }
+      }
+    }
+    short* orow_s = reinterpret_cast<short*>(out + b * (int64_t)out_w);
+    for (int c = lane; c < D; c += WAVE) {
+      orow_s[tri_n + c] = lds[c];
+    }
+    for (int c = tri_n + D + lane; c < out_w; c += WAVE) {
+      orow_s[c] = 0;
+    }
+  }
+}
+
+template <int FMAX>
+__global__ void dot_interact_bwd_packed(
+    const __hip_bfloat16* __restrict__ gout,
+    const __hip_bfloat16* __restrict__ bottom,
+    const __hip_bfloat16* __restrict__ packed, const int* __restrict__ perm,
+    __hip_bfloat16* __restrict__ gbottom, __hip_bfloat16* __restrict__ gpacked,
+    int64_t B, int F, int D, int out_w, int tri_n) {
+  extern __shared__ short lds_all[];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  short* lds = lds_all + wave * (FMAX * D + FMAX * FMAX);
+  short* gsym = lds + FMAX * D;
+  const int64_t n_waves = ((int64_t)gridDim.x * blockDim.x) >> 6;
+  const int64_t wave_id = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+
+  for (int64_t b = wave_id; b < B; b += n_waves) {
+    const int d8 = D / 8;
+    for (int i = lane; i < FMAX * d8; i += WAVE) {
+      const int row = i / d8;
+      const int col = (i % d8) * 8;
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (row == 0) {
+        v = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const short*>(bottom) + b * (int64_t)D + col);
+      } else if (row < F) {
+        const int64_t src_row = perm[row - 1];
+        v = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const short*>(packed) +
+            (src_row * B + b) * (int64_t)D + col);
+      }
+      *reinterpret_cast<bf16x8*>(&lds[row * D + col]) = v;
+    }
+    const __hip_bfloat16* grow = gout + b * (int64_t)out_w;
+    for (int idx = lane; idx < FMAX * FMAX; idx += WAVE) {
+      const int i = idx / FMAX, j = idx % FMAX;
+      float g = 0.f;
+      if (i < F && j < F && i != j) {
+        const int r = i > j ? i : j, c = i > j ? j : i;
+        g = float(grow[r * (r - 1) / 2 + c]);
+      }
+      __hip_bfloat16 hb(g);
+      gsym[idx] = *reinterpret_cast<short*>(&hb);
+    }
+
+    const int r16 = lane & 15;
+    const int khalf = lane >> 4;
+
+    for (int nj = 0; nj < D / 16; ++nj) {
+#pragma unroll
+      for (int mi = 0; mi < FMAX / 16; ++mi) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int k0 = 0; k0 < FMAX; k0 += 32) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &gsym[(mi * 16 + r16) * FMAX + k0 + khalf * 8]);
+          bf16x8 bfr;
+#pragma unroll
+          for (int r = 0; r < 8; ++r) {
+            bfr[r] = lds[(k0 + khalf * 8 + r) * D + nj * 16 + r16];
+          }
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int i = mi * 16 + (lane >> 4) * 4 + reg;
+          const int j = nj * 16 + (lane & 15);
+          if (i < F && j < D) {
+            float v = acc[reg];
+            if (i == 0) {
+              v += float(grow[tri_n + j]);
+              gbottom[b * (int64_t)D + j] = __hip_bfloat16(v);
+            } else {
+              const int64_t dst_row = perm[i - 1];
+              gpacked[(dst_row * B + b) * (int64_t)D + j] = __hip_bfloat16(v);
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
+void launch_dot_interact_fwd_packed(const void* bottom, const void* packed,
+                                    const int* perm, void* out, int64_t B,
+                                    int F, int D, int out_w, int tri_n,
+                                    hipStream_t stream) {
+  const int block = 256;
+  const int waves = block / WAVE;
+  int64_t blocks = (B + waves - 1) / waves;
+  if (blocks > 8192) blocks = 8192;
+  const size_t lds = (size_t)waves * 32 * (D + 8) * sizeof(short);
+  hipLaunchKernelGGL((dot_interact_fwd_packed<32>), dim3((int)blocks),
+                     dim3(block), lds, stream,
+                     (const __hip_bfloat16*)bottom,
+                     (const __hip_bfloat16*)packed, perm, (__hip_bfloat16*)out,
+                     B, F, D, out_w, tri_n);
+}
+
+void launch_dot_interact_bwd_packed(const void* gout, const void* bottom,
+                                    const void* packed, const int* perm,
+                                    void* gbottom, void* gpacked, int64_t B,
+                                    int F, int D, int out_w, int tri_n,
+                                    hipStream_t stream) {
+  const int block = 256;
+  const int waves = block / WAVE;
+  int64_t blocks = (B + waves - 1) / waves;
+  if (blocks > 8192) blocks = 8192;
+  const size_t lds = (size_t)waves * (32 * D + 32 * 32) * sizeof(short);
+  hipLaunchKernelGGL((dot_interact_bwd_packed<32>), dim3((int)blocks),
+                     dim3(block), lds, stream, (const __hip_bfloat16*)gout,
+                     (const __hip_bfloat16*)bottom,
+                     (const __hip_bfloat16*)packed, perm,
+                     (__hip_bfloat16*)gbottom, (__hip_bfloat16*)gpacked, B, F,
+                     D, out_w, tri_n);
+}
+
 void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,
                              int D, int out_w, int tri_n, hipStream_t stream) {
   const int block = 256;

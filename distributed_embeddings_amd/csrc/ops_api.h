@@ -80,6 +80,15 @@ void launch_sparse_row_update(void* weight, bool weight_bf16, float* state,
                               int64_t num_rows, int width, float lr, float eps,
                               bool adagrad, hipStream_t stream);
 
+void launch_dot_interact_fwd_packed(const void* bottom, const void* packed,
+                                    const int* perm, void* out, int64_t B,
+                                    int F, int D, int out_w, int tri_n,
+                                    hipStream_t stream);
+void launch_dot_interact_bwd_packed(const void* gout, const void* bottom,
+                                    const void* packed, const int* perm,
+                                    void* gbottom, void* gpacked, int64_t B,
+                                    int F, int D, int out_w, int tri_n,
+                                    hipStream_t stream);
 void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,
                              int D, int out_w, int tri_n, hipStream_t stream);
 void launch_dot_interact_bwd(const void* gout, const void* feats, void* gfeats,

@@ -23,6 +23,7 @@ from torch import nn
 
 from ..layers.embedding import Embedding, scaled_uniform_init
 from ..ops.dot_interact import dot_interact as fused_dot_interact
+from ..ops.dot_interact import dot_interact_packed
 from ..parallel import comm
 from ..parallel.dist_embedding import DistributedEmbedding
 
@@ -118,6 +119,19 @@ class DLRM(nn.Module):
             data_parallel_threshold=data_parallel_threshold,
             table_dtype=table_dtype)
 
+        # packed interaction fast path: the fused-group lookup output (or the
+        # a2a recv buffer) feeds dot_interact as one [F-1, B, D] view; the
+        # plan's worker-order permutation lives in a device buffer so output
+        # columns stay in input order at every world size.
+        if (self.embeddings.packed_forward_available()
+                and os.environ.get("DE_PACKED", "1") != "0"):
+            self.register_buffer(
+                "_dot_perm",
+                torch.tensor(self.embeddings.packed_order(), dtype=torch.int32),
+                persistent=False)
+        else:
+            self._dot_perm = None
+
     def local_cat_feature_ids(self) -> List[int]:
         if self.distributed and not self.dp_input:
             return self.embeddings.local_input_ids()
@@ -133,7 +147,15 @@ class DLRM(nn.Module):
             # id exchange overlaps the MLP GEMMs (ids carry no grad)
             handle = self.embeddings.redistribute_async(cats)
         bottom = self.bottom_mlp(numerical)
-        emb = self.embeddings(cats, output_dtype=bottom.dtype,
-                              async_handle=handle)
-        x = fused_dot_interact(emb, bottom, pad_to=self.interact_pad)
+        use_packed = self._dot_perm is not None and all(
+            isinstance(x, torch.Tensor) and x.dim() == 1 for x in cats)
+        if use_packed:
+            packed = self.embeddings.forward_packed(
+                cats, output_dtype=bottom.dtype, async_handle=handle)
+            x = dot_interact_packed(packed, bottom, self._dot_perm,
+                                    pad_to=self.interact_pad)
+        else:
+            emb = self.embeddings(cats, output_dtype=bottom.dtype,
+                                  async_handle=handle)
+            x = fused_dot_interact(emb, bottom, pad_to=self.interact_pad)
         return self.top_mlp(x)

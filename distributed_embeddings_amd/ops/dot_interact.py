@@ -39,6 +39,46 @@ def _torch_dot_interact(feats: torch.Tensor, pad_to: int) -> torch.Tensor:
     return torch.cat(parts, dim=1)
 
 
+class _DotInteractPacked(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, bottom, packed, perm, out_w):
+        ctx.save_for_backward(bottom, packed, perm)
+        return _backend.ops().dot_interact_fwd_packed(bottom, packed, perm, out_w)
+
+    @staticmethod
+    def backward(ctx, gout):
+        bottom, packed, perm = ctx.saved_tensors
+        gbottom, gpacked = _backend.ops().dot_interact_bwd_packed(
+            gout.contiguous(), bottom, packed, perm)
+        return gbottom, gpacked, None, None
+
+
+def dot_interact_packed(packed: torch.Tensor, bottom_mlp_out: torch.Tensor,
+                        perm: torch.Tensor, pad_to: int = 0) -> torch.Tensor:
+    """Interaction over feature-major packed embeddings (zero-copy input).
+
+    ``packed``: [P, B, D] — the fused-group lookup output (world==1) or the
+    mp->dp all-to-all recv buffer viewed in place (see
+    ``DistributedEmbedding.forward_packed``).  ``perm[f-1]`` maps model
+    feature f to its packed row, so output columns are in model input order
+    at every world size.
+    """
+    p, b, d = packed.shape
+    f = p + 1
+    width = f * (f - 1) // 2 + d
+    out_w = max(pad_to, width)
+    if (packed.is_cuda and packed.dtype == torch.bfloat16
+            and bottom_mlp_out.dtype == torch.bfloat16 and f <= 32
+            and d % 32 == 0):
+        return _DotInteractPacked.apply(bottom_mlp_out.contiguous(),
+                                        packed.contiguous(), perm, out_w)
+    # fallback (CPU / unsupported shapes): materialize [B, F, D] and reuse
+    # the plain path
+    feats = torch.cat([bottom_mlp_out.unsqueeze(0),
+                       packed.index_select(0, perm.long())]).transpose(0, 1)
+    return _torch_dot_interact(feats.contiguous(), out_w)
+
+
 def dot_interact(emb_outs: List[torch.Tensor], bottom_mlp_out: torch.Tensor,
                  pad_to: int = 0) -> torch.Tensor:
     """[tril(feats @ feats^T) | bottom | 0-pad], feats = [bottom] + emb_outs."""

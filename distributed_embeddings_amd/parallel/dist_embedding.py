@@ -575,10 +575,13 @@ class DistributedEmbedding(nn.Module):
                 out.append(vals.reshape(W * shape[0], *shape[1:]))
         return out
 
-    def _fused_group_lookup(self, pair_ids):
+    def _fused_group_lookup(self, pair_ids, unsplit=False):
         """Runs every local pair's lookup, one fused call per concat group.
 
         Returns per-pair 2-D outputs [rows, out_cols] in local pair order.
+        With ``unsplit=True`` (single concat group, packed fast path) the raw
+        group output [total_rows, width] is returned without the per-pair
+        split — zero extra copies.
         """
         plan = self.strategy
         groups = plan.local_concat_groups(self.rank)
@@ -614,6 +617,8 @@ class DistributedEmbedding(nn.Module):
                     emb = layer.csr_lookup(allids, splits, "sum")
                 if getattr(self, "_output_dtype", None) is not None:
                     emb = emb.to(self._output_dtype)
+                if unsplit:
+                    return emb
                 # torch.split (not manual narrow): its backward is ONE cat
                 # instead of per-slice zero-fill + accumulate.
                 parts = torch.split(emb, [n for _, _, n in metas])
@@ -689,6 +694,8 @@ class DistributedEmbedding(nn.Module):
                     out = layer.csr_lookup(allvals, allsplits, grp.combiner)
                 if getattr(self, "_output_dtype", None) is not None:
                     out = out.to(self._output_dtype)
+                if unsplit:
+                    return out
                 parts = torch.split(out, [nrows for _, nrows in metas])
                 for (j, nrows), part in zip(metas, parts):
                     outs[j] = part
@@ -851,6 +858,82 @@ class DistributedEmbedding(nn.Module):
         if key not in cache:
             cache[key] = comm.all_gather_int_vectors(list(my_pair_cols))
         return cache[key]
+
+    # ------------------------------------------------------ packed fast path
+
+    def packed_forward_available(self) -> bool:
+        """True when ``forward_packed`` applies: dp-input mode, every table in
+        ONE same-width table-parallel concat group per rank, no column
+        slicing, no dp/row groups.  (The DLRM flagship shape.)"""
+        plan = self.strategy
+        if not self.dp_input:
+            return False
+        if plan.dp_table_ids or plan.row_table_ids:
+            return False
+        if plan.sliced_out_ranges:
+            return False
+        widths = {s.width for r in range(self.world_size)
+                  for s in plan.rank_slices[r]}
+        if len(widths) != 1:
+            return False
+        return all(len(plan.rank_concat_groups[r]) == 1
+                   for r in range(self.world_size))
+
+    def packed_order(self) -> List[int]:
+        """``packed_order()[f]`` = packed row of input f (worker order).
+
+        Build this once into a device int32 buffer and hand it to
+        ``dot_interact_packed`` so interaction columns stay in model input
+        order at every world size / plan.
+        """
+        return list(self.strategy.rev_tp_order)
+
+    def forward_packed(self, inputs: Sequence[torch.Tensor],
+                       output_dtype: Optional[torch.dtype] = None,
+                       async_handle: Optional[dict] = None) -> torch.Tensor:
+        """Returns ALL lookups as one feature-major [P, b, width] tensor.
+
+        Zero-copy relative to :meth:`forward`: at world==1 this is a view of
+        the fused-group lookup output; at world>1 it is a view of the mp->dp
+        all-to-all recv buffer — the per-pair split + stack/merge copies of
+        the general path disappear.  Rows are in worker order (see
+        :meth:`packed_order`).  Requires :meth:`packed_forward_available` and
+        hotness-1 dense inputs (or 2-D with a combiner).
+        """
+        plan = self.strategy
+        W = self.world_size
+        self._output_dtype = output_dtype
+        col_in = plan.input_groups[1]
+        col_inputs = [inputs[i] for i in col_in]
+        for j, x in enumerate(col_inputs):
+            comb = plan.configs[plan.col_table_ids[plan.input_maps[1][j]]].combiner
+            if isinstance(x, Ragged) or x.dim() > 2 or (
+                    x.dim() == 2 and comb is None and x.shape[1] != 1):
+                raise ValueError("forward_packed requires hotness-1 dense "
+                                 "inputs (or 2-D with a combiner)")
+        b = col_inputs[0].shape[0]
+        self._validate_batch(b)
+        if async_handle is not None:
+            if async_handle["sig"] != tuple(tuple(x.shape) for x in col_inputs):
+                raise ValueError("async_handle does not match these inputs")
+            async_handle["work"].wait()
+            recv_ids = async_handle["recv"].to(async_handle["device"])
+            pair_ids = self._carve_dense_recv(
+                recv_ids, async_handle["my_sizes"], col_inputs)
+        else:
+            pair_ids = self._dp_to_mp_dense(col_inputs)
+        group_out = self._fused_group_lookup(pair_ids, unsplit=True)
+        D = group_out.shape[-1]
+        P_local = len(pair_ids)
+        if W == 1:
+            return group_out.view(P_local, b, D)
+        send = group_out.view(P_local, W, b * D).transpose(0, 1).reshape(-1)
+        all_cols = self._exchange_pair_cols([D] * P_local)
+        out_splits = [b * sum(all_cols[k]) for k in range(W)]
+        in_splits = [b * D * P_local] * W
+        recv = comm.all_to_all_single(send, out_splits, in_splits)
+        P_total = sum(len(c) for c in all_cols)
+        return recv.view(P_total, b, D)
 
     # --------------------------------------------------------- row slice path
 

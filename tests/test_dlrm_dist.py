@@ -198,3 +198,47 @@ def test_synthetic_model_world2_matches_world1():
         got = r2[rank]["out"]
         assert torch.allclose(got, ref, atol=1e-4), \
             f"rank{rank} err {(got - ref).abs().max()}"
+
+
+def _packed_vs_legacy_worker(rank, world):
+    """Packed interaction path == legacy stack path, world 2 (gloo)."""
+    import os
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd.models.dlrm import DLRM
+    sizes = [50, 60, 70, 40]
+    torch.manual_seed(0)
+    m = DLRM(sizes, embedding_dim=32, bottom_mlp_dims=(64, 32),
+             top_mlp_dims=(64, 1), num_numerical=4, strategy="basic")
+    de.broadcast_parameters(m)
+    assert m._dot_perm is not None
+    os.environ["DE_PACKED"] = "0"
+    try:
+        m2 = DLRM(sizes, embedding_dim=32, bottom_mlp_dims=(64, 32),
+                  top_mlp_dims=(64, 1), num_numerical=4, strategy="basic")
+    finally:
+        del os.environ["DE_PACKED"]
+    assert m2._dot_perm is None
+    m2.load_state_dict(m.state_dict())
+    m2.embeddings.set_weights(m.embeddings.get_weights(all_ranks=True))
+
+    g = torch.Generator().manual_seed(7)
+    B = world * 4
+    num = torch.rand(B, 4, generator=g)
+    cats = [torch.randint(0, s, (B,), generator=g) for s in sizes]
+    sl = slice(rank * 4, (rank + 1) * 4)
+    o1 = m(num[sl], [c[sl] for c in cats])
+    o2 = m2(num[sl], [c[sl] for c in cats])
+    err = float((o1 - o2).abs().max())
+    # grads flow through the packed view + a2a
+    o1.square().sum().backward()
+    return err
+
+
+def test_dlrm_packed_matches_legacy_world2():
+    errs = run_distributed(_packed_vs_legacy_worker, world=2)
+    assert max(errs) < 1e-5, errs
+
+
+def test_dlrm_packed_matches_legacy_world1():
+    err = _packed_vs_legacy_worker(0, 1)
+    assert err < 1e-6, err
