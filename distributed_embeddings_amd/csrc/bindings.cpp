@@ -77,7 +77,8 @@ void sort_ids_dispatch(torch::Tensor values, int64_t vocab, int64_t nnz,
 }
 
 torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
-                                 torch::Tensor row_splits, bool mean) {
+                                 torch::Tensor row_splits, bool mean,
+                                 bool out_bf16) {
   CHECK_CUDA(params); CHECK_CUDA(values); CHECK_CUDA(row_splits);
   CHECK_CONTIG(params); CHECK_CONTIG(values); CHECK_CONTIG(row_splits);
   TORCH_CHECK(params.dtype() == torch::kFloat32 ||
@@ -88,9 +89,13 @@ torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
   const int64_t num_rows = row_splits.numel() - 1;
   const int64_t vocab = params.size(0);
   const int width = (int)params.size(1);
-  // output accumulates in fp32 regardless of the table storage dtype
+  // accumulation is fp32 regardless of table/output storage dtype; bf16 out
+  // stores RNE-rounded results directly (no separate cast kernel).  The
+  // long-segment split needs fp32 atomics, so bf16-out launches reduce
+  // every row in-register instead (fine for bounded forward hotness).
   auto out = torch::empty({num_rows, width},
-                          params.options().dtype(torch::kFloat32));
+                          params.options().dtype(
+                              out_bf16 ? torch::kBFloat16 : torch::kFloat32));
   if (num_rows > 0) {
     const int64_t nnz_in = values.numel();
     auto long_rows = torch::empty({num_rows}, values.options());
@@ -101,7 +106,7 @@ torch::Tensor csr_lookup_forward(torch::Tensor params, torch::Tensor values,
                               params.dtype() == torch::kBFloat16,
                               values.data_ptr<int64_t>(),
                               row_splits.data_ptr<int64_t>(), nullptr,
-                              out.data_ptr<float>(), num_rows, nnz_in,
+                              out.data_ptr(), out_bf16, num_rows, nnz_in,
                               vocab, width, mean,
                               long_rows.data_ptr<int64_t>(),
                               long_count.data_ptr<int32_t>(),
@@ -126,14 +131,16 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
                                                int64_t vocab, bool mean) {
   CHECK_CUDA(grad_out); CHECK_CUDA(values); CHECK_CUDA(row_splits);
   CHECK_CONTIG(grad_out); CHECK_CONTIG(values); CHECK_CONTIG(row_splits);
-  TORCH_CHECK(grad_out.dtype() == torch::kFloat32, "grad_out must be fp32");
+  TORCH_CHECK(grad_out.dtype() == torch::kFloat32 ||
+                  grad_out.dtype() == torch::kBFloat16,
+              "grad_out must be fp32 or bf16");
   const int64_t nnz = values.numel();
   const int64_t num_rows = row_splits.numel() - 1;
   const int width = (int)grad_out.size(1);
   auto stream = current_stream();
   auto i64 = values.options();
   auto i32 = values.options().dtype(torch::kInt32);
-  auto f32 = grad_out.options();
+  auto f32 = grad_out.options().dtype(torch::kFloat32);
 
   if (nnz == 0) {
     return {torch::empty({0}, i64), torch::empty({0, width}, f32)};
@@ -209,10 +216,11 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
     auto long_count = torch::empty({1}, i32);
     auto work_items = torch::empty({nnz / 64 + 64}, i64);
     auto n_work = torch::empty({1}, i32);
-    launch_csr_lookup_forward(grad_out.data_ptr(), /*params_bf16=*/false,
+    launch_csr_lookup_forward(grad_out.data_ptr(),
+                              grad_out.dtype() == torch::kBFloat16,
                               srow.data_ptr<int64_t>(),
                               seg_tmp.data_ptr<int64_t>(), sw_ptr,
-                              unique_grad.data_ptr<float>(), nu, nnz,
+                              unique_grad.data_ptr<float>(), false, nu, nnz,
                               grad_out.size(0), width, /*mean=*/false,
                               long_rows.data_ptr<int64_t>(),
                               long_count.data_ptr<int32_t>(),
@@ -310,7 +318,8 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
   CHECK_CONTIG(grad_out);
   TORCH_CHECK(weight.dtype() == torch::kFloat32 ||
               weight.dtype() == torch::kBFloat16);
-  TORCH_CHECK(grad_out.dtype() == torch::kFloat32 &&
+  TORCH_CHECK((grad_out.dtype() == torch::kFloat32 ||
+               grad_out.dtype() == torch::kBFloat16) &&
               lr.dtype() == torch::kFloat32);
   const bool wbf16 = weight.dtype() == torch::kBFloat16;
   const int64_t num_rows = row_splits.numel() - 1;
@@ -321,7 +330,7 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
   auto stream = current_stream();
   auto i64 = values.options();
   auto i32 = values.options().dtype(torch::kInt32);
-  auto f32 = grad_out.options();
+  auto f32 = grad_out.options().dtype(torch::kFloat32);
 
   auto row_ids = torch::empty({nnz}, i32);
   torch::Tensor w;
@@ -393,7 +402,8 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
                                  (float)eps, sorted_ids.data_ptr<int64_t>(),
                                  seg_tmp.data_ptr<int64_t>(),
                                  srow.data_ptr<int64_t>(), sw_ptr,
-                                 grad_out.data_ptr<float>(),
+                                 grad_out.data_ptr(),
+                                 grad_out.dtype() == torch::kBFloat16,
                                  lr.data_ptr<float>(),
                                  nu_dev.data_ptr<int32_t>(), nnz, width,
                                  long_rows.data_ptr<int64_t>(),
@@ -482,7 +492,10 @@ torch::Tensor dot_interact_bwd(torch::Tensor gout, torch::Tensor feats) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("csr_lookup_forward", &csr_lookup_forward,
-        "CSR segmented gather-reduce forward (gfx950)");
+        "CSR segmented gather-reduce forward (gfx950)",
+        pybind11::arg("params"), pybind11::arg("values"),
+        pybind11::arg("row_splits"), pybind11::arg("mean"),
+        pybind11::arg("out_bf16") = false);
   m.def("csr_lookup_backward", &csr_lookup_backward,
         "sparse backward: sort + unique + segmented sum (gfx950)");
   m.def("row_to_split", &row_to_split, "COO rows -> CSR splits (gfx950)");

@@ -28,6 +28,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
+#include <cstdint>
 #include <cstring>
 #include <type_traits>
 
@@ -78,6 +79,29 @@ __device__ __forceinline__ void pt_load2(const bf16_t* p, float* o) {
   c.u = ((unsigned int)(unsigned short)v.x) << 16; o[0] = c.f;
   c.u = ((unsigned int)(unsigned short)v.y) << 16; o[1] = c.f;
 }
+// vectorized stores (RNE for bf16): 4/2 contiguous elements
+__device__ __forceinline__ void pt_store4(float* p, const float* v) {
+  *reinterpret_cast<float4*>(p) = make_float4(v[0], v[1], v[2], v[3]);
+}
+__device__ __forceinline__ void pt_store4(bf16_t* p, const float* v) {
+  short4 s;
+  __hip_bfloat16 h0(v[0]), h1(v[1]), h2(v[2]), h3(v[3]);
+  s.x = *reinterpret_cast<short*>(&h0);
+  s.y = *reinterpret_cast<short*>(&h1);
+  s.z = *reinterpret_cast<short*>(&h2);
+  s.w = *reinterpret_cast<short*>(&h3);
+  *reinterpret_cast<short4*>(p) = s;
+}
+__device__ __forceinline__ void pt_store2(float* p, const float* v) {
+  *reinterpret_cast<float2*>(p) = make_float2(v[0], v[1]);
+}
+__device__ __forceinline__ void pt_store2(bf16_t* p, const float* v) {
+  short2 s;
+  __hip_bfloat16 h0(v[0]), h1(v[1]);
+  s.x = *reinterpret_cast<short*>(&h0);
+  s.y = *reinterpret_cast<short*>(&h1);
+  *reinterpret_cast<short2*>(p) = s;
+}
 
 static inline int next_pow2(int v) {
   int p = 1;
@@ -104,12 +128,15 @@ static inline int64_t cdiv64(int64_t a, int64_t b) { return (a + b - 1) / b; }
 #define LONG_T 128
 
 // Narrow kernel A: width <= 64.  TILE = pow2 >= width; 64/TILE rows per wave.
-template <int TILE, bool MEAN, bool HAS_W, typename PT>
+// OT: output storage type (float, or bf16 when the consumer wants bf16
+// directly — saves the separate cast kernel + half the store traffic; the
+// long-row path requires fp32 atomics so bf16-out launches disable it).
+template <int TILE, bool MEAN, bool HAS_W, typename PT, typename OT>
 __global__ void csr_fwd_narrow(const PT* __restrict__ params,
                                const int64_t* __restrict__ values,
                                const int64_t* __restrict__ splits,
                                const float* __restrict__ per_id_w,
-                               float* __restrict__ out, int64_t num_rows,
+                               OT* __restrict__ out, int64_t num_rows,
                                int64_t vocab, int width, int64_t long_thresh,
                                int64_t* __restrict__ long_rows,
                                int32_t* __restrict__ long_count) {
@@ -125,7 +152,7 @@ __global__ void csr_fwd_narrow(const PT* __restrict__ params,
     if (row >= num_rows || tl >= width) continue;
     const int64_t s = splits[row], e = splits[row + 1];
     if (e - s > long_thresh) {
-      out[row * width + tl] = 0.f;
+      pt_store(&out[row * width + tl], 0.f);
       if (tl == 0) long_rows[atomicAdd(long_count, 1)] = row;
       continue;
     }
@@ -153,17 +180,17 @@ __global__ void csr_fwd_narrow(const PT* __restrict__ params,
       acc += w * pt_load(&params[idx * width + tl]);
     }
     if (MEAN && e > s) acc /= (float)(e - s);
-    out[row * width + tl] = acc;
+    pt_store(&out[row * width + tl], acc);
   }
 }
 
 // Wide kernel A: width > 64.  One wave per row; VEC elements per lane.
-template <int VEC, bool MEAN, bool HAS_W, typename PT>
+template <int VEC, bool MEAN, bool HAS_W, typename PT, typename OT>
 __global__ void csr_fwd_wide(const PT* __restrict__ params,
                              const int64_t* __restrict__ values,
                              const int64_t* __restrict__ splits,
                              const float* __restrict__ per_id_w,
-                             float* __restrict__ out, int64_t num_rows,
+                             OT* __restrict__ out, int64_t num_rows,
                              int64_t vocab, int width, int64_t long_thresh,
                              int64_t* __restrict__ long_rows,
                              int32_t* __restrict__ long_count,
@@ -183,7 +210,8 @@ __global__ void csr_fwd_wide(const PT* __restrict__ params,
     if (row >= num_rows) continue;
     const int64_t s = splits[row], e = splits[row + 1];
     if (e - s > long_thresh) {
-      for (int c = tl; c < width; c += tile_w) out[row * width + c] = 0.f;
+      for (int c = tl; c < width; c += tile_w)
+        pt_store(&out[row * width + c], 0.f);
       if (tl == 0) long_rows[atomicAdd(long_count, 1)] = row;
       continue;
     }
@@ -221,16 +249,17 @@ __global__ void csr_fwd_wide(const PT* __restrict__ params,
           if (col0 < width) acc[0] += w * pt_load(&rowp[0]);
         }
       }
-      float* outp = out + row * (int64_t)width + col0;
+      OT* outp = out + row * (int64_t)width + col0;
+#pragma unroll
+      for (int v = 0; v < VEC; ++v) acc[v] *= inv;
       if (VEC == 4 && col0 + 4 <= width) {
-        *reinterpret_cast<float4*>(outp) = make_float4(
-            acc[0] * inv, acc[1] * inv, acc[2] * inv, acc[3] * inv);
+        pt_store4(outp, acc);
       } else if (VEC == 2 && col0 + 2 <= width) {
-        *reinterpret_cast<float2*>(outp) = make_float2(acc[0] * inv, acc[1] * inv);
+        pt_store2(outp, acc);
       } else {
 #pragma unroll
         for (int v = 0; v < VEC; ++v)
-          if (col0 + v < width) outp[v] = acc[v] * inv;
+          if (col0 + v < width) pt_store(&outp[v], acc[v]);
       }
     }
   }
@@ -385,10 +414,10 @@ static int pick_grid(int64_t work_items, int block_waves) {
   return (int)blocks;
 }
 
-template <int TILE, int VEC, typename PT>
+template <int TILE, int VEC, typename PT, typename OT>
 static void launch_csr_pair(const PT* params, const int64_t* values,
                             const int64_t* splits, const float* per_id_w,
-                            float* out, int64_t num_rows, int64_t nnz,
+                            OT* out, int64_t num_rows, int64_t nnz,
                             int64_t vocab, int width, bool mean,
                             int64_t* long_rows, int32_t* long_count,
                             int64_t* work_items, int32_t* n_work,
@@ -416,29 +445,39 @@ static void launch_csr_pair(const PT* params, const int64_t* values,
     if (long_thresh < LONG_T) long_thresh = LONG_T;
     if (long_thresh > 8192) long_thresh = 8192;
   }
+  constexpr bool kFloatOut = std::is_same<OT, float>::value;
+  if (!kFloatOut) {
+    // bf16 out: the long-row combine needs fp32 atomics, so disable the
+    // split — every row reduces in-register in kernel A (callers request
+    // bf16 out only on bounded-hotness forwards)
+    long_thresh = INT64_MAX;
+  }
   hipMemsetAsync(long_count, 0, sizeof(int32_t), stream);
   hipMemsetAsync(n_work, 0, sizeof(int32_t), stream);
 #define LA(MEAN, HASW)                                                         \
   do {                                                                         \
     if constexpr (TILE > 0)                                                    \
       hipLaunchKernelGGL(                                                      \
-          (csr_fwd_narrow<(TILE > 0 ? TILE : 1), MEAN, HASW, PT>), grid,       \
+          (csr_fwd_narrow<(TILE > 0 ? TILE : 1), MEAN, HASW, PT, OT>), grid,   \
           dim3(block), 0, stream, params, values, splits, per_id_w, out,       \
           num_rows, vocab, width, long_thresh, long_rows, long_count);         \
     else                                                                       \
-      hipLaunchKernelGGL((csr_fwd_wide<(VEC > 0 ? VEC : 1), MEAN, HASW, PT>),  \
-                         grid, dim3(block), 0, stream, params, values, splits, \
-                         per_id_w, out, num_rows, vocab, width, long_thresh,   \
-                         long_rows, long_count, tile_w);                       \
+      hipLaunchKernelGGL(                                                      \
+          (csr_fwd_wide<(VEC > 0 ? VEC : 1), MEAN, HASW, PT, OT>),             \
+          grid, dim3(block), 0, stream, params, values, splits,                \
+          per_id_w, out, num_rows, vocab, width, long_thresh,                  \
+          long_rows, long_count, tile_w);                                      \
   } while (0)
 #define LB(MEAN, HASW)                                                         \
   do {                                                                         \
-    hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,    \
-                       long_rows, long_count, splits, work_items, n_work);     \
-    hipLaunchKernelGGL((csr_fwd_long<TILE, VEC, MEAN, HASW, PT>), dim3(2048),  \
-                       dim3(block), 0, stream, params, values, splits,         \
-                       per_id_w, out, vocab, width, long_rows, work_items,     \
-                       n_work, tile_w);                                        \
+    if constexpr (kFloatOut) {                                                 \
+      hipLaunchKernelGGL(expand_long_work, dim3(512), dim3(block), 0, stream,  \
+                         long_rows, long_count, splits, work_items, n_work);   \
+      hipLaunchKernelGGL((csr_fwd_long<TILE, VEC, MEAN, HASW, PT>),            \
+                         dim3(2048), dim3(block), 0, stream, params, values,   \
+                         splits, per_id_w, (float*)out, vocab, width,          \
+                         long_rows, work_items, n_work, tile_w);               \
+    }                                                                          \
   } while (0)
   if (mean) {
     if (per_id_w) { LA(true, true); LB(true, true); }
@@ -451,10 +490,10 @@ static void launch_csr_pair(const PT* params, const int64_t* values,
 #undef LB
 }
 
-template <typename PT>
+template <typename PT, typename OT>
 static void launch_csr_lookup_forward_t(const PT* params, const int64_t* values,
                                         const int64_t* splits,
-                                        const float* per_id_w, float* out,
+                                        const float* per_id_w, OT* out,
                                         int64_t num_rows, int64_t nnz,
                                         int64_t vocab, int width, bool mean,
                                         int64_t* long_rows, int32_t* long_count,
@@ -464,40 +503,41 @@ static void launch_csr_lookup_forward_t(const PT* params, const int64_t* values,
              width, mean, long_rows, long_count, work_items, n_work, stream
   if (width <= 64) {
     switch (next_pow2(width)) {
-      case 1: launch_csr_pair<1, 0, PT>(ARGS); break;
-      case 2: launch_csr_pair<2, 0, PT>(ARGS); break;
-      case 4: launch_csr_pair<4, 0, PT>(ARGS); break;
-      case 8: launch_csr_pair<8, 0, PT>(ARGS); break;
-      case 16: launch_csr_pair<16, 0, PT>(ARGS); break;
-      case 32: launch_csr_pair<32, 0, PT>(ARGS); break;
-      default: launch_csr_pair<64, 0, PT>(ARGS); break;
+      case 1: launch_csr_pair<1, 0, PT, OT>(ARGS); break;
+      case 2: launch_csr_pair<2, 0, PT, OT>(ARGS); break;
+      case 4: launch_csr_pair<4, 0, PT, OT>(ARGS); break;
+      case 8: launch_csr_pair<8, 0, PT, OT>(ARGS); break;
+      case 16: launch_csr_pair<16, 0, PT, OT>(ARGS); break;
+      case 32: launch_csr_pair<32, 0, PT, OT>(ARGS); break;
+      default: launch_csr_pair<64, 0, PT, OT>(ARGS); break;
     }
   } else if (width % 4 == 0) {
-    launch_csr_pair<0, 4, PT>(ARGS);
+    launch_csr_pair<0, 4, PT, OT>(ARGS);
   } else if (width % 2 == 0) {
-    launch_csr_pair<0, 2, PT>(ARGS);
+    launch_csr_pair<0, 2, PT, OT>(ARGS);
   } else {
-    launch_csr_pair<0, 1, PT>(ARGS);
+    launch_csr_pair<0, 1, PT, OT>(ARGS);
   }
 #undef ARGS
 }
 
 void launch_csr_lookup_forward(const void* params, bool params_bf16,
                                const int64_t* values, const int64_t* splits,
-                               const float* per_id_w, float* out,
+                               const float* per_id_w, void* out, bool out_bf16,
                                int64_t num_rows, int64_t nnz, int64_t vocab,
                                int width, bool mean, int64_t* long_rows,
                                int32_t* long_count, int64_t* work_items,
                                int32_t* n_work, hipStream_t stream) {
+#define FWD(PTT, OTT)                                                          \
+  launch_csr_lookup_forward_t<PTT, OTT>(                                       \
+      (const PTT*)params, values, splits, per_id_w, (OTT*)out, num_rows, nnz, \
+      vocab, width, mean, long_rows, long_count, work_items, n_work, stream)
   if (params_bf16) {
-    launch_csr_lookup_forward_t<bf16_t>(
-        (const bf16_t*)params, values, splits, per_id_w, out, num_rows, nnz,
-        vocab, width, mean, long_rows, long_count, work_items, n_work, stream);
+    if (out_bf16) FWD(bf16_t, bf16_t); else FWD(bf16_t, float);
   } else {
-    launch_csr_lookup_forward_t<float>(
-        (const float*)params, values, splits, per_id_w, out, num_rows, nnz,
-        vocab, width, mean, long_rows, long_count, work_items, n_work, stream);
+    if (out_bf16) FWD(float, bf16_t); else FWD(float, float);
   }
+#undef FWD
 }
 
 // ---------------------------------------------------------------------------
@@ -768,10 +808,10 @@ void launch_pad_seg_offsets(int64_t* seg, int64_t n, const int32_t* num_unique,
 // Segment-sum helper: reduces grad_out rows of segment [ks, ke) into
 // acc[V] per lane; NARROW (TILE>0): NSUB=64/TILE sub-tiles split k with a
 // cross-sub shuffle fold afterwards; WIDE: V floats per lane, k unrolled x2.
-template <int TILE, int VEC, bool HAS_W>
+template <int TILE, int VEC, bool HAS_W, typename GT>
 __device__ __forceinline__ void seg_grad_reduce(
     const int64_t* __restrict__ srow, const float* __restrict__ sw,
-    const float* __restrict__ grad_out, int width, int64_t ks, int64_t ke,
+    const GT* __restrict__ grad_out, int width, int64_t ks, int64_t ke,
     int lane, float* acc /* size V */) {
   if constexpr (TILE > 0) {
     constexpr int T = TILE > 0 ? TILE : 1;
@@ -785,11 +825,12 @@ __device__ __forceinline__ void seg_grad_reduce(
         const int64_t r0 = srow[k], r1 = srow[k + NSUB];
         const float w0 = HAS_W ? sw[k] : 1.f;
         const float w1 = HAS_W ? sw[k + NSUB] : 1.f;
-        a += w0 * grad_out[r0 * (int64_t)width + tl] +
-             w1 * grad_out[r1 * (int64_t)width + tl];
+        a += w0 * pt_load(&grad_out[r0 * (int64_t)width + tl]) +
+             w1 * pt_load(&grad_out[r1 * (int64_t)width + tl]);
       }
       if (k < ke) {
-        a += (HAS_W ? sw[k] : 1.f) * grad_out[srow[k] * (int64_t)width + tl];
+        a += (HAS_W ? sw[k] : 1.f) *
+             pt_load(&grad_out[srow[k] * (int64_t)width + tl]);
       }
     }
 #pragma unroll
@@ -805,14 +846,15 @@ __device__ __forceinline__ void seg_grad_reduce(
 // Short segments: one wave per segment, direct (non-atomic) update.
 // Grid-strides only over the REAL segment count (*nu_ptr), not the padded
 // nnz-sized buffer.
-template <int TILE, int VEC, bool HAS_W, bool ADAGRAD, typename PT>
+template <int TILE, int VEC, bool HAS_W, bool ADAGRAD, typename PT,
+          typename GT>
 __global__ void sorted_opt_update(PT* __restrict__ weight,
                                   float* __restrict__ state, float eps,
                                   const int64_t* __restrict__ sorted_ids,
                                   const int64_t* __restrict__ seg,
                                   const int64_t* __restrict__ srow,
                                   const float* __restrict__ sw,
-                                  const float* __restrict__ grad_out,
+                                  const GT* __restrict__ grad_out,
                                   const float* __restrict__ lr_ptr,
                                   const int32_t* __restrict__ nu_ptr,
                                   int64_t long_thresh, int width,
@@ -864,15 +906,16 @@ __global__ void sorted_opt_update(PT* __restrict__ weight,
         const int col0 = cbase + tl_w * V;
         for (int64_t k = s; k < e; ++k) {
           const float w = HAS_W ? sw[k] : 1.f;
-          const float* gp = grad_out + srow[k] * (int64_t)width + col0;
+          const GT* gp = grad_out + srow[k] * (int64_t)width + col0;
           if (V == 4 && col0 + 4 <= width) {
-            const float4 g4 = *reinterpret_cast<const float4*>(gp);
-            acc[0] += w * g4.x; acc[1] += w * g4.y;
-            acc[2] += w * g4.z; acc[3] += w * g4.w;
+            float g4[4];
+            pt_load4(gp, g4);
+            acc[0] += w * g4[0]; acc[1] += w * g4[1];
+            acc[2] += w * g4[2]; acc[3] += w * g4[3];
           } else {
 #pragma unroll
             for (int v = 0; v < V; ++v)
-              if (col0 + v < width) acc[v] += w * gp[v];
+              if (col0 + v < width) acc[v] += w * pt_load(&gp[v]);
           }
         }
         if (ADAGRAD) {
@@ -902,13 +945,13 @@ __global__ void sorted_opt_update(PT* __restrict__ weight,
 
 // Long segments, SGD: chunk partials atomically into the weight (linear).
 // Long segments, Adagrad: chunk partials into scratch rows, then finalize.
-template <int TILE, int VEC, bool HAS_W, bool TO_SCRATCH>
+template <int TILE, int VEC, bool HAS_W, bool TO_SCRATCH, typename GT>
 __global__ void sorted_opt_long(float* __restrict__ target,  // weight or scratch
                                 const int64_t* __restrict__ sorted_ids,
                                 const int64_t* __restrict__ seg,
                                 const int64_t* __restrict__ srow,
                                 const float* __restrict__ sw,
-                                const float* __restrict__ grad_out,
+                                const GT* __restrict__ grad_out,
                                 const float* __restrict__ lr_ptr, int width,
                                 const int64_t* __restrict__ long_rows,
                                 const int64_t* __restrict__ work_items,
@@ -954,15 +997,16 @@ __global__ void sorted_opt_long(float* __restrict__ target,  // weight or scratc
         const int col0 = cbase + tl_w * V;
         for (int64_t k = ks; k < ke; ++k) {
           const float w = HAS_W ? sw[k] : 1.f;
-          const float* gp = grad_out + srow[k] * (int64_t)width + col0;
+          const GT* gp = grad_out + srow[k] * (int64_t)width + col0;
           if (V == 4 && col0 + 4 <= width) {
-            const float4 g4 = *reinterpret_cast<const float4*>(gp);
-            acc[0] += w * g4.x; acc[1] += w * g4.y;
-            acc[2] += w * g4.z; acc[3] += w * g4.w;
+            float g4[4];
+            pt_load4(gp, g4);
+            acc[0] += w * g4[0]; acc[1] += w * g4[1];
+            acc[2] += w * g4[2]; acc[3] += w * g4[3];
           } else {
 #pragma unroll
             for (int v = 0; v < V; ++v)
-              if (col0 + v < width) acc[v] += w * gp[v];
+              if (col0 + v < width) acc[v] += w * pt_load(&gp[v]);
           }
         }
         float* tp = target + trow * (int64_t)width + col0;
@@ -1006,11 +1050,11 @@ __global__ void sorted_long_finalize(
 
 
 
-template <int TILE, int VEC, typename PT>
+template <int TILE, int VEC, typename PT, typename GT>
 static void launch_sorted_opt_pair(PT* weight, float* state, float eps,
                                    const int64_t* sorted_ids,
                                    const int64_t* seg, const int64_t* srow,
-                                   const float* sw, const float* grad_out,
+                                   const float* sw, const GT* grad_out,
                                    const float* lr, const int32_t* nu_ptr,
                                    int64_t max_segs, int width,
                                    int64_t* long_rows, int32_t* long_count,
@@ -1034,17 +1078,17 @@ static void launch_sorted_opt_pair(PT* weight, float* state, float eps,
   constexpr bool PT_F32 = std::is_same<PT, float>::value;
   const bool use_scratch = adagrad || !PT_F32;
 #define SU(HASW, ADA)                                                          \
-  hipLaunchKernelGGL((sorted_opt_update<TILE, VEC, HASW, ADA, PT>),            \
+  hipLaunchKernelGGL((sorted_opt_update<TILE, VEC, HASW, ADA, PT, GT>),        \
                      dim3(grid), dim3(block), 0, stream, weight, state, eps,   \
                      sorted_ids, seg, srow, sw, grad_out, lr, nu_ptr,          \
                      (int64_t)LONG_T, width, long_rows, long_count, tile_w)
 #define SL_SCRATCH(HASW)                                                       \
-  hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, true>), dim3(2048),     \
+  hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, true, GT>), dim3(2048), \
                      dim3(block), 0, stream, long_scratch, sorted_ids, seg,    \
                      srow, sw, grad_out, lr, width, long_rows, work_items,     \
                      n_work, tile_w)
 #define SL_DIRECT(HASW)                                                        \
-  hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, false>), dim3(2048),    \
+  hipLaunchKernelGGL((sorted_opt_long<TILE, VEC, HASW, false, GT>), dim3(2048),\
                      dim3(block), 0, stream, (float*)weight, sorted_ids, seg,  \
                      srow, sw, grad_out, lr, width, long_rows, work_items,     \
                      n_work, tile_w)
@@ -1071,11 +1115,11 @@ static void launch_sorted_opt_pair(PT* weight, float* state, float eps,
 #undef FIN
 }
 
-template <typename PT>
+template <typename PT, typename GT>
 static void launch_sorted_optimizer_update_t(
     PT* weight, float* state, float eps, const int64_t* sorted_ids,
     const int64_t* seg, const int64_t* srow, const float* sw,
-    const float* grad_out, const float* lr, const int32_t* nu_ptr,
+    const GT* grad_out, const float* lr, const int32_t* nu_ptr,
     int64_t max_segs, int width, int64_t* long_rows, int32_t* long_count,
     int64_t* work_items, int32_t* n_work, float* long_scratch, bool adagrad,
     hipStream_t stream) {
@@ -1084,20 +1128,20 @@ static void launch_sorted_optimizer_update_t(
              n_work, long_scratch, adagrad, stream
   if (width <= 64) {
     switch (next_pow2(width)) {
-      case 1: launch_sorted_opt_pair<1, 0, PT>(ARGS); break;
-      case 2: launch_sorted_opt_pair<2, 0, PT>(ARGS); break;
-      case 4: launch_sorted_opt_pair<4, 0, PT>(ARGS); break;
-      case 8: launch_sorted_opt_pair<8, 0, PT>(ARGS); break;
-      case 16: launch_sorted_opt_pair<16, 0, PT>(ARGS); break;
-      case 32: launch_sorted_opt_pair<32, 0, PT>(ARGS); break;
-      default: launch_sorted_opt_pair<64, 0, PT>(ARGS); break;
+      case 1: launch_sorted_opt_pair<1, 0, PT, GT>(ARGS); break;
+      case 2: launch_sorted_opt_pair<2, 0, PT, GT>(ARGS); break;
+      case 4: launch_sorted_opt_pair<4, 0, PT, GT>(ARGS); break;
+      case 8: launch_sorted_opt_pair<8, 0, PT, GT>(ARGS); break;
+      case 16: launch_sorted_opt_pair<16, 0, PT, GT>(ARGS); break;
+      case 32: launch_sorted_opt_pair<32, 0, PT, GT>(ARGS); break;
+      default: launch_sorted_opt_pair<64, 0, PT, GT>(ARGS); break;
     }
   } else if (width % 4 == 0) {
-    launch_sorted_opt_pair<0, 4, PT>(ARGS);
+    launch_sorted_opt_pair<0, 4, PT, GT>(ARGS);
   } else if (width % 2 == 0) {
-    launch_sorted_opt_pair<0, 2, PT>(ARGS);
+    launch_sorted_opt_pair<0, 2, PT, GT>(ARGS);
   } else {
-    launch_sorted_opt_pair<0, 1, PT>(ARGS);
+    launch_sorted_opt_pair<0, 1, PT, GT>(ARGS);
   }
 #undef ARGS
 }
@@ -1106,7 +1150,8 @@ void launch_sorted_optimizer_update(void* weight, bool weight_bf16,
                                     float* state, float eps,
                                     const int64_t* sorted_ids,
                                     const int64_t* seg, const int64_t* srow,
-                                    const float* sw, const float* grad_out,
+                                    const float* sw, const void* grad_out,
+                                    bool grad_bf16,
                                     const float* lr, const int32_t* nu_ptr,
                                     int64_t max_segs, int width,
                                     int64_t* long_rows, int32_t* long_count,
@@ -1119,17 +1164,17 @@ void launch_sorted_optimizer_update(void* weight, bool weight_bf16,
     hipMemsetAsync(long_scratch, 0,
                    sizeof(float) * scratch_rows * (int64_t)width, stream);
   }
+#define SOU(PTT, GTT)                                                          \
+  launch_sorted_optimizer_update_t<PTT, GTT>(                                  \
+      (PTT*)weight, state, eps, sorted_ids, seg, srow, sw,                     \
+      (const GTT*)grad_out, lr, nu_ptr, max_segs, width, long_rows,            \
+      long_count, work_items, n_work, long_scratch, adagrad, stream)
   if (weight_bf16) {
-    launch_sorted_optimizer_update_t<bf16_t>(
-        (bf16_t*)weight, state, eps, sorted_ids, seg, srow, sw, grad_out, lr,
-        nu_ptr, max_segs, width, long_rows, long_count, work_items, n_work,
-        long_scratch, adagrad, stream);
+    if (grad_bf16) SOU(bf16_t, bf16_t); else SOU(bf16_t, float);
   } else {
-    launch_sorted_optimizer_update_t<float>(
-        (float*)weight, state, eps, sorted_ids, seg, srow, sw, grad_out, lr,
-        nu_ptr, max_segs, width, long_rows, long_count, work_items, n_work,
-        long_scratch, adagrad, stream);
+    if (grad_bf16) SOU(float, bf16_t); else SOU(float, float);
   }
+#undef SOU
 }
 
 // ---------------------------------------------------------------------------

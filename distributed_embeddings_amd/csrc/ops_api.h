@@ -9,7 +9,7 @@
 
 void launch_csr_lookup_forward(const void* params, bool params_bf16,
                                const int64_t* values, const int64_t* splits,
-                               const float* per_id_w, float* out,
+                               const float* per_id_w, void* out, bool out_bf16,
                                int64_t num_rows, int64_t nnz, int64_t vocab,
                                int width, bool mean, int64_t* long_rows,
                                int32_t* long_count, int64_t* work_items,
@@ -67,7 +67,8 @@ void launch_sorted_optimizer_update(void* weight, bool weight_bf16,
                                     float* state, float eps,
                                     const int64_t* sorted_ids,
                                     const int64_t* seg, const int64_t* srow,
-                                    const float* sw, const float* grad_out,
+                                    const float* sw, const void* grad_out,
+                                    bool grad_bf16,
                                     const float* lr, const int32_t* nu_ptr,
                                     int64_t max_segs, int width,
                                     int64_t* long_rows, int32_t* long_count,

@@ -134,14 +134,20 @@ class Embedding(nn.Module):
             self._fused_lr.fill_(float(lr))
 
     def csr_lookup(self, values: torch.Tensor, row_splits: torch.Tensor,
-                   combiner: str) -> torch.Tensor:
-        """CSR lookup through this layer (fused-SGD aware)."""
+                   combiner: str, out_dtype=None) -> torch.Tensor:
+        """CSR lookup through this layer (fused-SGD aware).
+
+        ``out_dtype=torch.bfloat16`` makes the HIP kernel store bf16 directly
+        (fp32 accumulation; backward consumes bf16 grads natively) — no
+        separate cast kernel on the hot path."""
         if getattr(self, "_fused_lr", None) is not None and self.training:
             return csr_lookup_fused_optimizer(
                 self.weight, values, row_splits, combiner, self._fused_lr,
                 self._fused_state, self._fused_method == "adagrad",
-                self._fused_eps)
-        return embedding_lookup(self.weight, Ragged(values, row_splits), combiner)
+                self._fused_eps, out_dtype)
+        from ..ops.embedding_lookup import _CsrLookup
+        return _CsrLookup.apply(self.weight, values, row_splits, combiner,
+                                out_dtype)
 
     def get_config(self) -> dict:
         """Planner-facing config (reference uses keras ``get_config()``)."""

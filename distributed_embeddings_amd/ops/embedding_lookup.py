@@ -116,15 +116,20 @@ class _CsrLookup(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, weight, values, row_splits, combiner):
+    def forward(ctx, weight, values, row_splits, combiner, out_dtype=None):
         ctx.save_for_backward(values, row_splits)
         ctx.combiner = combiner
         ctx.vocab = weight.shape[0]
         ctx.width = weight.shape[1]
         ctx.wdtype = weight.dtype
         if weight.is_cuda:
-            return _backend.ops().csr_lookup_forward(weight, values, row_splits, combiner == "mean")
-        return _csr_lookup_ref(weight, values, row_splits, combiner)
+            # bf16 out is stored directly by the kernel (no cast kernel);
+            # the backward consumes bf16 grads natively too
+            return _backend.ops().csr_lookup_forward(
+                weight, values, row_splits, combiner == "mean",
+                out_dtype == torch.bfloat16)
+        out = _csr_lookup_ref(weight, values, row_splits, combiner)
+        return out.to(out_dtype) if out_dtype is not None else out
 
     @staticmethod
     def backward(ctx, grad_out):
@@ -144,7 +149,7 @@ class _CsrLookup(torch.autograd.Function):
             size=(ctx.vocab, ctx.width),
             is_coalesced=True,
         )
-        return grad_weight, None, None, None
+        return grad_weight, None, None, None, None
 
 
 def _csr_lookup_backward_ref(grad_out, values, row_splits, vocab, combiner):
@@ -161,8 +166,8 @@ def _csr_lookup_backward_ref(grad_out, values, row_splits, vocab, combiner):
     g = g[valid]
     unique_ids, inverse = torch.unique(vals, sorted=True, return_inverse=True)
     unique_grad = torch.zeros(unique_ids.numel(), grad_out.shape[1],
-                              dtype=grad_out.dtype, device=grad_out.device)
-    unique_grad.index_add_(0, inverse, g)
+                              dtype=torch.float32, device=grad_out.device)
+    unique_grad.index_add_(0, inverse, g.float())
     return unique_ids, unique_grad
 
 
@@ -180,15 +185,17 @@ class _CsrLookupFusedOptimizer(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, weight, values, row_splits, combiner, lr, state, adagrad,
-                eps):
+                eps, out_dtype=None):
         ctx.save_for_backward(weight, values, row_splits, lr, state)
         ctx.combiner = combiner
         ctx.adagrad = adagrad
         ctx.eps = eps
         if weight.is_cuda:
-            return _backend.ops().csr_lookup_forward(weight, values, row_splits,
-                                                     combiner == "mean")
-        return _csr_lookup_ref(weight, values, row_splits, combiner)
+            return _backend.ops().csr_lookup_forward(
+                weight, values, row_splits, combiner == "mean",
+                out_dtype == torch.bfloat16)
+        out = _csr_lookup_ref(weight, values, row_splits, combiner)
+        return out.to(out_dtype) if out_dtype is not None else out
 
     @staticmethod
     def backward(ctx, grad_out):
@@ -201,7 +208,8 @@ class _CsrLookupFusedOptimizer(torch.autograd.Function):
                     ctx.combiner == "mean", ctx.adagrad, ctx.eps)
             else:
                 unique_ids, unique_grad = _csr_lookup_backward_ref(
-                    grad_out, values, row_splits, weight.shape[0], ctx.combiner)
+                    grad_out.float(), values, row_splits, weight.shape[0],
+                    ctx.combiner)
                 lr_v = lr.item()
                 if ctx.adagrad:
                     state.index_add_(0, unique_ids, unique_grad * unique_grad)
@@ -212,19 +220,20 @@ class _CsrLookupFusedOptimizer(torch.autograd.Function):
                 else:
                     weight.index_add_(0, unique_ids,
                                       (unique_grad * (-lr_v)).to(weight.dtype))
-        return None, None, None, None, None, None, None, None
+        return (None,) * 9
 
 
-def csr_lookup_fused_sgd(weight, values, row_splits, combiner, lr):
+def csr_lookup_fused_sgd(weight, values, row_splits, combiner, lr,
+                         out_dtype=None):
     empty = torch.empty(0, dtype=torch.float32, device=weight.device)
     return _CsrLookupFusedOptimizer.apply(weight, values, row_splits, combiner,
-                                          lr, empty, False, 0.0)
+                                          lr, empty, False, 0.0, out_dtype)
 
 
 def csr_lookup_fused_optimizer(weight, values, row_splits, combiner, lr, state,
-                               adagrad, eps):
+                               adagrad, eps, out_dtype=None):
     return _CsrLookupFusedOptimizer.apply(weight, values, row_splits, combiner,
-                                          lr, state, adagrad, eps)
+                                          lr, state, adagrad, eps, out_dtype)
 
 
 def _dense_fixed_hotness(weight, ids, combiner):

@@ -614,7 +614,8 @@ class DistributedEmbedding(nn.Module):
                 else:
                     splits = torch.arange(allids.numel() + 1, device=allids.device,
                                           dtype=torch.long)
-                    emb = layer.csr_lookup(allids, splits, "sum")
+                    emb = layer.csr_lookup(allids, splits, "sum",
+                                           out_dtype=self._kernel_out_dtype(allids))
                 if getattr(self, "_output_dtype", None) is not None:
                     emb = emb.to(self._output_dtype)
                 if unsplit:
@@ -691,7 +692,8 @@ class DistributedEmbedding(nn.Module):
                                            Ragged(allvals.cpu(), allsplits.cpu()),
                                            grp.combiner).to(allvals.device)
                 else:
-                    out = layer.csr_lookup(allvals, allsplits, grp.combiner)
+                    out = layer.csr_lookup(allvals, allsplits, grp.combiner,
+                                           out_dtype=self._kernel_out_dtype(allvals))
                 if getattr(self, "_output_dtype", None) is not None:
                     out = out.to(self._output_dtype)
                 if unsplit:
@@ -700,6 +702,14 @@ class DistributedEmbedding(nn.Module):
                 for (j, nrows), part in zip(metas, parts):
                     outs[j] = part
         return outs
+
+    def _kernel_out_dtype(self, ids):
+        """bf16 when the caller asked for bf16 outputs and the lookup runs
+        the HIP kernel (the kernel stores bf16 directly; the later .to() then
+        no-ops).  None otherwise — CPU/gloo paths keep fp32 numerics."""
+        if getattr(self, "_output_dtype", None) == torch.bfloat16 and ids.is_cuda:
+            return torch.bfloat16
+        return None
 
     def _offset_vector(self, gi, spec, device, kind="elem"):
         """Cached fused-table offset vector for ``spec`` = [(offset, count)].

@@ -621,3 +621,76 @@ def test_dlrm_packed_gpu_matches_legacy():
     err = float((o1.float() - o2.float()).abs().max())
     assert err < 0.05, err
     o1.float().square().sum().backward()
+
+
+@requires_gpu
+@pytest.mark.parametrize("width", [16, 64, 96, 128, 256])
+def test_csr_forward_bf16_out(width):
+    """out_bf16=True stores bf16 directly — equals fp32 result RNE-rounded."""
+    from distributed_embeddings_amd.ops import _backend
+    ext = _backend.ops()
+    torch.manual_seed(width)
+    vocab = 700
+    w = torch.randn(vocab, width, device="cuda")
+    ids = torch.randint(0, vocab, (3000,), device="cuda")
+    splits = torch.arange(0, 3001, 3, device="cuda")
+    out32 = ext.csr_lookup_forward(w, ids, splits, False)
+    out16 = ext.csr_lookup_forward(w, ids, splits, False, True)
+    assert out16.dtype == torch.bfloat16
+    assert torch.equal(out16, out32.bfloat16())
+
+
+@requires_gpu
+def test_csr_backward_bf16_grad():
+    """The sparse backward consumes bf16 upstream grads natively."""
+    from distributed_embeddings_amd.ops import _backend
+    ext = _backend.ops()
+    torch.manual_seed(5)
+    vocab, width = 500, 64
+    ids = torch.randint(0, vocab, (4000,), device="cuda")
+    splits = torch.arange(0, 4001, 4, device="cuda")
+    g32 = torch.randn(1000, width, device="cuda")
+    g16 = g32.bfloat16()
+    u32, gr32 = ext.csr_lookup_backward(g16.float(), ids, splits, vocab, False)
+    u16, gr16 = ext.csr_lookup_backward(g16, ids, splits, vocab, False)
+    assert torch.equal(u32, u16)
+    assert gr16.dtype == torch.float32
+    assert torch.allclose(gr32, gr16, atol=1e-5), \
+        float((gr32 - gr16).abs().max())
+
+
+@requires_gpu
+def test_fused_sgd_bf16_grad_and_out():
+    """Fused in-backward SGD with bf16 lookup output: the update equals the
+    explicit fp32-grad update up to bf16 grad rounding."""
+    from distributed_embeddings_amd.ops.embedding_lookup import (
+        Ragged, csr_lookup_fused_sgd)
+    torch.manual_seed(7)
+    vocab, width = 300, 128
+    w0 = torch.randn(vocab, width, device="cuda")
+    ids = torch.randint(0, vocab, (2000,), device="cuda")
+    splits = torch.arange(0, 2001, 2, device="cuda")
+    lr = torch.tensor([0.5], device="cuda")
+
+    w_a = w0.clone().requires_grad_(True)
+    out = csr_lookup_fused_sgd(w_a, ids, splits, "sum", lr,
+                               out_dtype=torch.bfloat16)
+    assert out.dtype == torch.bfloat16
+    gout = torch.randn_like(out)  # bf16 upstream grad
+    out.backward(gout)
+
+    # oracle: same bf16 grads, fp32 math on CPU
+    w_b = w0.cpu().clone().requires_grad_(True)
+    out_b = torch.zeros(1000, width)
+    idc = ids.cpu()
+    for r in range(1000):
+        seg = idc[2 * r:2 * r + 2]
+        out_b[r] = w_b.data[seg].sum(0)
+    g = gout.float().cpu()
+    grad = torch.zeros(vocab, width)
+    for r in range(1000):
+        for k in idc[2 * r:2 * r + 2]:
+            grad[k] += g[r]
+    expect = w0.cpu() - 0.5 * grad
+    err = (w_a.detach().cpu() - expect).abs().max()
+    assert float(err) < 1e-4, float(err)
