@@ -428,47 +428,58 @@ torch::Tensor dot_interact_fwd(torch::Tensor feats, int64_t out_w) {
   return out;
 }
 
+// sample_major=false: packed is [P, B, D] (a2a recv layout, world>1);
+// sample_major=true: packed is [B, P, D] (world==1 zero-copy lookup output —
+// each sample's feature rows are adjacent, so reads/writes stay local).
 torch::Tensor dot_interact_fwd_packed(torch::Tensor bottom,
                                       torch::Tensor packed, torch::Tensor perm,
-                                      int64_t out_w) {
+                                      int64_t out_w, bool sample_major) {
   CHECK_CUDA(bottom); CHECK_CUDA(packed); CHECK_CUDA(perm);
   CHECK_CONTIG(bottom); CHECK_CONTIG(packed); CHECK_CONTIG(perm);
   TORCH_CHECK(bottom.dtype() == torch::kBFloat16 &&
               packed.dtype() == torch::kBFloat16, "bf16 required");
   TORCH_CHECK(perm.dtype() == torch::kInt32, "perm must be int32");
-  TORCH_CHECK(bottom.dim() == 2 && packed.dim() == 3, "bottom [B,D], packed [P,B,D]");
+  TORCH_CHECK(bottom.dim() == 2 && packed.dim() == 3, "bottom [B,D], packed 3-D");
   const int64_t B = bottom.size(0);
-  const int P = (int)packed.size(0), D = (int)bottom.size(1);
-  TORCH_CHECK(packed.size(1) == B && packed.size(2) == D, "shape mismatch");
+  const int D = (int)bottom.size(1);
+  const int P = (int)(sample_major ? packed.size(1) : packed.size(0));
+  TORCH_CHECK(packed.size(sample_major ? 0 : 1) == B && packed.size(2) == D,
+              "shape mismatch");
   TORCH_CHECK(perm.numel() == P, "perm must have P entries");
   const int F = P + 1;
   TORCH_CHECK(F <= 32 && D % 32 == 0, "F<=32 and D%32==0 required");
   const int tri_n = F * (F - 1) / 2;
   TORCH_CHECK(out_w >= tri_n + D, "out width too small");
+  const int64_t sb = sample_major ? P : 1;
+  const int64_t sp = sample_major ? 1 : B;
   auto out = torch::empty({B, out_w}, bottom.options());
   launch_dot_interact_fwd_packed(bottom.data_ptr(), packed.data_ptr(),
                                  perm.data_ptr<int>(), out.data_ptr(), B, F, D,
-                                 (int)out_w, tri_n, current_stream());
+                                 (int)out_w, tri_n, sb, sp, current_stream());
   return out;
 }
 
 std::vector<torch::Tensor> dot_interact_bwd_packed(torch::Tensor gout,
                                                    torch::Tensor bottom,
                                                    torch::Tensor packed,
-                                                   torch::Tensor perm) {
+                                                   torch::Tensor perm,
+                                                   bool sample_major) {
   CHECK_CUDA(gout); CHECK_CUDA(bottom); CHECK_CUDA(packed); CHECK_CUDA(perm);
   CHECK_CONTIG(gout); CHECK_CONTIG(bottom); CHECK_CONTIG(packed);
   TORCH_CHECK(gout.dtype() == torch::kBFloat16);
   const int64_t B = bottom.size(0);
-  const int P = (int)packed.size(0), D = (int)bottom.size(1);
+  const int D = (int)bottom.size(1);
+  const int P = (int)(sample_major ? packed.size(1) : packed.size(0));
   const int F = P + 1;
   const int tri_n = F * (F - 1) / 2;
+  const int64_t sb = sample_major ? P : 1;
+  const int64_t sp = sample_major ? 1 : B;
   auto gbottom = torch::empty_like(bottom);
   auto gpacked = torch::empty_like(packed);
   launch_dot_interact_bwd_packed(gout.data_ptr(), bottom.data_ptr(),
                                  packed.data_ptr(), perm.data_ptr<int>(),
                                  gbottom.data_ptr(), gpacked.data_ptr(), B, F,
-                                 D, (int)gout.size(1), tri_n,
+                                 D, (int)gout.size(1), tri_n, sb, sp,
                                  current_stream());
   return {gbottom, gpacked};
 }

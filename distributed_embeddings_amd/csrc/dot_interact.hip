@@ -186,12 +186,15 @@ __global__ void dot_interact_bwd(const __hip_bfloat16* __restrict__ gout,
 // column order identical at every world size.
 // ---------------------------------------------------------------------------
 
+// sb/sp: element-row strides of the packed block — feature-major [P,B,D]
+// uses (sb=1, sp=B); sample-major [B,P,D] (world==1 zero-copy layout) uses
+// (sb=P, sp=1), keeping every sample's feature rows adjacent in memory.
 template <int FMAX>
 __global__ void dot_interact_fwd_packed(
     const __hip_bfloat16* __restrict__ bottom,
     const __hip_bfloat16* __restrict__ packed,
     const int* __restrict__ perm, __hip_bfloat16* __restrict__ out, int64_t B,
-    int F, int D, int out_w, int tri_n) {
+    int F, int D, int out_w, int tri_n, int64_t sb, int64_t sp) {
   extern __shared__ short lds_all[];
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -213,7 +216,7 @@ __global__ void dot_interact_fwd_packed(
         const int64_t src_row = perm[row - 1];
         v = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const short*>(packed) +
-            (src_row * B + b) * (int64_t)D + col);
+            (src_row * sp + b * sb) * (int64_t)D + col);
       }
       *reinterpret_cast<bf16x8*>(&lds[row * ldst + col]) = v;
     }
@@ -259,7 +262,7 @@ __global__ void dot_interact_bwd_packed(
     const __hip_bfloat16* __restrict__ bottom,
     const __hip_bfloat16* __restrict__ packed, const int* __restrict__ perm,
     __hip_bfloat16* __restrict__ gbottom, __hip_bfloat16* __restrict__ gpacked,
-    int64_t B, int F, int D, int out_w, int tri_n) {
+    int64_t B, int F, int D, int out_w, int tri_n, int64_t sb, int64_t sp) {
   extern __shared__ short lds_all[];
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -281,7 +284,7 @@ __global__ void dot_interact_bwd_packed(
         const int64_t src_row = perm[row - 1];
         v = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const short*>(packed) +
-            (src_row * B + b) * (int64_t)D + col);
+            (src_row * sp + b * sb) * (int64_t)D + col);
       }
       *reinterpret_cast<bf16x8*>(&lds[row * D + col]) = v;
     }
@@ -326,7 +329,8 @@ __global__ void dot_interact_bwd_packed(
               gbottom[b * (int64_t)D + j] = __hip_bfloat16(v);
             } else {
               const int64_t dst_row = perm[i - 1];
-              gpacked[(dst_row * B + b) * (int64_t)D + j] = __hip_bfloat16(v);
+              gpacked[(dst_row * sp + b * sb) * (int64_t)D + j] =
+                  __hip_bfloat16(v);
             }
           }
         }
@@ -338,6 +342,7 @@ __global__ void dot_interact_bwd_packed(
 void launch_dot_interact_fwd_packed(const void* bottom, const void* packed,
                                     const int* perm, void* out, int64_t B,
                                     int F, int D, int out_w, int tri_n,
+                                    int64_t sb, int64_t sp,
                                     hipStream_t stream) {
   const int block = 256;
   const int waves = block / WAVE;
@@ -348,13 +353,14 @@ void launch_dot_interact_fwd_packed(const void* bottom, const void* packed,
                      dim3(block), lds, stream,
                      (const __hip_bfloat16*)bottom,
                      (const __hip_bfloat16*)packed, perm, (__hip_bfloat16*)out,
-                     B, F, D, out_w, tri_n);
+                     B, F, D, out_w, tri_n, sb, sp);
 }
 
 void launch_dot_interact_bwd_packed(const void* gout, const void* bottom,
                                     const void* packed, const int* perm,
                                     void* gbottom, void* gpacked, int64_t B,
                                     int F, int D, int out_w, int tri_n,
+                                    int64_t sb, int64_t sp,
                                     hipStream_t stream) {
   const int block = 256;
   const int waves = block / WAVE;
@@ -366,7 +372,7 @@ void launch_dot_interact_bwd_packed(const void* gout, const void* bottom,
                      (const __hip_bfloat16*)bottom,
                      (const __hip_bfloat16*)packed, perm,
                      (__hip_bfloat16*)gbottom, (__hip_bfloat16*)gpacked, B, F,
-                     D, out_w, tri_n);
+                     D, out_w, tri_n, sb, sp);
 }
 
 void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,

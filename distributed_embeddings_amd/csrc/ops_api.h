@@ -84,11 +84,13 @@ void launch_sparse_row_update(void* weight, bool weight_bf16, float* state,
 void launch_dot_interact_fwd_packed(const void* bottom, const void* packed,
                                     const int* perm, void* out, int64_t B,
                                     int F, int D, int out_w, int tri_n,
+                                    int64_t sb, int64_t sp,
                                     hipStream_t stream);
 void launch_dot_interact_bwd_packed(const void* gout, const void* bottom,
                                     const void* packed, const int* perm,
                                     void* gbottom, void* gpacked, int64_t B,
                                     int F, int D, int out_w, int tri_n,
+                                    int64_t sb, int64_t sp,
                                     hipStream_t stream);
 void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,
                              int D, int out_w, int tri_n, hipStream_t stream);

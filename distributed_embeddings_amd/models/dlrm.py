@@ -150,10 +150,11 @@ class DLRM(nn.Module):
         use_packed = self._dot_perm is not None and all(
             isinstance(x, torch.Tensor) and x.dim() == 1 for x in cats)
         if use_packed:
-            packed = self.embeddings.forward_packed(
+            packed, smaj = self.embeddings.forward_packed(
                 cats, output_dtype=bottom.dtype, async_handle=handle)
             x = dot_interact_packed(packed, bottom, self._dot_perm,
-                                    pad_to=self.interact_pad)
+                                    pad_to=self.interact_pad,
+                                    sample_major=smaj)
         else:
             emb = self.embeddings(cats, output_dtype=bottom.dtype,
                                   async_handle=handle)

@@ -41,29 +41,35 @@ def _torch_dot_interact(feats: torch.Tensor, pad_to: int) -> torch.Tensor:
 
 class _DotInteractPacked(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, bottom, packed, perm, out_w):
+    def forward(ctx, bottom, packed, perm, out_w, sample_major):
         ctx.save_for_backward(bottom, packed, perm)
-        return _backend.ops().dot_interact_fwd_packed(bottom, packed, perm, out_w)
+        ctx.sample_major = sample_major
+        return _backend.ops().dot_interact_fwd_packed(bottom, packed, perm,
+                                                      out_w, sample_major)
 
     @staticmethod
     def backward(ctx, gout):
         bottom, packed, perm = ctx.saved_tensors
         gbottom, gpacked = _backend.ops().dot_interact_bwd_packed(
-            gout.contiguous(), bottom, packed, perm)
-        return gbottom, gpacked, None, None
+            gout.contiguous(), bottom, packed, perm, ctx.sample_major)
+        return gbottom, gpacked, None, None, None
 
 
 def dot_interact_packed(packed: torch.Tensor, bottom_mlp_out: torch.Tensor,
-                        perm: torch.Tensor, pad_to: int = 0) -> torch.Tensor:
-    """Interaction over feature-major packed embeddings (zero-copy input).
+                        perm: torch.Tensor, pad_to: int = 0,
+                        sample_major: bool = False) -> torch.Tensor:
+    """Interaction over packed embeddings (zero-copy input).
 
-    ``packed``: [P, B, D] — the fused-group lookup output (world==1) or the
-    mp->dp all-to-all recv buffer viewed in place (see
-    ``DistributedEmbedding.forward_packed``).  ``perm[f-1]`` maps model
+    ``packed``: [P, B, D] feature-major (the mp->dp all-to-all recv buffer,
+    world>1) or [B, P, D] sample-major (the world==1 fused lookup output) —
+    see ``DistributedEmbedding.forward_packed``.  ``perm[f-1]`` maps model
     feature f to its packed row, so output columns are in model input order
     at every world size.
     """
-    p, b, d = packed.shape
+    if sample_major:
+        b, p, d = packed.shape
+    else:
+        p, b, d = packed.shape
     f = p + 1
     width = f * (f - 1) // 2 + d
     out_w = max(pad_to, width)
@@ -71,11 +77,13 @@ def dot_interact_packed(packed: torch.Tensor, bottom_mlp_out: torch.Tensor,
             and bottom_mlp_out.dtype == torch.bfloat16 and f <= 32
             and d % 32 == 0):
         return _DotInteractPacked.apply(bottom_mlp_out.contiguous(),
-                                        packed.contiguous(), perm, out_w)
+                                        packed.contiguous(), perm, out_w,
+                                        sample_major)
     # fallback (CPU / unsupported shapes): materialize [B, F, D] and reuse
     # the plain path
-    feats = torch.cat([bottom_mlp_out.unsqueeze(0),
-                       packed.index_select(0, perm.long())]).transpose(0, 1)
+    sel = packed.index_select(1, perm.long()) if sample_major else \
+        packed.index_select(0, perm.long()).transpose(0, 1)
+    feats = torch.cat([bottom_mlp_out.unsqueeze(1), sel], dim=1)
     return _torch_dot_interact(feats.contiguous(), out_w)
 
 

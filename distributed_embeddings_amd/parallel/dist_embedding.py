@@ -886,6 +886,9 @@ class DistributedEmbedding(nn.Module):
                   for s in plan.rank_slices[r]}
         if len(widths) != 1:
             return False
+        if any(g.cpu_offload for r in range(self.world_size)
+               for g in plan.rank_concat_groups[r]):
+            return False
         return all(len(plan.rank_concat_groups[r]) == 1
                    for r in range(self.world_size))
 
@@ -900,15 +903,18 @@ class DistributedEmbedding(nn.Module):
 
     def forward_packed(self, inputs: Sequence[torch.Tensor],
                        output_dtype: Optional[torch.dtype] = None,
-                       async_handle: Optional[dict] = None) -> torch.Tensor:
-        """Returns ALL lookups as one feature-major [P, b, width] tensor.
+                       async_handle: Optional[dict] = None):
+        """Returns ALL lookups as one packed tensor: ``(packed, sample_major)``.
 
-        Zero-copy relative to :meth:`forward`: at world==1 this is a view of
-        the fused-group lookup output; at world>1 it is a view of the mp->dp
-        all-to-all recv buffer — the per-pair split + stack/merge copies of
-        the general path disappear.  Rows are in worker order (see
-        :meth:`packed_order`).  Requires :meth:`packed_forward_available` and
-        hotness-1 dense inputs (or 2-D with a combiner).
+        Zero-copy relative to :meth:`forward`: at world==1 ``packed`` is
+        [b, P, width] SAMPLE-major — the fused lookup runs over
+        sample-interleaved ids so its output is the torch.stack layout
+        directly; at world>1 ``packed`` is [P, b, width] feature-major — a
+        view of the mp->dp all-to-all recv buffer.  Packed rows are in worker
+        order (see :meth:`packed_order`); the per-pair split + stack/merge
+        copies of the general path disappear.  Requires
+        :meth:`packed_forward_available` and hotness-1 dense inputs (or 2-D
+        with a combiner).
         """
         plan = self.strategy
         W = self.world_size
@@ -932,18 +938,63 @@ class DistributedEmbedding(nn.Module):
                 recv_ids, async_handle["my_sizes"], col_inputs)
         else:
             pair_ids = self._dp_to_mp_dense(col_inputs)
+        if W == 1:
+            # SAMPLE-major batch: ids interleaved per sample, so the lookup
+            # output IS [b, P, D] — the torch.stack layout with zero copies,
+            # and every sample's feature rows stay adjacent for the
+            # interaction kernel (feature-major scatters 256B segments
+            # across the whole block: measured 2-3x slower kernels).
+            return self._packed_lookup_sample_major(pair_ids, b), True
         group_out = self._fused_group_lookup(pair_ids, unsplit=True)
         D = group_out.shape[-1]
         P_local = len(pair_ids)
-        if W == 1:
-            return group_out.view(P_local, b, D)
         send = group_out.view(P_local, W, b * D).transpose(0, 1).reshape(-1)
         all_cols = self._exchange_pair_cols([D] * P_local)
         out_splits = [b * sum(all_cols[k]) for k in range(W)]
         in_splits = [b * D * P_local] * W
         recv = comm.all_to_all_single(send, out_splits, in_splits)
         P_total = sum(len(c) for c in all_cols)
-        return recv.view(P_total, b, D)
+        return recv.view(P_total, b, D), False
+
+    def _packed_lookup_sample_major(self, pair_ids, b):
+        """One fused CSR lookup over sample-interleaved ids -> [b, P, D]."""
+        plan = self.strategy
+        layer = self.col_layers[0]
+        grp = plan.local_concat_groups(self.rank)[0]
+        parts = [ids.view(b, -1) for ids in pair_ids]
+        mat = torch.cat(parts, dim=1)            # [b, sum_h] — ids only
+        allids = mat.reshape(-1)
+        device = allids.device
+        espec = tuple((self._pair_row_offset[j], int(p.shape[1]))
+                      for j, p in enumerate(parts))
+        key = ("smaj", espec, b, str(device))
+        cache = getattr(self, "_off_cache", None)
+        if cache is None:
+            cache = self._off_cache = {}
+        if key not in cache:
+            offp = None
+            if any(off for off, _ in espec):
+                offp = torch.cat([
+                    torch.full((h,), off, dtype=torch.long, device=device)
+                    for off, h in espec]).repeat(b)
+            if grp.combiner is None:
+                splits = torch.arange(allids.numel() + 1, device=device,
+                                      dtype=torch.long)
+            else:
+                lens = torch.tensor([h for _, h in espec], dtype=torch.long,
+                                    device=device).repeat(b)
+                splits = torch.zeros(lens.numel() + 1, dtype=torch.long,
+                                     device=device)
+                torch.cumsum(lens, 0, out=splits[1:])
+            cache[key] = (offp, splits)
+        offp, splits = cache[key]
+        if offp is not None:
+            allids = allids + offp
+        out = layer.csr_lookup(allids, splits, grp.combiner or "sum",
+                               out_dtype=self._kernel_out_dtype(allids))
+        if getattr(self, "_output_dtype", None) is not None:
+            out = out.to(self._output_dtype)
+        return out.view(b, len(pair_ids), -1)
 
     # --------------------------------------------------------- row slice path
 

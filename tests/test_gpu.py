@@ -694,3 +694,34 @@ def test_fused_sgd_bf16_grad_and_out():
     expect = w0.cpu() - 0.5 * grad
     err = (w_a.detach().cpu() - expect).abs().max()
     assert float(err) < 1e-4, float(err)
+
+
+@requires_gpu
+def test_dot_interact_packed_sample_major():
+    """Sample-major [B, P, D] packed layout (world==1 zero-copy) == oracle."""
+    from distributed_embeddings_amd.ops.dot_interact import (
+        _torch_dot_interact, dot_interact_packed)
+    torch.manual_seed(4)
+    B, P, D = 256, 26, 128
+    perm = torch.tensor(torch.randperm(P).tolist(), dtype=torch.int32,
+                        device="cuda")
+    packed = torch.randn(B, P, D, device="cuda").bfloat16().requires_grad_(True)
+    bottom = torch.randn(B, D, device="cuda").bfloat16().requires_grad_(True)
+    out = dot_interact_packed(packed, bottom, perm, pad_to=512,
+                              sample_major=True)
+    feats = torch.cat([bottom.detach().unsqueeze(1),
+                       packed.detach().index_select(1, perm.long())], dim=1)
+    ref = _torch_dot_interact(feats.float().contiguous(), 512)
+    err = (out.float() - ref).abs().max() / ref.abs().max()
+    assert float(err) < 0.02, float(err)
+    gout = torch.randn_like(out)
+    out.backward(gout)
+    f2 = feats.clone().requires_grad_(True)
+    _torch_dot_interact(f2, 512).backward(gout)
+    errb = (bottom.grad.float() - f2.grad[:, 0, :].float()).abs().max()
+    assert float(errb) < 0.5, float(errb)
+    inv = torch.empty(P, dtype=torch.long)
+    inv[perm.long().cpu()] = torch.arange(P)
+    gp_ref = f2.grad[:, 1:, :].index_select(1, inv.cuda())
+    errp = (packed.grad.float() - gp_ref.float()).abs().max()
+    assert float(errp) < 0.5, float(errp)
