@@ -939,12 +939,15 @@ class DistributedEmbedding(nn.Module):
         else:
             pair_ids = self._dp_to_mp_dense(col_inputs)
         if W == 1:
-            # SAMPLE-major batch: ids interleaved per sample, so the lookup
-            # output IS [b, P, D] — the torch.stack layout with zero copies,
-            # and every sample's feature rows stay adjacent for the
-            # interaction kernel (feature-major scatters 256B segments
-            # across the whole block: measured 2-3x slower kernels).
-            return self._packed_lookup_sample_major(pair_ids, b), True
+            import os
+            if os.environ.get("DE_PACKED_SMAJ") == "1":
+                # sample-interleaved ids give a [b, P, D] output directly,
+                # but MEASURED SLOWER end-to-end: they destroy the lookup's
+                # per-table cache locality (pair-major streams each table's
+                # whole batch through one hot region).  Kept for measurement.
+                return self._packed_lookup_sample_major(pair_ids, b), True
+            group_out = self._fused_group_lookup(pair_ids, unsplit=True)
+            return group_out.view(len(pair_ids), b, -1), False
         group_out = self._fused_group_lookup(pair_ids, unsplit=True)
         D = group_out.shape[-1]
         P_local = len(pair_ids)
