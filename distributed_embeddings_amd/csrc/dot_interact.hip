@@ -289,11 +289,13 @@ __global__ void dot_interact_bwd_packed(
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & (WAVE - 1);
   const int tid = threadIdx.x;
-  // per wave: [FMAX][D] feats + [FMAX][FMAX] gsym + [FMAX][D] grad
-  const int per_wave = FMAX * D * 2 + FMAX * FMAX;
+  // per wave: [FMAX][D] feats + [FMAX][FMAX] gsym.  The feats region is
+  // REUSED for the grad: each nj column chunk is overwritten only after
+  // both mi MFMA passes buffered their results in registers (keeps LDS at
+  // 40KB/workgroup -> 4 workgroups/CU).
+  const int per_wave = FMAX * D + FMAX * FMAX;
   short* lds = lds_all + wave * per_wave;
   short* gsym = lds + FMAX * D;
-  short* gbuf = gsym + FMAX * FMAX;
   const int d8 = D / 8;
 
   for (int64_t b0 = (int64_t)blockIdx.x * DI_WPB; b0 < B;
@@ -337,6 +339,7 @@ __global__ void dot_interact_bwd_packed(
       const int r16 = lane & 15;
       const int khalf = lane >> 4;
       for (int nj = 0; nj < D / 16; ++nj) {
+        f32x4 accs[FMAX / 16];
 #pragma unroll
         for (int mi = 0; mi < FMAX / 16; ++mi) {
           f32x4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -351,16 +354,19 @@ __global__ void dot_interact_bwd_packed(
             }
             acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
           }
+          accs[mi] = acc;
+        }
+        // both mi passes read columns nj*16.. — now overwrite them in place
+#pragma unroll
+        for (int mi = 0; mi < FMAX / 16; ++mi) {
 #pragma unroll
           for (int reg = 0; reg < 4; ++reg) {
             const int i = mi * 16 + (lane >> 4) * 4 + reg;
             const int j = nj * 16 + (lane & 15);
-            if (i < FMAX && j < D) {
-              float v = acc[reg];
-              if (i == 0 && j < D) v += float(grow[tri_n + j]);
-              __hip_bfloat16 hb(v);
-              gbuf[i * D + j] = *reinterpret_cast<short*>(&hb);
-            }
+            float v = accs[mi][reg];
+            if (i == 0) v += float(grow[tri_n + j]);
+            __hip_bfloat16 hb(v);
+            lds[i * D + j] = *reinterpret_cast<short*>(&hb);
           }
         }
       }
@@ -376,7 +382,7 @@ __global__ void dot_interact_bwd_packed(
       const int64_t b = b0 + w;
       if (b >= B || row >= F) continue;
       const bf16x8 v = *reinterpret_cast<const bf16x8*>(
-          &lds_all[w * per_wave + FMAX * D + FMAX * FMAX + row * D + col]);
+          &lds_all[w * per_wave + row * D + col]);
       if (row == 0) {
         *reinterpret_cast<bf16x8*>(
             reinterpret_cast<short*>(gbottom) + b * (int64_t)D + col) = v;
@@ -397,7 +403,7 @@ void launch_dot_interact_fwd_packed(const void* bottom, const void* packed,
                                     hipStream_t stream) {
   const int block = DI_WPB * WAVE;
   int64_t blocks = (B + DI_WPB - 1) / DI_WPB;
-  if (blocks > 8192) blocks = 8192;
+  if (blocks > 32768) blocks = 32768;
   const size_t lds = (size_t)DI_WPB * 32 * (D + 8) * sizeof(short);
   hipLaunchKernelGGL((dot_interact_fwd_packed<32>), dim3((int)blocks),
                      dim3(block), lds, stream,
@@ -414,8 +420,8 @@ void launch_dot_interact_bwd_packed(const void* gout, const void* bottom,
                                     hipStream_t stream) {
   const int block = DI_WPB * WAVE;
   int64_t blocks = (B + DI_WPB - 1) / DI_WPB;
-  if (blocks > 8192) blocks = 8192;
-  const size_t lds = (size_t)DI_WPB * (32 * D * 2 + 32 * 32) * sizeof(short);
+  if (blocks > 32768) blocks = 32768;
+  const size_t lds = (size_t)DI_WPB * (32 * D + 32 * 32) * sizeof(short);
   hipLaunchKernelGGL((dot_interact_bwd_packed<32>), dim3((int)blocks),
                      dim3(block), lds, stream, (const __hip_bfloat16*)gout,
                      (const __hip_bfloat16*)bottom,
