@@ -270,3 +270,50 @@ def test_async_redistribute_world2():
 def test_async_redistribute_world1_noop():
     r = _overlap_worker(0, 1)
     assert r["match"]
+
+
+# ------------------------------------------------------------ packed forward
+
+def _packed_forward_worker(rank, world):
+    """forward_packed == forward (rows permuted by packed_order), world 2."""
+    import distributed_embeddings_amd as de
+    sizes = [40, 60, 25, 33]
+    tables = [de.TableConfig(s, 8, None) for s in sizes]
+    model = de.DistributedEmbedding(tables, strategy="basic")
+    assert model.packed_forward_available()
+    g = torch.Generator().manual_seed(5)
+    weights = [torch.randn(s, 8, generator=g) for s in sizes]
+    model.set_weights([w.numpy() for w in weights])
+    gi = torch.Generator().manual_seed(41)
+    full = [torch.randint(0, s, (world * 4,), generator=gi) for s in sizes]
+    local = [x[rank * 4:(rank + 1) * 4] for x in full]
+    packed, smaj = model.forward_packed(local)
+    outs = model(local)
+    order = model.packed_order()
+    ok = True
+    for f in range(len(sizes)):
+        row = packed[:, order[f], :] if smaj else packed[order[f]]
+        ok = ok and torch.allclose(row, outs[f], atol=1e-6)
+    # grads flow through the packed view
+    packed.sum().backward()
+    return ok
+
+
+def test_forward_packed_world2():
+    assert all(run_distributed(_packed_forward_worker, world=2))
+
+
+def test_forward_packed_world1():
+    assert _packed_forward_worker(0, 1)
+
+
+def test_forward_packed_unavailable_cases():
+    import distributed_embeddings_amd as de
+    # mixed widths -> no packed path
+    m = de.DistributedEmbedding([de.TableConfig(10, 8, "sum"),
+                                 de.TableConfig(10, 16, "sum")])
+    assert not m.packed_forward_available()
+    # row-sliced tables -> no packed path
+    m2 = de.DistributedEmbedding([de.TableConfig(1000, 8, "sum")],
+                                 row_slice_threshold=10)
+    assert not m2.packed_forward_available()
