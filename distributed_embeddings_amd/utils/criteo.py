@@ -102,7 +102,13 @@ class SyntheticDLRMData:
                  num_batches: int = 100, num_numerical: int = 13,
                  device="cpu", rank: int = 0,
                  feature_ids: Optional[Sequence[int]] = None,
-                 dp_input: bool = True, pool: int = 4, world: int = 1):
+                 dp_input: bool = True, pool: int = 4, world: int = 1,
+                 learnable: bool = False):
+        """``learnable=True`` makes labels a deterministic function of the
+        first categorical feature (id parity), so a correctly-scaled
+        optimizer drives the BCE well below the ln(2) random floor — the
+        convergence smoke for the lr contract (random labels cannot
+        distinguish a working optimizer from a frozen one)."""
         from .input_gen import make_batch
         self.num_batches = num_batches
         self.pool = []
@@ -115,7 +121,10 @@ class SyntheticDLRMData:
             cats = [c.to(device) for c in
                     make_batch(sizes, [1] * len(sizes), bs, generator=g)]
             num = torch.rand(local_bs, num_numerical, device=device)
-            labels = torch.randint(0, 2, (local_bs, 1), device=device).float()
+            if learnable and cats:
+                labels = (cats[0][:local_bs] % 2).view(-1, 1).float()
+            else:
+                labels = torch.randint(0, 2, (local_bs, 1), device=device).float()
             self.pool.append((num, cats, labels))
 
     def __len__(self):

@@ -48,6 +48,9 @@ def parse_args():
     p.add_argument("--decay-steps", type=int, default=30000)
     p.add_argument("--table-size-cap", type=int, default=None,
                    help="cap per-table vocab (small-memory smoke runs)")
+    p.add_argument("--learnable-labels", action="store_true",
+                   help="synthetic labels = f(ids): loss must drop below "
+                        "ln(2) if the optimizer scaling is right")
     return p.parse_args()
 
 
@@ -109,7 +112,8 @@ def main():
         data = SyntheticDLRMData(table_sizes, local_bs, num_batches=args.num_batches,
                                  device=device, rank=rank,
                                  feature_ids=feature_ids,
-                                 dp_input=args.dp_input or world == 1)
+                                 dp_input=args.dp_input or world == 1,
+                                 learnable=args.learnable_labels)
 
     # loss below is sum(BCE)/global_batch and dp grads are SUMMED
     # (average=False), so each grad is already the global-batch mean —
