@@ -141,8 +141,8 @@ def main():
             logits = model(num, cats)
             loss = loss_fn(logits.float(), labels)
         loss.backward()
+        sched.step()   # set THIS step's lr (warmup starts at step 0)
         opt.step()
-        sched.step()
         if rank == 0 and step % 50 == 0:
             l = float(loss.detach())
             print(f"step {step:5d} loss {l:.4f} "
