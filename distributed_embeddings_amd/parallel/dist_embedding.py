@@ -782,35 +782,45 @@ class DistributedEmbedding(nn.Module):
 
         outs = self._fused_group_lookup(pair_ids)  # per pair [W*b, c]
 
-        # mp->dp output all-to-all (parity: reference :868-878).
-        if outs:
-            send = torch.cat([o.reshape(W, -1) for o in outs], dim=1).reshape(-1)
+        if W == 1:
+            # the all-to-all is a passthrough: the pair outputs ARE the
+            # worker outputs — skip the send-side cat + recv split entirely
+            # (a 116-slice batched cat costs real time on many-table models)
+            worker_outs = outs
         else:
-            # empty-contribution rank: buffer must match peers' device/dtype
-            # for the RCCL collective, and must REQUIRE GRAD so this rank
-            # enters the reverse all-to-all in backward alongside its peers
-            send = torch.empty(
-                0,
-                dtype=getattr(self, "_output_dtype", None) or self.table_dtype,
-                device=self._comm_device(),
-                requires_grad=torch.is_grad_enabled())
-        my_cols = sum(o.shape[1] for o in outs) if outs else 0
-        in_splits = [local_b * my_cols] * W
-        # Per-pair output column counts of every rank (static after first call;
-        # out_cols may exceed the slice width for no-combiner multi-hot inputs).
-        my_pair_cols = [o.shape[1] for o in outs]
-        all_cols = self._exchange_pair_cols(my_pair_cols)
-        out_splits = [local_b * sum(all_cols[k]) for k in range(W)]
-        recv = comm.all_to_all_single(send, out_splits, in_splits)
+            # mp->dp output all-to-all (parity: reference :868-878).
+            if outs:
+                send = torch.cat([o.reshape(W, -1) for o in outs],
+                                 dim=1).reshape(-1)
+            else:
+                # empty-contribution rank: buffer must match peers'
+                # device/dtype for the RCCL collective, and must REQUIRE GRAD
+                # so this rank enters the reverse all-to-all in backward
+                send = torch.empty(
+                    0,
+                    dtype=getattr(self, "_output_dtype", None) or self.table_dtype,
+                    device=self._comm_device(),
+                    requires_grad=torch.is_grad_enabled())
+            my_cols = sum(o.shape[1] for o in outs) if outs else 0
+            in_splits = [local_b * my_cols] * W
+            # Per-pair output column counts of every rank (static after first
+            # call; out_cols may exceed the slice width for no-combiner
+            # multi-hot inputs).
+            my_pair_cols = [o.shape[1] for o in outs]
+            all_cols = self._exchange_pair_cols(my_pair_cols)
+            out_splits = [local_b * sum(all_cols[k]) for k in range(W)]
+            recv = comm.all_to_all_single(send, out_splits, in_splits)
 
-        # split per source rank, then per pair (pair-major layout within each
-        # source block, matching the send layout); reorder to input order.
-        chunks = torch.split(recv, out_splits)
-        worker_outs = []
-        for k in range(W):
-            sizes = [local_b * c for c in all_cols[k]]
-            parts = torch.split(chunks[k], sizes)
-            worker_outs.extend(p.view(local_b, c) for p, c in zip(parts, all_cols[k]))
+            # split per source rank, then per pair (pair-major layout within
+            # each source block, matching the send layout); reorder to input
+            # order.
+            chunks = torch.split(recv, out_splits)
+            worker_outs = []
+            for k in range(W):
+                sizes = [local_b * c for c in all_cols[k]]
+                parts = torch.split(chunks[k], sizes)
+                worker_outs.extend(p.view(local_b, c)
+                                   for p, c in zip(parts, all_cols[k]))
         ordered = [worker_outs[i] for i in plan.rev_tp_order]
         # per ordered entry, the slice width (ordered slices of one input are
         # in col_offset order — worker order == rank walk order in the plan)
