@@ -725,3 +725,37 @@ def test_dot_interact_packed_sample_major():
     gp_ref = f2.grad[:, 1:, :].index_select(1, inv.cuda())
     errp = (packed.grad.float() - gp_ref.float()).abs().max()
     assert float(errp) < 0.5, float(errp)
+
+
+@requires_gpu
+def test_kernel_determinism_short_segments():
+    """Same inputs -> bitwise-identical results, run to run: the short-segment
+    forward, the sorted sparse backward and the fused SGD update are
+    atomics-free (only power-law mega-segments take the atomic-combine long
+    path, which is float-order nondeterministic by design)."""
+    from distributed_embeddings_amd.ops import _backend
+    from distributed_embeddings_amd.ops.embedding_lookup import csr_lookup_fused_sgd
+    ext = _backend.ops()
+    torch.manual_seed(11)
+    vocab, width, rows, hot = 2_000_000, 128, 50_000, 4
+    w0 = torch.randn(vocab, width, device="cuda")
+    ids = torch.randint(0, vocab, (rows * hot,), device="cuda")
+    splits = torch.arange(0, rows * hot + 1, hot, device="cuda")
+    g = torch.randn(rows, width, device="cuda")
+
+    o1 = ext.csr_lookup_forward(w0, ids, splits, False)
+    o2 = ext.csr_lookup_forward(w0, ids, splits, False)
+    assert torch.equal(o1, o2)
+
+    u1, g1 = ext.csr_lookup_backward(g, ids, splits, vocab, False)
+    u2, g2 = ext.csr_lookup_backward(g, ids, splits, vocab, False)
+    assert torch.equal(u1, u2) and torch.equal(g1, g2)
+
+    lr = torch.tensor([0.1], device="cuda")
+    wa = w0.clone().requires_grad_(True)
+    wb = w0.clone().requires_grad_(True)
+    for wx in (wa, wb):
+        from distributed_embeddings_amd.ops.embedding_lookup import Ragged
+        out = csr_lookup_fused_sgd(wx, ids, splits, "sum", lr)
+        out.backward(g)
+    assert torch.equal(wa.detach(), wb.detach())

@@ -145,3 +145,53 @@ def test_invalid_strategy():
 def test_world1_passthrough():
     plan = DistEmbeddingStrategy(make([10, 20]), 1)
     assert len(plan.rank_slices[0]) == 2
+
+
+def test_planner_handles_jumbo_and_colossal_scales():
+    """BASELINE scale table (tiny 4.2 GiB ... colossal 22.3 TiB): host-only
+    plans on an 8-GPU 288 GB/GPU node.  'large' (774 GiB) fits entirely in
+    HBM via row+table sharding; jumbo (3.1 TiB) and colossal (22.3 TiB)
+    exceed aggregate HBM and must land within budget via CPU offload of the
+    overflow (parity: reference offload is a TP-group feature; row-sliced
+    tables are never offloaded)."""
+    from distributed_embeddings_amd.models.config import synthetic_models
+    from distributed_embeddings_amd.models.synthetic import expand_tables
+    from distributed_embeddings_amd.parallel.strategy import (
+        DistEmbeddingStrategy, TableConfig)
+
+    budget = int(0.9 * 288e9) // 4  # fp32 elements per GPU
+
+    def resident_elems(plan, r):
+        col = sum(plan.configs[s.table_id].input_dim * s.width
+                  for s in plan.rank_slices[r] if not s._offload)
+        row = sum(plan.row_shards[t][r].rows * plan.configs[t].output_dim
+                  for t in plan.row_table_ids)
+        return col + row
+
+    # large: fits HBM outright (row-slice the >=2e9-element tables)
+    tables, imap, _ = expand_tables(synthetic_models["large"])
+    cfgs = [TableConfig(r, w, "sum") for r, w in tables]
+    plan = DistEmbeddingStrategy(
+        cfgs, 8, strategy="memory_optimized", input_table_map=imap,
+        row_slice_threshold=2_000_000_000, gpu_embedding_size=budget)
+    for r in range(8):
+        assert resident_elems(plan, r) <= budget
+        assert not any(s._offload for s in plan.rank_slices[r])
+
+    # jumbo / colossal: beyond aggregate HBM -> overflow offloads to CPU
+    for name in ("jumbo", "colossal"):
+        tables, imap, _ = expand_tables(synthetic_models[name])
+        cfgs = [TableConfig(r, w, "sum") for r, w in tables]
+        total = sum(r * w for r, w in tables)
+        assert total > 8 * budget  # the scale genuinely demands offload
+        plan = DistEmbeddingStrategy(
+            cfgs, 8, strategy="memory_optimized", input_table_map=imap,
+            gpu_embedding_size=budget)
+        offloaded = 0
+        for r in range(8):
+            assert resident_elems(plan, r) <= budget, (name, r)
+            offloaded += sum(plan.configs[s.table_id].input_dim * s.width
+                             for s in plan.rank_slices[r] if s._offload)
+        assert offloaded >= total - 8 * budget
+        # every table is placed exactly once
+        assert all(plan.table_slices[t] for t in plan.col_table_ids)
