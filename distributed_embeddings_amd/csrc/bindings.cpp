@@ -156,7 +156,7 @@ std::vector<torch::Tensor> csr_lookup_backward(torch::Tensor grad_out,
     w = torch::empty({nnz}, f32);
     w_ptr = w.data_ptr<float>();
   }
-  launch_expand_row_ids(row_splits.data_ptr<int64_t>(), num_rows,
+  launch_expand_row_ids(row_splits.data_ptr<int64_t>(), num_rows, nnz,
                         row_ids.data_ptr<int32_t>(), w_ptr, mean, stream);
 
   // 3. radix sort (ids, position) — end_bit covers [0, vocab] inclusive.
@@ -339,7 +339,7 @@ void csr_fused_optimizer_apply(torch::Tensor weight, torch::Tensor state,
     w = torch::empty({nnz}, f32);
     w_ptr = w.data_ptr<float>();
   }
-  launch_expand_row_ids(row_splits.data_ptr<int64_t>(), num_rows,
+  launch_expand_row_ids(row_splits.data_ptr<int64_t>(), num_rows, nnz,
                         row_ids.data_ptr<int32_t>(), w_ptr, mean, stream);
   auto sorted_ids = torch::empty({nnz}, i64);
   auto sorted_pos = torch::empty({nnz}, i32);

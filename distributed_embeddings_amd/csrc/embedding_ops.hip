@@ -656,10 +656,44 @@ size_t csr_backward_temp_bytes(int64_t nnz, int64_t vocab) {
 
 // Orchestration is done on the host side (bindings.cpp) because output
 // allocation needs num_unique; these launchers expose the pieces.
+// Short segments (the common forward-input shape: hotness 1..tens): one
+// THREAD per row — a wave per hotness-1 row left 63 of 64 lanes idle and
+// made this trivial expansion ~100x slower than its write bandwidth.
+template <bool MEAN>
+__global__ void expand_row_ids_thread(const int64_t* __restrict__ splits,
+                                      int64_t num_rows,
+                                      int32_t* __restrict__ row_ids,
+                                      float* __restrict__ w) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < num_rows; row += stride) {
+    const int64_t s = splits[row], e = splits[row + 1];
+    const float iw = (MEAN && e > s) ? 1.f / (float)(e - s) : 1.f;
+    for (int64_t k = s; k < e; ++k) {
+      row_ids[k] = (int32_t)row;
+      if (MEAN) w[k] = iw;
+    }
+  }
+}
+
 void launch_expand_row_ids(const int64_t* splits, int64_t num_rows,
-                           int32_t* row_ids, float* w, bool mean,
+                           int64_t nnz, int32_t* row_ids, float* w, bool mean,
                            hipStream_t stream) {
   const int block = 256;
+  const int64_t ave = num_rows > 0 ? nnz / num_rows : 0;
+  if (ave <= 16) {
+    int64_t blocks = cdiv64(num_rows, block);
+    if (blocks > 16384) blocks = 16384;
+    if (blocks < 1) blocks = 1;
+    if (mean)
+      hipLaunchKernelGGL(expand_row_ids_thread<true>, dim3((int)blocks),
+                         dim3(block), 0, stream, splits, num_rows, row_ids, w);
+    else
+      hipLaunchKernelGGL(expand_row_ids_thread<false>, dim3((int)blocks),
+                         dim3(block), 0, stream, splits, num_rows, row_ids,
+                         nullptr);
+    return;
+  }
   const int grid = pick_grid(num_rows, block / WAVE);
   if (mean)
     hipLaunchKernelGGL(expand_row_ids<true>, dim3(grid), dim3(block), 0,

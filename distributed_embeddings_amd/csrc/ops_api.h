@@ -21,7 +21,7 @@ void launch_row_to_split(const int64_t* rows, int64_t nnz, int64_t num_rows,
 size_t csr_backward_temp_bytes(int64_t nnz, int64_t vocab);
 
 void launch_expand_row_ids(const int64_t* splits, int64_t num_rows,
-                           int32_t* row_ids, float* w, bool mean,
+                           int64_t nnz, int32_t* row_ids, float* w, bool mean,
                            hipStream_t stream);
 
 hipError_t run_inclusive_scan_i32(void* temp, size_t temp_bytes,
