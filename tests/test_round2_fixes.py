@@ -317,3 +317,24 @@ def test_forward_packed_unavailable_cases():
     m2 = de.DistributedEmbedding([de.TableConfig(1000, 8, "sum")],
                                  row_slice_threshold=10)
     assert not m2.packed_forward_available()
+
+
+def test_forward_packed_sample_major_env(monkeypatch):
+    """DE_PACKED_SMAJ=1 (measurement-only layout) returns [b, P, D] equal to
+    the per-input outputs."""
+    import distributed_embeddings_amd as de
+    monkeypatch.setenv("DE_PACKED_SMAJ", "1")
+    sizes = [40, 60, 25]
+    tables = [de.TableConfig(s, 8, None) for s in sizes]
+    model = de.DistributedEmbedding(tables, strategy="basic")
+    g = torch.Generator().manual_seed(5)
+    weights = [torch.randn(s, 8, generator=g) for s in sizes]
+    model.set_weights([w.numpy() for w in weights])
+    ids = [torch.randint(0, s, (4,), generator=g) for s in sizes]
+    packed, smaj = model.forward_packed(ids)
+    assert smaj and packed.shape == (4, 3, 8)
+    outs = model(ids)
+    order = model.packed_order()
+    for f in range(3):
+        assert torch.allclose(packed[:, order[f], :], outs[f], atol=1e-6)
+    packed.sum().backward()
