@@ -197,9 +197,20 @@ __global__ void dot_interact_bwd(const __hip_bfloat16* __restrict__ gout,
 // its [F, D] grad in LDS (overwriting nothing the MFMA still needs) and
 // writes it back cooperatively the same way.
 
-#define DI_WPB 4  // waves (= samples) per block
+// waves (= samples) per block: templated; DI_WPB is the measured default,
+// DE_DI_WPB env selects 2/4/8 for measurement (tools/bench_interact.py)
+#define DI_WPB 4
+static int di_wpb() {
+  static int v = [] {
+    const char* e = getenv("DE_DI_WPB");
+    if (!e) return DI_WPB;
+    int i = atoi(e);
+    return (i == 2 || i == 4 || i == 8) ? i : DI_WPB;
+  }();
+  return v;
+}
 
-template <int FMAX>
+template <int FMAX, int WPB>
 __global__ void dot_interact_fwd_packed(
     const __hip_bfloat16* __restrict__ bottom,
     const __hip_bfloat16* __restrict__ packed,
@@ -213,13 +224,13 @@ __global__ void dot_interact_fwd_packed(
   short* lds = lds_all + wave * FMAX * ldst;
   const int d8 = D / 8;
 
-  for (int64_t b0 = (int64_t)blockIdx.x * DI_WPB; b0 < B;
-       b0 += (int64_t)gridDim.x * DI_WPB) {
+  for (int64_t b0 = (int64_t)blockIdx.x * WPB; b0 < B;
+       b0 += (int64_t)gridDim.x * WPB) {
     // cooperative load: chunk i -> (row, sample w, col); per row the WPB
     // samples' segments are contiguous in the feature-major layout
-    for (int i = tid; i < FMAX * DI_WPB * d8; i += DI_WPB * WAVE) {
-      const int row = i / (DI_WPB * d8);
-      const int rem = i % (DI_WPB * d8);
+    for (int i = tid; i < FMAX * WPB * d8; i += WPB * WAVE) {
+      const int row = i / (WPB * d8);
+      const int rem = i % (WPB * d8);
       const int w = rem / d8;
       const int col = (rem % d8) * 8;
       const int64_t b = b0 + w;
@@ -278,7 +289,7 @@ __global__ void dot_interact_fwd_packed(
   }
 }
 
-template <int FMAX>
+template <int FMAX, int WPB>
 __global__ void dot_interact_bwd_packed(
     const __hip_bfloat16* __restrict__ gout,
     const __hip_bfloat16* __restrict__ bottom,
@@ -298,11 +309,11 @@ __global__ void dot_interact_bwd_packed(
   short* gsym = lds + FMAX * D;
   const int d8 = D / 8;
 
-  for (int64_t b0 = (int64_t)blockIdx.x * DI_WPB; b0 < B;
-       b0 += (int64_t)gridDim.x * DI_WPB) {
-    for (int i = tid; i < FMAX * DI_WPB * d8; i += DI_WPB * WAVE) {
-      const int row = i / (DI_WPB * d8);
-      const int rem = i % (DI_WPB * d8);
+  for (int64_t b0 = (int64_t)blockIdx.x * WPB; b0 < B;
+       b0 += (int64_t)gridDim.x * WPB) {
+    for (int i = tid; i < FMAX * WPB * d8; i += WPB * WAVE) {
+      const int row = i / (WPB * d8);
+      const int rem = i % (WPB * d8);
       const int w = rem / d8;
       const int col = (rem % d8) * 8;
       const int64_t b = b0 + w;
@@ -374,9 +385,9 @@ __global__ void dot_interact_bwd_packed(
     __syncthreads();
 
     // cooperative write-back: same (row, sample, col) mapping as the load
-    for (int i = tid; i < FMAX * DI_WPB * d8; i += DI_WPB * WAVE) {
-      const int row = i / (DI_WPB * d8);
-      const int rem = i % (DI_WPB * d8);
+    for (int i = tid; i < FMAX * WPB * d8; i += WPB * WAVE) {
+      const int row = i / (WPB * d8);
+      const int rem = i % (WPB * d8);
       const int w = rem / d8;
       const int col = (rem % d8) * 8;
       const int64_t b = b0 + w;
@@ -401,15 +412,19 @@ void launch_dot_interact_fwd_packed(const void* bottom, const void* packed,
                                     int F, int D, int out_w, int tri_n,
                                     int64_t sb, int64_t sp,
                                     hipStream_t stream) {
-  const int block = DI_WPB * WAVE;
-  int64_t blocks = (B + DI_WPB - 1) / DI_WPB;
+  const int wpb = di_wpb();
+  const int block = wpb * WAVE;
+  int64_t blocks = (B + wpb - 1) / wpb;
   if (blocks > 32768) blocks = 32768;
-  const size_t lds = (size_t)DI_WPB * 32 * (D + 8) * sizeof(short);
-  hipLaunchKernelGGL((dot_interact_fwd_packed<32>), dim3((int)blocks),
-                     dim3(block), lds, stream,
-                     (const __hip_bfloat16*)bottom,
-                     (const __hip_bfloat16*)packed, perm, (__hip_bfloat16*)out,
-                     B, F, D, out_w, tri_n, sb, sp);
+  const size_t lds = (size_t)wpb * 32 * (D + 8) * sizeof(short);
+#define LFWD(W)                                                                \
+  hipLaunchKernelGGL((dot_interact_fwd_packed<32, W>), dim3((int)blocks),      \
+                     dim3(block), lds, stream,                                 \
+                     (const __hip_bfloat16*)bottom,                            \
+                     (const __hip_bfloat16*)packed, perm, (__hip_bfloat16*)out,\
+                     B, F, D, out_w, tri_n, sb, sp)
+  if (wpb == 2) LFWD(2); else if (wpb == 8) LFWD(8); else LFWD(4);
+#undef LFWD
 }
 
 void launch_dot_interact_bwd_packed(const void* gout, const void* bottom,
@@ -418,16 +433,20 @@ void launch_dot_interact_bwd_packed(const void* gout, const void* bottom,
                                     int F, int D, int out_w, int tri_n,
                                     int64_t sb, int64_t sp,
                                     hipStream_t stream) {
-  const int block = DI_WPB * WAVE;
-  int64_t blocks = (B + DI_WPB - 1) / DI_WPB;
+  const int wpb = di_wpb();
+  const int block = wpb * WAVE;
+  int64_t blocks = (B + wpb - 1) / wpb;
   if (blocks > 32768) blocks = 32768;
-  const size_t lds = (size_t)DI_WPB * (32 * D + 32 * 32) * sizeof(short);
-  hipLaunchKernelGGL((dot_interact_bwd_packed<32>), dim3((int)blocks),
-                     dim3(block), lds, stream, (const __hip_bfloat16*)gout,
-                     (const __hip_bfloat16*)bottom,
-                     (const __hip_bfloat16*)packed, perm,
-                     (__hip_bfloat16*)gbottom, (__hip_bfloat16*)gpacked, B, F,
-                     D, out_w, tri_n, sb, sp);
+  const size_t lds = (size_t)wpb * (32 * D + 32 * 32) * sizeof(short);
+#define LBWD(W)                                                                \
+  hipLaunchKernelGGL((dot_interact_bwd_packed<32, W>), dim3((int)blocks),      \
+                     dim3(block), lds, stream, (const __hip_bfloat16*)gout,    \
+                     (const __hip_bfloat16*)bottom,                            \
+                     (const __hip_bfloat16*)packed, perm,                      \
+                     (__hip_bfloat16*)gbottom, (__hip_bfloat16*)gpacked, B, F, \
+                     D, out_w, tri_n, sb, sp)
+  if (wpb == 2) LBWD(2); else if (wpb == 8) LBWD(8); else LBWD(4);
+#undef LBWD
 }
 
 void launch_dot_interact_fwd(const void* feats, void* out, int64_t B, int F,

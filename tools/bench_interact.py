@@ -24,7 +24,19 @@ def timeit(fn, iters=30):
     return (time.perf_counter() - t0) / iters * 1e6
 
 
+def sweep_wpb():
+    import subprocess
+    env = dict(os.environ, _DI_WORKER="1")
+    for w in (2, 4, 8):
+        e = dict(env, DE_DI_WPB=str(w))
+        print(f"--- WPB={w} ---", flush=True)
+        subprocess.run([sys.executable, __file__], env=e, check=True)
+
+
 def main():
+    if "_DI_WORKER" not in os.environ and os.environ.get("DI_SWEEP") == "1":
+        sweep_wpb()
+        return
     torch.manual_seed(0)
     B, P, D = 65536, 26, 128
     F = P + 1
