@@ -199,15 +199,16 @@ __global__ void dot_interact_bwd(const __hip_bfloat16* __restrict__ gout,
 
 // waves (= samples) per block: templated; DI_WPB is the measured default,
 // DE_DI_WPB env selects 2/4/8 for measurement (tools/bench_interact.py)
-#define DI_WPB 4
-static int di_wpb() {
+// measured per-direction optima (tools/bench_interact.py sweep): fwd 2,
+// bwd 8 — the deltas are small; DE_DI_WPB overrides both for measurement
+static int di_wpb(int dflt) {
   static int v = [] {
     const char* e = getenv("DE_DI_WPB");
-    if (!e) return DI_WPB;
+    if (!e) return 0;
     int i = atoi(e);
-    return (i == 2 || i == 4 || i == 8) ? i : DI_WPB;
+    return (i == 2 || i == 4 || i == 8) ? i : 0;
   }();
-  return v;
+  return v ? v : dflt;
 }
 
 template <int FMAX, int WPB>
@@ -412,7 +413,7 @@ void launch_dot_interact_fwd_packed(const void* bottom, const void* packed,
                                     int F, int D, int out_w, int tri_n,
                                     int64_t sb, int64_t sp,
                                     hipStream_t stream) {
-  const int wpb = di_wpb();
+  const int wpb = di_wpb(2);
   const int block = wpb * WAVE;
   int64_t blocks = (B + wpb - 1) / wpb;
   if (blocks > 32768) blocks = 32768;
@@ -433,7 +434,7 @@ void launch_dot_interact_bwd_packed(const void* gout, const void* bottom,
                                     int F, int D, int out_w, int tri_n,
                                     int64_t sb, int64_t sp,
                                     hipStream_t stream) {
-  const int wpb = di_wpb();
+  const int wpb = di_wpb(8);
   const int block = wpb * WAVE;
   int64_t blocks = (B + wpb - 1) / wpb;
   if (blocks > 32768) blocks = 32768;
