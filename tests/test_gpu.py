@@ -759,3 +759,46 @@ def test_kernel_determinism_short_segments():
         out = csr_lookup_fused_sgd(wx, ids, splits, "sum", lr)
         out.backward(g)
     assert torch.equal(wa.detach(), wb.detach())
+
+
+@requires_gpu
+@pytest.mark.parametrize("mean", [False, True])
+def test_fused_adagrad_bf16_grad(mean):
+    """Fused Adagrad with bf16 upstream grads (scratch+finalize GT path)
+    equals the fp32-grad update."""
+    from distributed_embeddings_amd.ops import _backend
+    ext = _backend.ops()
+    torch.manual_seed(9)
+    vocab, width = 400, 64
+    ids = torch.randint(0, vocab, (3000,), device="cuda")
+    splits = torch.arange(0, 3001, 3, device="cuda")
+    g16 = torch.randn(1000, width, device="cuda").bfloat16()
+    lr = torch.tensor([0.3], device="cuda")
+
+    w_a = torch.randn(vocab, width, device="cuda")
+    st_a = torch.zeros_like(w_a)
+    w_b, st_b = w_a.clone(), st_a.clone()
+    ext.csr_fused_optimizer_apply(w_a, st_a, ids, splits, g16, lr,
+                                  mean, True, 1e-10)
+    ext.csr_fused_optimizer_apply(w_b, st_b, ids, splits, g16.float(), lr,
+                                  mean, True, 1e-10)
+    torch.cuda.synchronize()
+    assert torch.allclose(w_a, w_b, atol=1e-5), \
+        float((w_a - w_b).abs().max())
+    assert torch.allclose(st_a, st_b, atol=1e-4)
+
+
+@requires_gpu
+def test_csr_backward_bf16_grad_mean():
+    """bf16 grads through the MEAN-combiner backward (per-id weights + GT)."""
+    from distributed_embeddings_amd.ops import _backend
+    ext = _backend.ops()
+    torch.manual_seed(12)
+    vocab, width = 300, 96
+    ids = torch.randint(0, vocab, (5000,), device="cuda")
+    splits = torch.arange(0, 5001, 5, device="cuda")
+    g16 = torch.randn(1000, width, device="cuda").bfloat16()
+    u1, g1 = ext.csr_lookup_backward(g16, ids, splits, vocab, True)
+    u2, g2 = ext.csr_lookup_backward(g16.float(), ids, splits, vocab, True)
+    assert torch.equal(u1, u2)
+    assert torch.allclose(g1, g2, atol=1e-5), float((g1 - g2).abs().max())
