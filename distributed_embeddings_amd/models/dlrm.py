@@ -24,7 +24,6 @@ from torch import nn
 from ..layers.embedding import Embedding, scaled_uniform_init
 from ..ops.dot_interact import dot_interact as fused_dot_interact
 from ..ops.dot_interact import dot_interact_packed
-from ..ops.fused_linear import FusedLinearReLU
 from ..parallel import comm
 from ..parallel.dist_embedding import DistributedEmbedding
 
@@ -54,14 +53,12 @@ def _mlp(sizes: Sequence[int], in_dim: int, final_linear: bool) -> nn.Sequential
     layers: List[nn.Module] = []
     d = in_dim
     for i, s in enumerate(sizes):
-        if final_linear and i == len(sizes) - 1:
-            lin = nn.Linear(d, s)
-            nn.init.xavier_normal_(lin.weight)
-            nn.init.normal_(lin.bias, std=math.sqrt(1.0 / s))
-            layers.append(lin)
-        else:
-            # one hipBLASLt call with bias+ReLU epilogue per hidden layer
-            layers.append(FusedLinearReLU(d, s))
+        lin = nn.Linear(d, s)
+        nn.init.xavier_normal_(lin.weight)
+        nn.init.normal_(lin.bias, std=math.sqrt(1.0 / s))
+        layers.append(lin)
+        if not (final_linear and i == len(sizes) - 1):
+            layers.append(nn.ReLU(inplace=True))
         d = s
     return nn.Sequential(*layers)
 

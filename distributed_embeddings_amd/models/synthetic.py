@@ -77,11 +77,10 @@ class SyntheticModel(nn.Module):
             import math
             total_width = math.ceil(total_width / self.interact_stride)
         mlp_in = total_width + model_config.num_numerical_features
-        from ..ops.fused_linear import FusedLinearReLU
         mods: List[nn.Module] = []
         d = mlp_in
         for s in model_config.mlp_sizes:
-            mods.append(FusedLinearReLU(d, s))  # bias+ReLU GEMM epilogue
+            mods += [nn.Linear(d, s), nn.ReLU(inplace=True)]
             d = s
         mods.append(nn.Linear(d, 1))
         self.mlp = nn.Sequential(*mods)
