@@ -50,8 +50,10 @@ class IntegerLookup(nn.Module):
       max_tokens: vocabulary budget (value 0 is reserved for OOV).
       auto_grow: if True, reaching 80% load (or overflowing) doubles
         ``max_tokens`` and rehashes instead of mapping new keys to OOV.
-        Costs one device sync per call on GPU; keep False (reference
-        parity) on hot serving paths.
+        Growth checks track a sound upper bound on the vocabulary size, so
+        the device sync happens only when that bound nears capacity
+        (amortized away by each doubling); False = fixed capacity
+        (reference parity).
     """
 
     def __init__(self, max_tokens: int = 100000, auto_grow: bool = False,
@@ -72,6 +74,12 @@ class IntegerLookup(nn.Module):
         counts = torch.zeros(self.max_tokens + 1, dtype=torch.int32, device=device)
         counts[0] = 1
         self.register_buffer("counts", counts)
+        # auto_grow bookkeeping: vocabulary_size() <= _known_used + _unseen
+        # always holds (each looked-up element can add at most one value), so
+        # growth checks only SYNC when this upper bound nears capacity —
+        # steady state runs without any device sync.
+        self._known_used = 1
+        self._unseen = 0
 
     # ------------------------------------------------------------------ CPU
 
@@ -165,6 +173,7 @@ class IntegerLookup(nn.Module):
                 self.table_values.numpy()[hit_slot[upg]] = vals[upg]
             # overflow (auto_grow off): remaining keys stay at value 0
         out = vals[inverse]
+        self._unseen += int(flat.size)
         self.counts += torch.from_numpy(
             np.bincount(out, minlength=self.counts.numel()).astype(np.int32))
         return torch.from_numpy(out).view(keys.shape)
@@ -212,27 +221,40 @@ class IntegerLookup(nn.Module):
             keys = keys.long()
         if not keys.is_cuda:
             return self._forward_cpu(keys)
+        n_keys = keys.numel()
         if self.auto_grow:
-            # proactive growth at 80% load (one D2H sync — documented cost)
-            while self.vocabulary_size() > _GROW_AT * self.max_tokens:
-                self._grow()
+            # proactive growth at 80% load; the sync happens only when the
+            # sound upper bound (known_used + elements seen since the last
+            # exact count) approaches it — amortized away by each doubling
+            if self._known_used + self._unseen + n_keys > \
+                    _GROW_AT * self.max_tokens:
+                used = self.vocabulary_size()  # one D2H sync (rare)
+                self._known_used, self._unseen = used, 0
+                while used > _GROW_AT * self.max_tokens:
+                    self._grow()
         out = _backend.ops().integer_lookup(
             keys.contiguous().reshape(-1), self.table_keys, self.table_values,
             self.counts, self.max_tokens)
-        if self.auto_grow and self.vocabulary_size() >= self.max_tokens + 1:
-            # Table filled mid-batch: out==0 elements are unresolved keys
-            # (in auto_grow mode nothing is true-OOV).  Retract only their
-            # counts[0] contribution, grow, resolve just that subset —
-            # resolved keys keep their counts, so the counts>0 free-value
-            # invariant holds throughout.
-            unresolved = out == 0
-            n0 = int(unresolved.sum().item())
-            if n0:
-                self.counts[0] -= n0
-                self._grow()
-                out = out.clone()
-                out[unresolved] = self.forward(
-                    keys.contiguous().reshape(-1)[unresolved]).reshape(-1)
+        self._unseen += n_keys
+        if self.auto_grow and \
+                self._known_used + self._unseen >= self.max_tokens + 1:
+            used = self.vocabulary_size()
+            self._known_used, self._unseen = used, 0
+            if used >= self.max_tokens + 1:
+                # Table filled mid-batch: out==0 elements are unresolved keys
+                # (in auto_grow mode nothing is true-OOV).  Retract only
+                # their counts[0] contribution, grow, resolve just that
+                # subset — resolved keys keep their counts, so the counts>0
+                # free-value invariant holds throughout.
+                unresolved = out == 0
+                n0 = int(unresolved.sum().item())
+                if n0:
+                    self.counts[0] -= n0
+                    self._known_used -= n0
+                    self._grow()
+                    out = out.clone()
+                    out[unresolved] = self.forward(
+                        keys.contiguous().reshape(-1)[unresolved]).reshape(-1)
         return out.view(keys.shape)
 
     # --------------------------------------------------------- (de)serialize
