@@ -249,8 +249,9 @@ class IntegerLookup(nn.Module):
                 unresolved = out == 0
                 n0 = int(unresolved.sum().item())
                 if n0:
+                    # counts[0] retraction does NOT change vocabulary_size()
+                    # (value 0 stays pre-claimed), so _known_used is kept
                     self.counts[0] -= n0
-                    self._known_used -= n0
                     self._grow()
                     out = out.clone()
                     out[unresolved] = self.forward(
