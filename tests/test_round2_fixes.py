@@ -338,3 +338,12 @@ def test_forward_packed_sample_major_env(monkeypatch):
     for f in range(3):
         assert torch.allclose(packed[:, order[f], :], outs[f], atol=1e-6)
     packed.sum().backward()
+
+
+def test_forward_packed_world4():
+    assert all(run_distributed(_packed_forward_worker, world=4))
+
+
+def test_async_redistribute_world4():
+    r = run_distributed(_overlap_worker, world=4)
+    assert all(o["match"] for o in r)
