@@ -916,15 +916,16 @@ class DistributedEmbedding(nn.Module):
                        async_handle: Optional[dict] = None):
         """Returns ALL lookups as one packed tensor: ``(packed, sample_major)``.
 
-        Zero-copy relative to :meth:`forward`: at world==1 ``packed`` is
-        [b, P, width] SAMPLE-major — the fused lookup runs over
-        sample-interleaved ids so its output is the torch.stack layout
-        directly; at world>1 ``packed`` is [P, b, width] feature-major — a
-        view of the mp->dp all-to-all recv buffer.  Packed rows are in worker
-        order (see :meth:`packed_order`); the per-pair split + stack/merge
-        copies of the general path disappear.  Requires
-        :meth:`packed_forward_available` and hotness-1 dense inputs (or 2-D
-        with a combiner).
+        Zero-copy relative to :meth:`forward`: ``packed`` is
+        [P, b, width] feature-major — at world==1 a view of the fused-group
+        lookup output, at world>1 a view of the mp->dp all-to-all recv
+        buffer.  Packed rows are in worker order (see :meth:`packed_order`);
+        the per-pair split + stack/merge copies of the general path
+        disappear.  Requires :meth:`packed_forward_available` and hotness-1
+        dense inputs (or 2-D with a combiner).  (``DE_PACKED_SMAJ=1``
+        switches world==1 to a sample-major [b, P, width] layout — measured
+        slower end-to-end, kept for experiments; hence the boolean in the
+        return value.)
         """
         plan = self.strategy
         W = self.world_size
