@@ -265,6 +265,14 @@ class DistributedEmbedding(nn.Module):
         plan = self.strategy
         self._output_dtype = output_dtype
         dp_in, col_in, row_in = plan.input_groups
+        for x in inputs:
+            if isinstance(x, torch.Tensor) and x.layout == torch.sparse_coo:
+                # parity: the reference rejects SparseTensor inputs to the
+                # distributed wrapper (dist_model_parallel.py:263-265)
+                raise ValueError(
+                    "sparse COO inputs are not supported by "
+                    "DistributedEmbedding — convert to Ragged "
+                    "(single-table Embedding layers do accept sparse)")
         if self.dp_input:
             if len(inputs) != len(plan.input_table_map):
                 raise ValueError("wrong number of inputs")

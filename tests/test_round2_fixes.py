@@ -347,3 +347,14 @@ def test_forward_packed_world4():
 def test_async_redistribute_world4():
     r = run_distributed(_overlap_worker, world=4)
     assert all(o["match"] for o in r)
+
+
+def test_sparse_input_rejected_by_wrapper():
+    """Parity: the reference rejects SparseTensor inputs to the distributed
+    wrapper (dist_model_parallel.py:263-265); single-table layers accept."""
+    import distributed_embeddings_amd as de
+    m = de.DistributedEmbedding([de.TableConfig(10, 4, "sum")])
+    sp = torch.sparse_coo_tensor(torch.tensor([[0, 0], [1, 1]]),
+                                 torch.tensor([1, 2]), (2, 3))
+    with pytest.raises(ValueError, match="sparse COO"):
+        m([sp])
