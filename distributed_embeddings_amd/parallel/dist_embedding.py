@@ -74,7 +74,26 @@ def _layer_to_config(layer) -> TableConfig:
         import dataclasses
         fields = {f.name for f in dataclasses.fields(TableConfig)}
         return TableConfig(**{k: v for k, v in layer.items() if k in fields})
+    # custom user layers, duck-typed via get_config() or attributes (parity:
+    # reference accepts any layer whose get_config() yields
+    # input_dim/output_dim — dist_model_parallel_test.py:50-68, 501-511)
+    if hasattr(layer, "get_config"):
+        cfg = layer.get_config()
+        if isinstance(cfg, dict) and "input_dim" in cfg and "output_dim" in cfg:
+            return TableConfig(int(cfg["input_dim"]), int(cfg["output_dim"]),
+                               cfg.get("combiner"))
+    if hasattr(layer, "input_dim") and hasattr(layer, "output_dim"):
+        return TableConfig(int(layer.input_dim), int(layer.output_dim),
+                           getattr(layer, "combiner", None))
     raise TypeError(f"unsupported layer/config type {type(layer)}")
+
+
+def _layer_weight(obj, rows, cols):
+    """The handed-in layer's weight, if it looks like an embedding table."""
+    w = getattr(obj, "weight", None)
+    if isinstance(w, torch.Tensor) and tuple(w.shape) == (rows, cols):
+        return w
+    return None
 
 
 class DistributedEmbedding(nn.Module):
@@ -139,10 +158,10 @@ class DistributedEmbedding(nn.Module):
         # layers — same contract as the col-group copy below).
         with torch.no_grad():
             for local_t, t in enumerate(plan.dp_table_ids):
-                src = embeddings[t] if isinstance(
-                    embeddings[t], (Embedding, nn.Embedding)) else None
-                if src is not None:
-                    self.dp_layers[local_t].weight.copy_(src.weight)
+                w = _layer_weight(embeddings[t], configs[t].input_dim,
+                                  configs[t].output_dim)
+                if w is not None:
+                    self.dp_layers[local_t].weight.copy_(w)
 
         # ---- column/table-parallel fused variables ----
         # The fused group runs at EVERY world size (world==1 included): one
@@ -169,11 +188,11 @@ class DistributedEmbedding(nn.Module):
                     cfg = configs[m.table_id]
                     dst = lyr.weight[m.concat_row_offset:
                                      m.concat_row_offset + cfg.input_dim]
-                    src_layer = embeddings[m.table_id] if isinstance(
-                        embeddings[m.table_id], (Embedding, nn.Embedding)) else None
-                    if src_layer is not None:
-                        dst.copy_(src_layer.weight[:, m.col_offset:
-                                                   m.col_offset + m.width])
+                    src_w = _layer_weight(embeddings[m.table_id],
+                                          cfg.input_dim, cfg.output_dim)
+                    if src_w is not None:
+                        dst.copy_(src_w[:, m.col_offset:
+                                        m.col_offset + m.width])
                     elif cfg.initializer is not None:
                         if m.width == cfg.output_dim:
                             if dst.dtype == torch.float32:
@@ -200,12 +219,12 @@ class DistributedEmbedding(nn.Module):
             lyr._oob_zero = True
             for p in lyr.parameters():
                 p.de_local = True
-            src = embeddings[shard.table_id] if isinstance(
-                embeddings[shard.table_id], (Embedding, nn.Embedding)) else None
-            if src is not None and shard.rows > 0:
+            src_w = _layer_weight(embeddings[shard.table_id], cfg.input_dim,
+                                  cfg.output_dim)
+            if src_w is not None and shard.rows > 0:
                 with torch.no_grad():
                     lyr.weight[:shard.rows].copy_(
-                        src.weight[shard.row_offset:shard.row_offset + shard.rows])
+                        src_w[shard.row_offset:shard.row_offset + shard.rows])
             row_layers.append(lyr)
         self.row_layers = nn.ModuleList(row_layers)
 

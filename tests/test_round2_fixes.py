@@ -358,3 +358,30 @@ def test_sparse_input_rejected_by_wrapper():
                                  torch.tensor([1, 2]), (2, 3))
     with pytest.raises(ValueError, match="sparse COO"):
         m([sp])
+
+
+class _CustomEmbedding(torch.nn.Module):
+    """User layer duck-typed like the reference CustomEmbedding
+    (dist_model_parallel_test.py:50-68)."""
+
+    def __init__(self, input_dim, output_dim):
+        super().__init__()
+        self.weight = torch.nn.Parameter(torch.randn(input_dim, output_dim))
+
+    def get_config(self):
+        return {"input_dim": self.weight.shape[0],
+                "output_dim": self.weight.shape[1]}
+
+    def forward(self, ids):
+        return self.weight[ids]
+
+
+def test_custom_user_layer_accepted():
+    """Parity: custom layers with a get_config() are sharded like any table
+    and their weights are preserved (reference :501-511)."""
+    import distributed_embeddings_amd as de
+    lyr = _CustomEmbedding(30, 8)
+    model = de.DistributedEmbedding([lyr, de.TableConfig(40, 8, None)])
+    ids = torch.randint(0, 30, (4,))
+    outs = model([ids, torch.randint(0, 40, (4,))])
+    assert torch.allclose(outs[0], lyr.weight[ids], atol=1e-6)
