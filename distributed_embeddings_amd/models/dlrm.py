@@ -131,6 +131,7 @@ class DLRM(nn.Module):
                 persistent=False)
         else:
             self._dot_perm = None
+        self._emb_stream = None  # lazy side stream (world==1 overlap)
 
     def local_cat_feature_ids(self) -> List[int]:
         if self.distributed and not self.dp_input:
@@ -146,9 +147,33 @@ class DLRM(nn.Module):
             # post the dp->mp ID all-to-all before the bottom MLP: the xGMI
             # id exchange overlaps the MLP GEMMs (ids carry no grad)
             handle = self.embeddings.redistribute_async(cats)
-        bottom = self.bottom_mlp(numerical)
         use_packed = self._dot_perm is not None and all(
             isinstance(x, torch.Tensor) and x.dim() == 1 for x in cats)
+        want = torch.bfloat16 if (numerical.is_cuda and
+                                  torch.is_autocast_enabled()) \
+            else numerical.dtype
+        if use_packed and not self.distributed and numerical.is_cuda and \
+                os.environ.get("DE_STREAM_OVERLAP", "1") != "0":
+            # world==1: the (bandwidth-bound) lookup runs on a side stream
+            # concurrently with the (MFMA-bound) bottom MLP; autograd keeps
+            # the same streams in backward, overlapping the table update
+            # with the MLP grads as well.  hipGraph capture spans the fork.
+            if self._emb_stream is None:
+                self._emb_stream = torch.cuda.Stream()
+            main = torch.cuda.current_stream()
+            self._emb_stream.wait_stream(main)
+            with torch.cuda.stream(self._emb_stream):
+                packed, smaj = self.embeddings.forward_packed(
+                    cats, output_dtype=want)
+            bottom = self.bottom_mlp(numerical)
+            main.wait_stream(self._emb_stream)
+            if not torch.cuda.is_current_stream_capturing():
+                packed.record_stream(main)
+            x = dot_interact_packed(packed, bottom, self._dot_perm,
+                                    pad_to=self.interact_pad,
+                                    sample_major=smaj)
+            return self.top_mlp(x)
+        bottom = self.bottom_mlp(numerical)
         if use_packed:
             packed, smaj = self.embeddings.forward_packed(
                 cats, output_dtype=bottom.dtype, async_handle=handle)
