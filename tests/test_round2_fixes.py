@@ -385,3 +385,57 @@ def test_custom_user_layer_accepted():
     ids = torch.randint(0, 30, (4,))
     outs = model([ids, torch.randint(0, 40, (4,))])
     assert torch.allclose(outs[0], lyr.weight[ids], atol=1e-6)
+
+
+# ------------------------------------------------- dynamic shapes / eval mode
+
+def _dynamic_batch_worker(rank, world):
+    """Consecutive calls with CHANGING batch sizes: every shape-keyed cache
+    (dp->mp splits, pair-cols exchange, offset/splits vectors) must re-key,
+    at world>1 consistently across ranks (serving batches vary)."""
+    import distributed_embeddings_amd as de
+    sizes = [40, 60, 25]
+    tables = [de.TableConfig(s, 8, "sum") for s in sizes]
+    model = de.DistributedEmbedding(tables, strategy="basic")
+    g = torch.Generator().manual_seed(5)
+    weights = [torch.randn(s, 8, generator=g) for s in sizes]
+    model.set_weights([w.numpy() for w in weights])
+    ok = True
+    for b in (4, 6, 2, 6, 4):
+        gi = torch.Generator().manual_seed(100 + b)
+        full = [torch.randint(0, s, (world * b, 2), generator=gi) for s in sizes]
+        local = [x[rank * b:(rank + 1) * b] for x in full]
+        outs = model(local)
+        for t in range(3):
+            ref = weights[t][full[t]].sum(1)[rank * b:(rank + 1) * b]
+            ok = ok and torch.allclose(outs[t], ref, atol=1e-5)
+    return ok
+
+
+def test_dynamic_batch_sizes_world2():
+    assert all(run_distributed(_dynamic_batch_worker, world=2))
+
+
+def test_eval_mode_matches_train_outputs():
+    """eval() + no_grad forward == training-mode forward values (packed and
+    general paths; fused-optimizer layers must NOT update in eval)."""
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd.models.dlrm import DLRM
+    torch.manual_seed(0)
+    sizes = [30, 40, 50]
+    m = DLRM(sizes, embedding_dim=32, bottom_mlp_dims=(16, 32),
+             top_mlp_dims=(16, 1), num_numerical=4, strategy="basic")
+    m.embeddings.enable_fused_sgd(0.5)
+    num = torch.rand(4, 4)
+    cats = [torch.randint(0, s, (4,)) for s in sizes]
+    m.train()
+    w_before = [w.copy() for w in m.embeddings.get_weights()]
+    with torch.no_grad():
+        o_train = m(num, cats)
+    m.eval()
+    with torch.no_grad():
+        o_eval = m(num, cats)
+    assert torch.allclose(o_train, o_eval, atol=1e-6)
+    w_after = m.embeddings.get_weights()
+    for a, b in zip(w_before, w_after):
+        assert (a == b).all()  # no updates without backward
