@@ -439,3 +439,28 @@ def test_eval_mode_matches_train_outputs():
     w_after = m.embeddings.get_weights()
     for a, b in zip(w_before, w_after):
         assert (a == b).all()  # no updates without backward
+
+
+# ------------------------------------------- distributed IntegerLookup (dp)
+
+def _integer_lookup_dp_worker(rank, world):
+    """BASELINE config #5 shape: IntegerLookup is a data-parallel layer —
+    build the vocab on rank 0, broadcast the hash buffers, then every rank
+    resolves the same keys to the same values."""
+    import distributed_embeddings_amd as de
+    lk = de.IntegerLookup(max_tokens=100)
+    if rank == 0:
+        lk(torch.tensor([111, 222, 333, 444]))
+    de.broadcast_parameters(lk)
+    out = lk(torch.tensor([444, 111, 333]))
+    # a NEW key after the broadcast gets the next free value consistently
+    v_new = int(lk(torch.tensor([999]))[0])
+    return {"out": out.tolist(), "new": v_new,
+            "vocab": lk.get_vocabulary()}
+
+
+def test_integer_lookup_dp_world2():
+    r = run_distributed(_integer_lookup_dp_worker, world=2)
+    assert r[0]["out"] == r[1]["out"] == [4, 1, 3]
+    assert r[0]["new"] == r[1]["new"] == 5
+    assert r[0]["vocab"] == r[1]["vocab"]
