@@ -164,3 +164,17 @@ def test_dlrm_example_with_binary_dataset(tmp_path):
                                 "--table-size-cap", str(cap),
                                 "--dataset-path", str(tmp_path)])
     assert "loss" in out
+
+
+def test_serving_example_smoke(tmp_path):
+    """Serving demo runs on CPU with capped tables (latency path + shapes)."""
+    import subprocess
+    import sys
+    import pathlib
+    ex = pathlib.Path(__file__).parent.parent / "examples" / "serving.py"
+    r = subprocess.run(
+        [sys.executable, str(ex), "--table-size-cap", "500",
+         "--batch-size", "32", "--iters", "3", "--warmup", "1"],
+        capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0, r.stderr
+    assert "p50" in r.stdout
