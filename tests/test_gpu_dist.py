@@ -272,3 +272,40 @@ def test_hybrid_all_modes_world2_gpu():
     outs = run_gpu2(_hybrid_all_modes_worker)
     for o in outs:
         assert max(o["errs"]) < 1e-4, o
+
+
+def _dlrm_world2_worker(rank, world):
+    """Full DLRM at world 2 with CUDA tensors: packed feature-major MFMA
+    interaction + async id overlap + staged collectives — the exact path the
+    multi-GPU SCALE bench runs (modulo gloo-vs-RCCL transport)."""
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd.models.dlrm import DLRM
+    torch.manual_seed(0)
+    sizes = [500, 600, 700, 400, 300]
+    with torch.device("cuda"):
+        m = DLRM(sizes, embedding_dim=128, bottom_mlp_dims=(64, 128),
+                 top_mlp_dims=(64, 1), num_numerical=4,
+                 strategy="memory_balanced")
+    de.broadcast_parameters(m)
+    assert m._dot_perm is not None  # packed path active
+    g = torch.Generator().manual_seed(7)
+    B = world * 8
+    num = torch.rand(B, 4, generator=g).cuda()
+    cats = [torch.randint(0, s, (B,), generator=g).cuda() for s in sizes]
+    sl = slice(rank * 8, (rank + 1) * 8)
+    opt = de.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.01), average=False)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out = m(num[sl], [c[sl] for c in cats])
+    (out.float().square().sum() / B).backward()
+    opt.step()
+    return out.detach().float().cpu()
+
+
+def test_dlrm_world2_gpu_matches_world1():
+    o2 = run_gpu2(_dlrm_world2_worker)
+    o1 = run_gpu2(_dlrm_world2_worker, world=1)[0]
+    for rank in range(2):
+        ref = o1[rank * 8:(rank + 1) * 8]
+        err = float((o2[rank] - ref).abs().max())
+        assert err < 0.05, err  # bf16 autocast tolerance
