@@ -289,10 +289,11 @@ def _dlrm_world2_worker(rank, world):
     de.broadcast_parameters(m)
     assert m._dot_perm is not None  # packed path active
     g = torch.Generator().manual_seed(7)
-    B = world * 8
+    B = 16  # fixed GLOBAL batch so world-1 and world-2 runs are comparable
+    lb = B // world
     num = torch.rand(B, 4, generator=g).cuda()
     cats = [torch.randint(0, s, (B,), generator=g).cuda() for s in sizes]
-    sl = slice(rank * 8, (rank + 1) * 8)
+    sl = slice(rank * lb, (rank + 1) * lb)
     opt = de.DistributedOptimizer(
         torch.optim.SGD(m.parameters(), lr=0.01), average=False)
     with torch.autocast("cuda", dtype=torch.bfloat16):
