@@ -374,7 +374,7 @@ def test_distributed_embedding_module_gpu_world1():
             weights[3][x3.cpu()]]
     for t in (0, 2, 3):
         err = (outs[t].float().cpu() - refs[t]).abs().max()
-        assert float(err) < 0.5, f"table {t}: {err}"
+        assert float(err.detach()) < 0.5, f"table {t}: {err}"
     # ragged row check
     from distributed_embeddings_amd.ops.embedding_lookup import _csr_lookup_ref
     ref1 = _csr_lookup_ref(weights[1], vals.cpu(),
