@@ -152,19 +152,20 @@ class DLRM(nn.Module):
         want = torch.bfloat16 if (numerical.is_cuda and
                                   torch.is_autocast_enabled()) \
             else numerical.dtype
-        if use_packed and not self.distributed and numerical.is_cuda and \
+        if use_packed and self.dp_input and numerical.is_cuda and \
                 os.environ.get("DE_STREAM_OVERLAP", "1") != "0":
-            # world==1: the (bandwidth-bound) lookup runs on a side stream
-            # concurrently with the (MFMA-bound) bottom MLP; autograd keeps
-            # the same streams in backward, overlapping the table update
-            # with the MLP grads as well.  hipGraph capture spans the fork.
+            # The (bandwidth-bound) lookup chain — and at world>1 the mp->dp
+            # output all-to-all — runs on a side stream concurrently with
+            # the (MFMA-bound) bottom MLP; autograd keeps the same streams
+            # in backward, so the reverse all-to-all + table update overlap
+            # the MLP grads too.  hipGraph capture spans the fork (world==1).
             if self._emb_stream is None:
                 self._emb_stream = torch.cuda.Stream()
             main = torch.cuda.current_stream()
             self._emb_stream.wait_stream(main)
             with torch.cuda.stream(self._emb_stream):
                 packed, smaj = self.embeddings.forward_packed(
-                    cats, output_dtype=want)
+                    cats, output_dtype=want, async_handle=handle)
             bottom = self.bottom_mlp(numerical)
             main.wait_stream(self._emb_stream)
             if not torch.cuda.is_current_stream_capturing():
