@@ -128,7 +128,12 @@ class DistributedOptimizer:
     """Wraps an optimizer: averages dp grads across ranks before each step.
 
     Parity: reference ``DistributedOptimizer`` (``:1270-1300``) — mp variables
-    (marked ``de_local``) stay local; everything else allreduces.
+    (marked ``de_local``) stay local; everything else allreduces.  With
+    ``overlap=True`` (default) the allreduce of each bucket is posted
+    asynchronously as soon as its gradients are accumulated during backward
+    (Horovod-tape-style overlap); ``step()`` drains the in-flight buckets.
+    The overlap path assumes ONE backward per step — for gradient
+    accumulation across several backwards pass ``overlap=False``.
 
     Usage::
 
@@ -137,10 +142,62 @@ class DistributedOptimizer:
     """
 
     def __init__(self, optimizer: torch.optim.Optimizer, bucket_bytes: int = 64 << 20,
-                 average: bool = True):
+                 average: bool = True, overlap: bool = True):
         self.optimizer = optimizer
         self.bucket_bytes = bucket_bytes
         self.average = average
+        self.overlap = overlap and comm.world_size() > 1
+        self._bucket: List[torch.nn.Parameter] = []
+        self._bucket_bytes_now = 0
+        self._inflight = []  # (work, flat, params)
+        self._seen = set()
+        if self.overlap:
+            for g in self.optimizer.param_groups:
+                for p in g["params"]:
+                    if p.requires_grad and not is_local_param(p):
+                        p.register_post_accumulate_grad_hook(self._on_grad_ready)
+
+    def _on_grad_ready(self, p):
+        if id(p) in self._seen or p.grad is None:
+            return
+        self._seen.add(id(p))
+        g = p.grad
+        if g.layout != torch.strided:
+            p.grad = g = g.to_dense()  # sparse_as_dense parity
+        self._bucket.append(p)
+        self._bucket_bytes_now += g.numel() * g.element_size()
+        if self._bucket_bytes_now >= self.bucket_bytes:
+            self._flush_bucket()
+
+    def _flush_bucket(self):
+        if not self._bucket:
+            return
+        # one flat buffer per (dtype, device) in the bucket
+        groups = {}
+        for p in self._bucket:
+            groups.setdefault((p.grad.dtype, p.grad.device), []).append(p)
+        for ps in groups.values():
+            dev = comm.backend_device()
+            flat = torch.cat([p.grad.reshape(-1) for p in ps]).to(dev)
+            work = dist.all_reduce(flat, async_op=True)
+            self._inflight.append((work, flat, ps))
+        self._bucket = []
+        self._bucket_bytes_now = 0
+
+    def _drain(self):
+        self._flush_bucket()
+        world = comm.world_size()
+        for work, flat, ps in self._inflight:
+            work.wait()
+            if self.average:
+                flat /= world
+            pos = 0
+            for p in ps:
+                n = p.grad.numel()
+                p.grad.copy_(flat[pos:pos + n].view_as(p.grad))
+                pos += n
+        self._inflight = []
+        self._seen = set()
 
     @property
     def param_groups(self):
@@ -156,8 +213,14 @@ class DistributedOptimizer:
         self.optimizer.zero_grad(set_to_none=set_to_none)
 
     def step(self, closure=None):
-        params = [p for g in self.optimizer.param_groups for p in g["params"]]
-        allreduce_gradients(params, self.bucket_bytes, self.average)
+        if self.overlap:
+            self._drain()
+            # params whose hooks never fired this step (unused / no grad)
+            # plus de_local exclusions are already correct: nothing to do
+        else:
+            params = [p for g in self.optimizer.param_groups
+                      for p in g["params"]]
+            allreduce_gradients(params, self.bucket_bytes, self.average)
         return self.optimizer.step(closure)
 
 
