@@ -51,6 +51,10 @@ def parse_args():
     p.add_argument("--learnable-labels", action="store_true",
                    help="synthetic labels = f(ids): loss must drop below "
                         "ln(2) if the optimizer scaling is right")
+    p.add_argument("--fused-optimizer", action="store_true",
+                   help="in-backward fused SGD on the model-parallel tables "
+                        "(no grad tensors, no host syncs; the schedule "
+                        "updates the device-resident lr)")
     return p.parse_args()
 
 
@@ -119,13 +123,18 @@ def main():
     # (average=False), so each grad is already the global-batch mean —
     # lr is used as-is (parity: reference lr=24 with a mean loss,
     # examples/dlrm/main.py).
+    fused_modules = []
+    if args.fused_optimizer:
+        model.embeddings.enable_fused_sgd(args.learning_rate)
+        fused_modules.append(model.embeddings)
     opt = de.DistributedOptimizer(SparseEmbeddingOptimizer(
         model.parameters(), lr=args.learning_rate, method="sgd"),
         average=False)
     sched = WarmupPolyDecay(opt, base_lr=args.learning_rate,
                             warmup_steps=args.warmup_steps,
                             decay_start=args.decay_start,
-                            decay_steps=args.decay_steps)
+                            decay_steps=args.decay_steps,
+                            fused_modules=fused_modules)
     de.broadcast_parameters(model)
     _loss_sum = torch.nn.BCEWithLogitsLoss(reduction="sum")
     loss_fn = lambda lg, lb: _loss_sum(lg, lb) / args.batch_size
