@@ -144,13 +144,16 @@ def main():
     for step, (num, cats, labels) in enumerate(data):
         if step >= args.num_batches:
             break
+        # set THIS step's lr before forward/backward: fused in-backward
+        # updates consume the device-resident lr during backward, and the
+        # dense optimizer reads it at opt.step() (warmup starts at step 0)
+        sched.step()
         opt.zero_grad(set_to_none=True)
         with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
                             enabled=torch.cuda.is_available()):
             logits = model(num, cats)
             loss = loss_fn(logits.float(), labels)
         loss.backward()
-        sched.step()   # set THIS step's lr (warmup starts at step 0)
         opt.step()
         if rank == 0 and step % 50 == 0:
             l = float(loss.detach())
