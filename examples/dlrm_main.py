@@ -51,6 +51,7 @@ def parse_args():
     p.add_argument("--learnable-labels", action="store_true",
                    help="synthetic labels = f(ids): loss must drop below "
                         "ln(2) if the optimizer scaling is right")
+    p.add_argument("--seed", type=int, default=1234)
     p.add_argument("--fused-optimizer", action="store_true",
                    help="in-backward fused SGD on the model-parallel tables "
                         "(no grad tensors, no host syncs; the schedule "
@@ -97,6 +98,7 @@ def main():
     if torch.cuda.is_available():
         torch.cuda.set_device(local_rank)
 
+    torch.manual_seed(args.seed + rank)
     table_sizes = CRITEO_1TB_TABLE_SIZES
     if args.table_size_cap:
         table_sizes = [min(s, args.table_size_cap) for s in table_sizes]
