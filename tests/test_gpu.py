@@ -802,3 +802,26 @@ def test_csr_backward_bf16_grad_mean():
     u2, g2 = ext.csr_lookup_backward(g16.float(), ids, splits, vocab, True)
     assert torch.equal(u1, u2)
     assert torch.allclose(g1, g2, atol=1e-5), float((g1 - g2).abs().max())
+
+
+@requires_gpu
+def test_csr_forward_bf16_out_mega_segment():
+    """bf16-out disables the long-segment split (fp32 atomics unavailable);
+    a power-law mega-row must still reduce correctly in-register."""
+    from distributed_embeddings_amd.ops import _backend
+    ext = _backend.ops()
+    torch.manual_seed(3)
+    vocab, width = 100, 128
+    w = torch.randn(vocab, width, device="cuda")
+    # row 0: 50k ids (mega-segment), rows 1..64: hotness 3
+    big = torch.randint(0, vocab, (50_000,), device="cuda")
+    rest = torch.randint(0, vocab, (64 * 3,), device="cuda")
+    ids = torch.cat([big, rest])
+    splits = torch.cat([torch.tensor([0], device="cuda"),
+                        torch.tensor([50_000], device="cuda"),
+                        50_000 + 3 * torch.arange(1, 65, device="cuda")])
+    out16 = ext.csr_lookup_forward(w, ids, splits, True, True)  # mean, bf16
+    out32 = ext.csr_lookup_forward(w, ids, splits, True)
+    assert out16.dtype == torch.bfloat16
+    err = (out16.float() - out32).abs().max()
+    assert float(err) < 0.02, float(err)
