@@ -66,7 +66,11 @@ def _tp_dense_worker(rank, world):
     loss.backward()
     refs = [weights[t][full[t]].sum(1)[rank * 4:(rank + 1) * 4] for t in range(3)]
     errs = [float((o.float().cpu() - r).abs().max()) for o, r in zip(outs, refs)]
-    return {"errs": errs}
+    # chunked column-slice checkpoint reassembly with device buffers
+    got = model.get_weights(all_ranks=True, chunk_elements=512)
+    werrs = [float((torch.as_tensor(a) - w).abs().max())
+             for a, w in zip(got, weights)]
+    return {"errs": errs, "werrs": werrs}
 
 
 def _tp_ragged_worker(rank, world):
@@ -236,6 +240,7 @@ def test_tp_dense_bf16_a2a_world2_gpu():
     outs = run_gpu2(_tp_dense_worker)
     for o in outs:
         assert max(o["errs"]) < 0.1, o  # bf16 round-trip tolerance
+        assert max(o["werrs"]) < 1e-5, o
 
 
 @requires_gpu
