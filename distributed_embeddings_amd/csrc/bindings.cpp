@@ -466,10 +466,17 @@ std::vector<torch::Tensor> dot_interact_bwd_packed(torch::Tensor gout,
                                                    bool sample_major) {
   CHECK_CUDA(gout); CHECK_CUDA(bottom); CHECK_CUDA(packed); CHECK_CUDA(perm);
   CHECK_CONTIG(gout); CHECK_CONTIG(bottom); CHECK_CONTIG(packed);
-  TORCH_CHECK(gout.dtype() == torch::kBFloat16);
+  CHECK_CONTIG(perm);
+  TORCH_CHECK(gout.dtype() == torch::kBFloat16 &&
+              bottom.dtype() == torch::kBFloat16 &&
+              packed.dtype() == torch::kBFloat16, "bf16 required");
+  TORCH_CHECK(perm.dtype() == torch::kInt32, "perm must be int32");
   const int64_t B = bottom.size(0);
   const int D = (int)bottom.size(1);
   const int P = (int)(sample_major ? packed.size(1) : packed.size(0));
+  TORCH_CHECK(perm.numel() == P, "perm must have P entries");
+  TORCH_CHECK(packed.size(sample_major ? 0 : 1) == B && packed.size(2) == D,
+              "shape mismatch");
   const int F = P + 1;
   const int tri_n = F * (F - 1) / 2;
   const int64_t sb = sample_major ? P : 1;
