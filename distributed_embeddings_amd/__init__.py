@@ -23,6 +23,8 @@ from .parallel.grad import (
     broadcast_variables,
 )
 from .parallel import comm
+from .parallel.checkpoint import (load_embedding_checkpoint,
+                                  save_embedding_checkpoint)
 
 __version__ = "0.1.0"
 
@@ -42,6 +44,8 @@ __all__ = [
     "BroadcastParametersOnFirstStep",
     "allreduce_gradients",
     "broadcast_parameters",
+    "save_embedding_checkpoint",
+    "load_embedding_checkpoint",
     "broadcast_variables",
     "comm",
 ]

@@ -211,3 +211,39 @@ def test_get_weights_rank0_only_world2():
     deadlock, rank 0 gets the full tables."""
     results = run_distributed(_get_weights_rank0_worker, world=2)
     assert results[0] < 1e-6
+
+
+def _ckpt_dir_worker(rank, world, tmpdir):
+    """save_embedding_checkpoint -> load into a DIFFERENTLY-sharded model
+    (world stays, strategy changes) == original tables."""
+    import distributed_embeddings_amd as de
+    sizes = [50, 700, 33]
+    tables = [de.TableConfig(s, 8, "sum") for s in sizes]
+    m1 = de.DistributedEmbedding(tables, strategy="basic",
+                                 row_slice_threshold=5000,
+                                 data_parallel_threshold=300)
+    g = torch.Generator().manual_seed(23)
+    weights = [torch.randn(s, 8, generator=g) for s in sizes]
+    m1.set_weights([w.numpy() for w in weights])
+    de.save_embedding_checkpoint(m1, tmpdir, chunk_elements=64)
+    m2 = de.DistributedEmbedding(tables, strategy="memory_balanced")
+    de.load_embedding_checkpoint(m2, tmpdir, chunk_elements=64)
+    out = m2.get_weights(all_ranks=True)
+    return [torch.as_tensor(w) for w in out]
+
+
+def test_checkpoint_dir_roundtrip_world2(tmp_path):
+    outs = run_distributed(_ckpt_dir_worker, world=2, args=(str(tmp_path),))
+    g = torch.Generator().manual_seed(23)
+    expect = [torch.randn(s, 8, generator=g) for s in [50, 700, 33]]
+    for o in outs:
+        for got, want in zip(o, expect):
+            assert torch.equal(got, want)
+
+
+def test_checkpoint_dir_roundtrip_world1(tmp_path):
+    out = _ckpt_dir_worker(0, 1, str(tmp_path))
+    g = torch.Generator().manual_seed(23)
+    for got, want in zip(out, [torch.randn(s, 8, generator=g)
+                               for s in [50, 700, 33]]):
+        assert torch.equal(got, want)
