@@ -43,6 +43,9 @@ def parse_args():
     p.add_argument("--eval", action="store_true", help="run AUC eval at end")
     p.add_argument("--dump-embeddings", default=None,
                    help="npz path: dump full tables via get_weights")
+    p.add_argument("--checkpoint-dir", default=None,
+                   help="directory: save per-table .npy via "
+                        "save_embedding_checkpoint (reshardable on load)")
     p.add_argument("--warmup-steps", type=int, default=8000)
     p.add_argument("--decay-start", type=int, default=70000)
     p.add_argument("--decay-steps", type=int, default=30000)
@@ -185,6 +188,10 @@ def main():
         if rank == 0:
             np.savez(args.dump_embeddings, *weights)
             print(f"embeddings dumped to {args.dump_embeddings}")
+    if args.checkpoint_dir:
+        de.save_embedding_checkpoint(model.embeddings, args.checkpoint_dir)
+        if rank == 0:
+            print(f"checkpoint written to {args.checkpoint_dir}")
 
 
 if __name__ == "__main__":

@@ -32,7 +32,8 @@ def parse_args():
     p.add_argument("--iters", type=int, default=200)
     p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--weights", default=None,
-                   help="npz from dlrm_main --dump-embeddings (else random)")
+                   help="npz from dlrm_main --dump-embeddings, or a "
+                        "checkpoint directory (table_*.npy; mmap-loaded)")
     p.add_argument("--table-dtype", default="fp32", choices=["fp32", "bf16"])
     p.add_argument("--table-size-cap", type=int, default=None)
     p.add_argument("--graph", action="store_true",
@@ -51,8 +52,12 @@ def main():
         model = DLRM(sizes, embedding_dim=128, table_dtype=tdt)
     model.eval()
     if args.weights:
-        with np.load(args.weights) as z:
-            model.embeddings.set_weights([z[k] for k in z.files])
+        if os.path.isdir(args.weights):
+            from distributed_embeddings_amd import load_embedding_checkpoint
+            load_embedding_checkpoint(model.embeddings, args.weights)
+        else:
+            with np.load(args.weights) as z:
+                model.embeddings.set_weights([z[k] for k in z.files])
 
     b = args.batch_size
     g = torch.Generator().manual_seed(7)
