@@ -323,7 +323,12 @@ class DistributedEmbedding(nn.Module):
         if isinstance(w, str):
             w = np.load(w, mmap_mode="r")  # parity: reference mmap path :911,919,950
         if isinstance(w, np.ndarray):
-            w = torch.from_numpy(np.ascontiguousarray(w))
+            import warnings
+            with warnings.catch_warnings():
+                # mmap'd checkpoints are read-only; the tensor is only ever a
+                # copy_ SOURCE here, so the non-writable warning is moot
+                warnings.simplefilter("ignore", UserWarning)
+                w = torch.from_numpy(np.ascontiguousarray(w))
         return w
 
     def set_weights(self, weights: Sequence, chunk_elements: int = 128 * 1024 * 1024):
