@@ -825,3 +825,35 @@ def test_csr_forward_bf16_out_mega_segment():
     assert out16.dtype == torch.bfloat16
     err = (out16.float() - out32).abs().max()
     assert float(err) < 0.02, float(err)
+
+
+@requires_gpu
+def test_inference_graph_capture_matches_eager():
+    """Serving path: hipGraph-captured eval scoring == eager scoring."""
+    import distributed_embeddings_amd as de
+    from distributed_embeddings_amd.models.dlrm import DLRM
+    torch.manual_seed(0)
+    sizes = [500, 600, 700]
+    with torch.device("cuda"):
+        m = DLRM(sizes, embedding_dim=128, bottom_mlp_dims=(32, 128),
+                 top_mlp_dims=(32, 1), num_numerical=4)
+    m.eval()
+    num = torch.rand(256, 4, device="cuda")
+    cats = [torch.randint(0, s, (256,), device="cuda") for s in sizes]
+
+    def score():
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            return torch.sigmoid(m(num, cats).float())
+
+    with torch.inference_mode():
+        eager = score().clone()
+        for _ in range(3):
+            out = score()
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            out = score()
+        g.replay()
+        torch.cuda.synchronize()
+    assert torch.allclose(out, eager, atol=1e-3), \
+        float((out - eager).abs().max())
