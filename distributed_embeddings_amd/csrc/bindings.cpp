@@ -44,6 +44,8 @@ void sort_ids_dispatch(torch::Tensor values, int64_t vocab, int64_t nnz,
                        int end_bit, torch::Tensor sorted_ids,
                        torch::Tensor sorted_pos, hipStream_t stream) {
   TORCH_CHECK(end_bit <= 32, "fused table vocab must fit 32 bits");
+  TORCH_CHECK(nnz < (int64_t(1) << 32),
+              "sort packs positions into 32 bits: nnz must be < 2^32");
   auto u64 = values.options().dtype(torch::kUInt64);
   auto i32 = values.options().dtype(torch::kInt32);
   auto packed = torch::empty({nnz}, u64);
